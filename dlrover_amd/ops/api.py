@@ -1,0 +1,325 @@
+"""Autograd-wrapped ops: HIP kernels on GPU, torch fp32 references on CPU."""
+
+from typing import Optional, Tuple
+
+import torch
+
+_EXT = None
+_EXT_ERR: Optional[str] = None
+
+
+class ExtensionMissingError(RuntimeError):
+    pass
+
+
+def hip_ops():
+    """Return the compiled extension, importing it lazily. Raises loudly on a
+    GPU box if the .so is missing (the driver checks native code is loaded)."""
+    global _EXT, _EXT_ERR
+    if _EXT is None and _EXT_ERR is None:
+        try:
+            from dlrover_amd.ops import _hip_ops as ext  # built in-tree
+
+            _EXT = ext
+        except ImportError as e:  # pragma: no cover - build problem
+            _EXT_ERR = str(e)
+    if _EXT is None:
+        raise ExtensionMissingError(
+            "dlrover_amd._hip_ops is not built — run "
+            "`PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace` "
+            f"(import error: {_EXT_ERR})"
+        )
+    return _EXT
+
+
+def hip_ops_available() -> bool:
+    try:
+        hip_ops()
+        return True
+    except ExtensionMissingError:
+        return False
+
+
+def _use_hip(*tensors: torch.Tensor) -> bool:
+    on_gpu = any(t.is_cuda for t in tensors if isinstance(t, torch.Tensor))
+    if on_gpu:
+        hip_ops()  # raises if missing: no silent eager fallback on GPU
+        return True
+    return False
+
+
+# ---------------------------------------------------------------------------
+# reference implementations (CPU path + GPU numerics oracle)
+# ---------------------------------------------------------------------------
+
+
+def rmsnorm_ref(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
+    xf = x.float()
+    inv = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    return (xf * inv * w.float()).to(x.dtype)
+
+
+def swiglu_ref(gate_up: torch.Tensor) -> torch.Tensor:
+    gate, up = gate_up.float().chunk(2, dim=-1)
+    return (torch.nn.functional.silu(gate) * up).to(gate_up.dtype)
+
+
+def rope_ref(
+    x: torch.Tensor, pos: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor
+) -> torch.Tensor:
+    """x [..., n_heads, D]; pos [n_tokens]; cos/sin [max_pos, D/2] fp32."""
+    d = x.shape[-1]
+    xf = x.float()
+    x1, x2 = xf[..., : d // 2], xf[..., d // 2 :]
+    shape = [1] * (x.dim() - 3) + [-1, 1, d // 2]
+    c = cos[pos].view(*shape)
+    s = sin[pos].view(*shape)
+    o1 = x1 * c - x2 * s
+    o2 = x2 * c + x1 * s
+    return torch.cat([o1, o2], dim=-1).to(x.dtype)
+
+
+def causal_softmax_ref(
+    scores: torch.Tensor, scale: float, q_offset: int = 0
+) -> torch.Tensor:
+    s = scores.float() * scale
+    q_len, k_len = s.shape[-2], s.shape[-1]
+    qpos = torch.arange(q_len, device=s.device).unsqueeze(-1) + q_offset
+    kpos = torch.arange(k_len, device=s.device).unsqueeze(0)
+    s = s.masked_fill(kpos > qpos, float("-inf"))
+    return torch.softmax(s, dim=-1).to(scores.dtype)
+
+
+# ---------------------------------------------------------------------------
+# autograd functions
+# ---------------------------------------------------------------------------
+
+
+class _RMSNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, eps):
+        if _use_hip(x):
+            y, invrms = hip_ops().rmsnorm_fwd(x.contiguous(), w.contiguous(), eps)
+            ctx.save_for_backward(x, w, invrms)
+            ctx.use_hip = True
+        else:
+            xf = x.float()
+            invrms = torch.rsqrt(xf.pow(2).mean(-1) + eps).reshape(-1)
+            y = rmsnorm_ref(x, w, eps)
+            ctx.save_for_backward(x, w, invrms)
+            ctx.use_hip = False
+        ctx.eps = eps
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, invrms = ctx.saved_tensors
+        if ctx.use_hip:
+            dx, dw = hip_ops().rmsnorm_bwd(dy.contiguous(), x, w, invrms)
+            return dx, dw, None
+        h = x.shape[-1]
+        xf, dyf, wf = x.float(), dy.float(), w.float()
+        inv = invrms.view(*x.shape[:-1], 1)
+        dot = (dyf * wf * xf).sum(-1, keepdim=True)
+        dx = dyf * wf * inv - xf * (dot * inv.pow(3) / h)
+        dw = (dyf * xf * inv).reshape(-1, h).sum(0)
+        return dx.to(x.dtype), dw.to(w.dtype), None
+
+
+def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    return _RMSNorm.apply(x, w, eps)
+
+
+class _SwiGLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gate_up):
+        ctx.save_for_backward(gate_up)
+        if _use_hip(gate_up):
+            return hip_ops().swiglu_fwd(gate_up.contiguous())
+        return swiglu_ref(gate_up)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (gate_up,) = ctx.saved_tensors
+        if _use_hip(gate_up):
+            return hip_ops().swiglu_bwd(dy.contiguous(), gate_up.contiguous())
+        gate, up = gate_up.float().chunk(2, dim=-1)
+        sig = torch.sigmoid(gate)
+        dyf = dy.float()
+        dg = dyf * up * sig * (1 + gate * (1 - sig))
+        du = dyf * gate * sig
+        return torch.cat([dg, du], dim=-1).to(gate_up.dtype)
+
+
+def swiglu(gate_up: torch.Tensor) -> torch.Tensor:
+    """out = silu(gate_up[..., :I]) * gate_up[..., I:]"""
+    return _SwiGLU.apply(gate_up)
+
+
+class _RoPE(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, pos, cos, sin):
+        ctx.save_for_backward(pos, cos, sin)
+        if _use_hip(x):
+            out = x.contiguous().clone()
+            hip_ops().rope_apply(out, pos.int(), cos, sin, False)
+            return out
+        return rope_ref(x, pos, cos, sin)
+
+    @staticmethod
+    def backward(ctx, dy):
+        pos, cos, sin = ctx.saved_tensors
+        if _use_hip(dy):
+            dx = dy.contiguous().clone()
+            hip_ops().rope_apply(dx, pos.int(), cos, sin, True)
+            return dx, None, None, None
+        # inverse rotation
+        return rope_ref(dy, pos, cos, -sin), None, None, None
+
+
+def rope_rotate(
+    x: torch.Tensor, pos: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor
+) -> torch.Tensor:
+    """Apply rotate-half RoPE. x: [..., n_tokens, n_heads, head_dim]."""
+    return _RoPE.apply(x, pos, cos, sin)
+
+
+def build_rope_cache(
+    max_pos: int, head_dim: int, base: float = 500000.0, device="cpu"
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Host-precomputed fp32 cos/sin tables [max_pos, head_dim/2]
+    (on-device trig would turn RoPE VALU-bound — guide Appendix B)."""
+    inv_freq = 1.0 / (
+        base ** (torch.arange(0, head_dim, 2, dtype=torch.float32) / head_dim)
+    )
+    t = torch.arange(max_pos, dtype=torch.float32)
+    freqs = torch.outer(t, inv_freq)
+    return freqs.cos().to(device), freqs.sin().to(device)
+
+
+class _CausalSoftmax(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, scores, scale, q_offset):
+        if _use_hip(scores):
+            probs = scores.contiguous()
+            hip_ops().causal_softmax_fwd(
+                probs, scores.shape[-2], q_offset, scale
+            )
+        else:
+            probs = causal_softmax_ref(scores, scale, q_offset)
+        ctx.save_for_backward(probs)
+        ctx.scale = scale
+        return probs
+
+    @staticmethod
+    def backward(ctx, dprobs):
+        (probs,) = ctx.saved_tensors
+        if _use_hip(probs):
+            ds = dprobs.contiguous().clone()
+            hip_ops().causal_softmax_bwd(ds, probs, ctx.scale)
+            return ds, None, None
+        pf, df = probs.float(), dprobs.float()
+        dot = (pf * df).sum(-1, keepdim=True)
+        return (pf * (df - dot) * ctx.scale).to(probs.dtype), None, None
+
+
+def causal_softmax(
+    scores: torch.Tensor, scale: float, q_offset: int = 0
+) -> torch.Tensor:
+    """In one fused pass: probs = softmax(scale * scores + causal_mask).
+
+    NOTE (GPU path): consumes ``scores`` in place — do not reuse it.
+    """
+    return _CausalSoftmax.apply(scores, scale, q_offset)
+
+
+class _CrossEntropy(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, targets, ignore_index):
+        n_valid = int((targets != ignore_index).sum())
+        grad_scale = 1.0 / max(n_valid, 1)
+        if _use_hip(logits):
+            work = logits.contiguous()
+            losses = hip_ops().cross_entropy_fwd_bwd(
+                work, targets.int(), ignore_index, grad_scale, True
+            )
+            ctx.save_for_backward(work)  # now holds dlogits (pre-scaled)
+            ctx.use_hip = True
+        else:
+            lf = logits.float()
+            losses = torch.nn.functional.cross_entropy(
+                lf.view(-1, lf.shape[-1]),
+                targets.view(-1).long(),
+                ignore_index=ignore_index,
+                reduction="none",
+            )
+            dl = torch.softmax(lf, dim=-1)
+            t = targets.view(-1)
+            valid = t != ignore_index
+            onehot = torch.zeros_like(dl.view(-1, dl.shape[-1]))
+            onehot[valid, t[valid].long()] = 1.0
+            dl = (dl.view(-1, dl.shape[-1]) - onehot) * grad_scale
+            dl[~valid] = 0
+            ctx.save_for_backward(dl.to(logits.dtype).view_as(logits))
+            ctx.use_hip = False
+        ctx.n_valid = n_valid
+        return losses.sum() * grad_scale
+
+    @staticmethod
+    def backward(ctx, dloss):
+        (dlogits,) = ctx.saved_tensors
+        return dlogits * dloss, None, None
+
+
+def cross_entropy_loss(
+    logits: torch.Tensor, targets: torch.Tensor, ignore_index: int = -100
+) -> torch.Tensor:
+    """Mean CE over non-ignored tokens. GPU path fuses loss+grad in one
+    kernel and reuses the logits buffer for dlogits (no extra V-sized
+    allocation). The logits tensor is consumed."""
+    return _CrossEntropy.apply(logits, targets, ignore_index)
+
+
+def fused_adamw_step(
+    param_f32: torch.Tensor,
+    grad: torch.Tensor,
+    exp_avg: torch.Tensor,
+    exp_avg_sq: torch.Tensor,
+    param_bf16: Optional[torch.Tensor],
+    lr: float,
+    beta1: float,
+    beta2: float,
+    eps: float,
+    weight_decay: float,
+    step: int,
+    grad_scale: float = 1.0,
+):
+    """Single fused AdamW update; see FusedAdamW for the optimizer class."""
+    if param_f32.is_cuda:
+        hip_ops().adamw_step(
+            param_f32,
+            grad.contiguous(),
+            exp_avg,
+            exp_avg_sq,
+            param_bf16,
+            lr,
+            beta1,
+            beta2,
+            eps,
+            weight_decay,
+            step,
+            grad_scale,
+        )
+        return
+    # CPU reference
+    g = grad.float() * grad_scale
+    exp_avg.mul_(beta1).add_(g, alpha=1 - beta1)
+    exp_avg_sq.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+    bc1 = 1 - beta1**step
+    bc2 = 1 - beta2**step
+    denom = (exp_avg_sq / bc2).sqrt().add_(eps)
+    param_f32.add_(
+        exp_avg / bc1 / denom + weight_decay * param_f32, alpha=-lr
+    )
+    if param_bf16 is not None:
+        param_bf16.copy_(param_f32.to(param_bf16.dtype))

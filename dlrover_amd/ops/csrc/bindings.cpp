@@ -1,0 +1,192 @@
+// Python bindings for the dlrover_amd HIP kernels.
+//
+// Host-only translation layer: validates tensors, allocates outputs through
+// the torch caching allocator, and forwards raw pointers + the current HIP
+// stream to the extern "C" launchers in the .hip translation units.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include <tuple>
+
+extern "C" {
+void rmsnorm_fwd_launch(const void*, const void*, void*, void*, int, int,
+                        float, void*);
+void rmsnorm_bwd_launch(const void*, const void*, const void*, const void*,
+                        void*, void*, void*, int, int, int, void*);
+void rope_launch(void*, const void*, const void*, const void*, long long, int,
+                 int, int, void*);
+void swiglu_fwd_launch(const void*, void*, long long, int, void*);
+void swiglu_bwd_launch(const void*, const void*, void*, long long, int, void*);
+void adamw_launch(void*, const void*, void*, void*, void*, long long, float,
+                  float, float, float, float, int, int, float, void*);
+void causal_softmax_fwd_launch(void*, long long, int, int, int, float, void*);
+void causal_softmax_bwd_launch(void*, const void*, long long, int, float,
+                               void*);
+void cross_entropy_launch(void*, const void*, void*, long long, int, int,
+                          float, int, void*);
+}
+
+namespace {
+
+void* cur_stream() {
+  return (void*)at::cuda::getCurrentCUDAStream().stream();
+}
+
+void check_bf16(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16, name, " must be bf16");
+}
+
+std::tuple<at::Tensor, at::Tensor> rmsnorm_fwd(const at::Tensor& x,
+                                               const at::Tensor& w,
+                                               double eps) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  const int hidden = (int)x.size(-1);
+  TORCH_CHECK(hidden % 8 == 0, "hidden must be a multiple of 8");
+  const long long n_rows = x.numel() / hidden;
+  auto y = at::empty_like(x);
+  auto invrms = at::empty({n_rows}, x.options().dtype(at::kFloat));
+  rmsnorm_fwd_launch(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                     invrms.data_ptr(), (int)n_rows, hidden, (float)eps,
+                     cur_stream());
+  return {y, invrms};
+}
+
+std::tuple<at::Tensor, at::Tensor> rmsnorm_bwd(const at::Tensor& dy,
+                                               const at::Tensor& x,
+                                               const at::Tensor& w,
+                                               const at::Tensor& invrms) {
+  check_bf16(dy, "dy");
+  check_bf16(x, "x");
+  const int hidden = (int)x.size(-1);
+  const long long n_rows = x.numel() / hidden;
+  const int n_partials = 64;
+  auto dx = at::empty_like(x);
+  auto dw = at::empty_like(w);
+  auto partial =
+      at::empty({n_partials, hidden}, x.options().dtype(at::kFloat));
+  rmsnorm_bwd_launch(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                     invrms.data_ptr(), dx.data_ptr(), partial.data_ptr(),
+                     dw.data_ptr(), (int)n_rows, hidden, n_partials,
+                     cur_stream());
+  return {dx, dw};
+}
+
+void rope_apply(at::Tensor& x, const at::Tensor& pos, const at::Tensor& cos,
+                const at::Tensor& sin, bool backward) {
+  check_bf16(x, "x");
+  TORCH_CHECK(pos.scalar_type() == at::kInt && pos.is_cuda(),
+              "pos must be int32 on GPU");
+  TORCH_CHECK(cos.scalar_type() == at::kFloat && cos.is_contiguous(),
+              "cos table must be contiguous fp32");
+  const int head_dim = (int)x.size(-1);
+  const int n_heads = (int)x.size(-2);
+  TORCH_CHECK(head_dim % 8 == 0, "head_dim must be a multiple of 8");
+  const long long n_tokens = x.numel() / ((long long)n_heads * head_dim);
+  TORCH_CHECK(pos.numel() == n_tokens, "pos must have one entry per token");
+  rope_launch(x.data_ptr(), pos.data_ptr(), cos.data_ptr(), sin.data_ptr(),
+              n_tokens, n_heads, head_dim, backward ? 1 : 0, cur_stream());
+}
+
+at::Tensor swiglu_fwd(const at::Tensor& gate_up) {
+  check_bf16(gate_up, "gate_up");
+  const int two_inner = (int)gate_up.size(-1);
+  TORCH_CHECK(two_inner % 16 == 0, "inner dim must be a multiple of 8");
+  const int inner = two_inner / 2;
+  const long long n_rows = gate_up.numel() / two_inner;
+  auto sizes = gate_up.sizes().vec();
+  sizes.back() = inner;
+  auto out = at::empty(sizes, gate_up.options());
+  swiglu_fwd_launch(gate_up.data_ptr(), out.data_ptr(), n_rows, inner,
+                    cur_stream());
+  return out;
+}
+
+at::Tensor swiglu_bwd(const at::Tensor& dy, const at::Tensor& gate_up) {
+  check_bf16(dy, "dy");
+  check_bf16(gate_up, "gate_up");
+  const int inner = (int)gate_up.size(-1) / 2;
+  const long long n_rows = gate_up.numel() / (2LL * inner);
+  auto d_gate_up = at::empty_like(gate_up);
+  swiglu_bwd_launch(dy.data_ptr(), gate_up.data_ptr(), d_gate_up.data_ptr(),
+                    n_rows, inner, cur_stream());
+  return d_gate_up;
+}
+
+void adamw_step(at::Tensor& param, const at::Tensor& grad, at::Tensor& m,
+                at::Tensor& v, c10::optional<at::Tensor> param_bf16,
+                double lr, double beta1, double beta2, double eps,
+                double weight_decay, int64_t step, double grad_scale) {
+  TORCH_CHECK(param.scalar_type() == at::kFloat && param.is_contiguous(),
+              "master param must be contiguous fp32");
+  TORCH_CHECK(grad.is_cuda() && grad.is_contiguous(), "grad must be GPU");
+  TORCH_CHECK(grad.numel() == param.numel(), "grad/param numel mismatch");
+  const bool gbf16 = grad.scalar_type() == at::kBFloat16;
+  TORCH_CHECK(gbf16 || grad.scalar_type() == at::kFloat,
+              "grad must be bf16 or fp32");
+  void* pb = nullptr;
+  if (param_bf16.has_value()) {
+    check_bf16(*param_bf16, "param_bf16");
+    pb = param_bf16->data_ptr();
+  }
+  adamw_launch(param.data_ptr(), grad.data_ptr(), m.data_ptr(), v.data_ptr(),
+               pb, param.numel(), (float)lr, (float)beta1, (float)beta2,
+               (float)eps, (float)weight_decay, (int)step, gbf16 ? 1 : 0,
+               (float)grad_scale, cur_stream());
+}
+
+void causal_softmax_fwd(at::Tensor& scores, int64_t q_len, int64_t q_offset,
+                        double scale) {
+  check_bf16(scores, "scores");
+  const int row_len = (int)scores.size(-1);
+  TORCH_CHECK(row_len % 8 == 0, "row_len must be a multiple of 8");
+  const long long n_rows = scores.numel() / row_len;
+  causal_softmax_fwd_launch(scores.data_ptr(), n_rows, row_len, (int)q_len,
+                            (int)q_offset, (float)scale, cur_stream());
+}
+
+void causal_softmax_bwd(at::Tensor& dscores, const at::Tensor& probs,
+                        double scale) {
+  check_bf16(dscores, "dscores");
+  check_bf16(probs, "probs");
+  const int row_len = (int)dscores.size(-1);
+  const long long n_rows = dscores.numel() / row_len;
+  causal_softmax_bwd_launch(dscores.data_ptr(), probs.data_ptr(), n_rows,
+                            row_len, (float)scale, cur_stream());
+}
+
+at::Tensor cross_entropy_fwd_bwd(at::Tensor& logits, const at::Tensor& targets,
+                                 int64_t ignore_index, double grad_scale,
+                                 bool compute_grad) {
+  check_bf16(logits, "logits");
+  TORCH_CHECK(targets.scalar_type() == at::kInt && targets.is_cuda(),
+              "targets must be int32 on GPU");
+  const int vocab = (int)logits.size(-1);
+  const long long n_rows = logits.numel() / vocab;
+  TORCH_CHECK(targets.numel() == n_rows, "one target per row");
+  auto losses = at::empty({n_rows}, logits.options().dtype(at::kFloat));
+  cross_entropy_launch(logits.data_ptr(), targets.data_ptr(),
+                       losses.data_ptr(), n_rows, vocab, (int)ignore_index,
+                       (float)grad_scale, compute_grad ? 1 : 0, cur_stream());
+  return losses;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "dlrover_amd CDNA4 (gfx950) kernels";
+  m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused RMSNorm forward (bf16)");
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, "fused RMSNorm backward (bf16)");
+  m.def("rope_apply", &rope_apply, "in-place RoPE rotate-half (bf16)");
+  m.def("swiglu_fwd", &swiglu_fwd, "fused SwiGLU forward (bf16)");
+  m.def("swiglu_bwd", &swiglu_bwd, "fused SwiGLU backward (bf16)");
+  m.def("adamw_step", &adamw_step, "fused AdamW with fp32 master weights");
+  m.def("causal_softmax_fwd", &causal_softmax_fwd,
+        "in-place fused scale+causal-mask+softmax (bf16)");
+  m.def("causal_softmax_bwd", &causal_softmax_bwd,
+        "in-place softmax backward (bf16)");
+  m.def("cross_entropy_fwd_bwd", &cross_entropy_fwd_bwd,
+        "fused CE loss + in-place dlogits (bf16)");
+}

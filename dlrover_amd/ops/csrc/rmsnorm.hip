@@ -1,0 +1,130 @@
+// Fused RMSNorm forward/backward for bf16 rows, fp32 accumulation.
+//
+// The reference delegates normalisation to Megatron/HF —
+// this is part of the MI355X-native hot path (BASELINE.json north star).
+// Memory-bound: target HBM ceiling via short8 vector loads (guide G13).
+//
+// Layout: x [N, H] bf16 row-major, w [H] bf16, y [N, H] bf16,
+//         invrms [N] fp32 saved for backward.
+#include "kern_common.h"
+
+extern "C" {
+
+// one block per row (grid-stride over rows), 256 threads, 8 elems/thread/iter
+__global__ void rmsnorm_fwd_kernel(
+    const short* __restrict__ x, const short* __restrict__ w,
+    short* __restrict__ y, float* __restrict__ invrms,
+    int n_rows, int hidden, float eps) {
+  __shared__ float scratch[16];
+  const int vecs = hidden >> 3;  // hidden % 8 == 0 enforced host-side
+  for (int row = blockIdx.x; row < n_rows; row += gridDim.x) {
+    const short* xr = x + (long long)row * hidden;
+    short* yr = y + (long long)row * hidden;
+    float ssq = 0.f;
+    for (int v = threadIdx.x; v < vecs; v += blockDim.x) {
+      float xv[8];
+      load8(xr + v * 8, xv);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ssq += xv[j] * xv[j];
+    }
+    ssq = block_reduce_sum(ssq, scratch);
+    const float inv = rsqrtf(ssq / hidden + eps);
+    if (threadIdx.x == 0 && invrms) invrms[row] = inv;
+    for (int v = threadIdx.x; v < vecs; v += blockDim.x) {
+      float xv[8], wv[8];
+      load8(xr + v * 8, xv);
+      load8(w + v * 8, wv);
+      float out[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) out[j] = xv[j] * inv * wv[j];
+      store8(yr + v * 8, out);
+    }
+    __syncthreads();
+  }
+}
+
+// backward:
+//   dx = inv * w * dy - x * inv^3 / H * sum_j(dy_j * w_j * x_j)
+//   dw = sum_rows(dy * x * inv)   (fp32 partials, reduced by a second kernel)
+__global__ void rmsnorm_bwd_kernel(
+    const short* __restrict__ dy, const short* __restrict__ x,
+    const short* __restrict__ w, const float* __restrict__ invrms,
+    short* __restrict__ dx, float* __restrict__ dw_partial,
+    int n_rows, int hidden, int n_partials) {
+  __shared__ float scratch[16];
+  const int vecs = hidden >> 3;
+  float* dwp = dw_partial + (long long)(blockIdx.x % n_partials) * hidden;
+  for (int row = blockIdx.x; row < n_rows; row += gridDim.x) {
+    const short* dyr = dy + (long long)row * hidden;
+    const short* xr = x + (long long)row * hidden;
+    short* dxr = dx + (long long)row * hidden;
+    const float inv = invrms[row];
+    // pass 1: dot = sum(dy * w * x)
+    float dot = 0.f;
+    for (int v = threadIdx.x; v < vecs; v += blockDim.x) {
+      float dyv[8], wv[8], xv[8];
+      load8(dyr + v * 8, dyv);
+      load8(w + v * 8, wv);
+      load8(xr + v * 8, xv);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dot += dyv[j] * wv[j] * xv[j];
+    }
+    dot = block_reduce_sum(dot, scratch);
+    const float k = dot * inv * inv * inv / hidden;
+    // pass 2: dx + dw partial accumulation
+    for (int v = threadIdx.x; v < vecs; v += blockDim.x) {
+      float dyv[8], wv[8], xv[8], out[8];
+      load8(dyr + v * 8, dyv);
+      load8(w + v * 8, wv);
+      load8(xr + v * 8, xv);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        out[j] = dyv[j] * wv[j] * inv - xv[j] * k;
+        atomicAdd(&dwp[v * 8 + j], dyv[j] * xv[j] * inv);
+      }
+      store8(dxr + v * 8, out);
+    }
+    __syncthreads();
+  }
+}
+
+// reduce [n_partials, H] fp32 -> dw [H] bf16
+__global__ void reduce_partials_kernel(
+    const float* __restrict__ partials, short* __restrict__ out,
+    int n_partials, int hidden) {
+  for (int col = blockIdx.x * blockDim.x + threadIdx.x; col < hidden;
+       col += gridDim.x * blockDim.x) {
+    float acc = 0.f;
+    for (int p = 0; p < n_partials; ++p)
+      acc += partials[(long long)p * hidden + col];
+    out[col] = f2bf(acc);
+  }
+}
+
+void rmsnorm_fwd_launch(const void* x, const void* w, void* y, void* invrms,
+                        int n_rows, int hidden, float eps, hipStream_t stream) {
+  int grid = n_rows < 2048 ? n_rows : 2048;
+  if (grid < 1) grid = 1;
+  hipLaunchKernelGGL(rmsnorm_fwd_kernel, dim3(grid), dim3(256), 0, stream,
+                     (const short*)x, (const short*)w, (short*)y,
+                     (float*)invrms, n_rows, hidden, eps);
+}
+
+void rmsnorm_bwd_launch(const void* dy, const void* x, const void* w,
+                        const void* invrms, void* dx, void* dw_partial,
+                        void* dw, int n_rows, int hidden, int n_partials,
+                        hipStream_t stream) {
+  int grid = n_rows < 2048 ? n_rows : 2048;
+  if (grid < 1) grid = 1;
+  hipMemsetAsync(dw_partial, 0, (size_t)n_partials * hidden * sizeof(float),
+                 stream);
+  hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(grid), dim3(256), 0, stream,
+                     (const short*)dy, (const short*)x, (const short*)w,
+                     (const float*)invrms, (short*)dx, (float*)dw_partial,
+                     n_rows, hidden, n_partials);
+  hipLaunchKernelGGL(reduce_partials_kernel, dim3(grid_capped(hidden, 256)),
+                     dim3(256), 0, stream, (const float*)dw_partial,
+                     (short*)dw, n_partials, hidden);
+}
+
+}  // extern "C"

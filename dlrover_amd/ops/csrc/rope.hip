@@ -1,0 +1,59 @@
+// RoPE (rotate-half / Llama convention) forward+backward, bf16.
+//
+// cos/sin tables are precomputed on HOST (fp32 [max_pos, D/2]) — on-device
+// sinf/cosf turns a memory-bound op VALU-bound (guide Appendix B,
+// "trig-heavy ops"). Backward is the inverse rotation (sin sign flip), so one
+// kernel serves both.
+//
+// Layout: x [n_tokens, n_heads, head_dim] bf16 contiguous;
+//         positions [n_tokens] int32; cos/sin [max_pos, head_dim/2] fp32.
+#include "kern_common.h"
+
+extern "C" {
+
+__global__ void rope_kernel(
+    short* __restrict__ x, const int* __restrict__ pos,
+    const float* __restrict__ cos_tab, const float* __restrict__ sin_tab,
+    long long n_tokens, int n_heads, int head_dim, float sin_sign) {
+  const int half = head_dim >> 1;
+  const int vecs_per_head = half >> 2;  // 4 pairs per thread
+  const long long total = n_tokens * n_heads * vecs_per_head;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    const int v = (int)(i % vecs_per_head);
+    const long long th = i / vecs_per_head;
+    const int h = (int)(th % n_heads);
+    const long long t = th / n_heads;
+    const int d0 = v * 4;
+    short* base = x + (t * n_heads + h) * (long long)head_dim;
+    short4_t v1 = *reinterpret_cast<short4_t*>(base + d0);
+    short4_t v2 = *reinterpret_cast<short4_t*>(base + half + d0);
+    const float* cr = cos_tab + (long long)pos[t] * half + d0;
+    const float* sr = sin_tab + (long long)pos[t] * half + d0;
+    float4_t c = *reinterpret_cast<const float4_t*>(cr);
+    float4_t s = *reinterpret_cast<const float4_t*>(sr);
+    short4_t o1, o2;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float x1 = bf2f(v1[j]);
+      const float x2 = bf2f(v2[j]);
+      const float sj = s[j] * sin_sign;
+      o1[j] = f2bf(x1 * c[j] - x2 * sj);
+      o2[j] = f2bf(x2 * c[j] + x1 * sj);
+    }
+    *reinterpret_cast<short4_t*>(base + d0) = o1;
+    *reinterpret_cast<short4_t*>(base + half + d0) = o2;
+  }
+}
+
+void rope_launch(void* x, const void* pos, const void* cos_tab,
+                 const void* sin_tab, long long n_tokens, int n_heads,
+                 int head_dim, int backward, hipStream_t stream) {
+  const long long total = n_tokens * n_heads * (head_dim >> 3);
+  hipLaunchKernelGGL(rope_kernel, dim3(grid_capped(total, 256)), dim3(256), 0,
+                     stream, (short*)x, (const int*)pos,
+                     (const float*)cos_tab, (const float*)sin_tab, n_tokens,
+                     n_heads, head_dim, backward ? -1.f : 1.f);
+}
+
+}  // extern "C"

@@ -1,0 +1,145 @@
+"""GPU numerics: each HIP kernel vs a plain PyTorch fp32 reference.
+
+bf16 tolerances: one bf16 ulp at |x|~1 is ~0.8%, reductions accumulate in
+fp32 inside the kernels so results should sit well inside 2e-2 relative.
+"""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _require_native():
+    from dlrover_amd.ops.api import hip_ops
+
+    hip_ops()  # raises on a GPU box if the extension didn't build
+
+
+def _dev():
+    return torch.device("cuda:0")
+
+
+def test_rmsnorm_gpu_matches_fp32_ref():
+    from dlrover_amd.ops import rmsnorm
+
+    torch.manual_seed(0)
+    x = torch.randn(64, 4096, device=_dev(), dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(4096, device=_dev(), dtype=torch.bfloat16, requires_grad=True)
+    y = rmsnorm(x, w, 1e-5)
+    (y.float().pow(2)).sum().backward()
+
+    x2 = x.detach().float().requires_grad_(True)
+    w2 = w.detach().float().requires_grad_(True)
+    inv = torch.rsqrt(x2.pow(2).mean(-1, keepdim=True) + 1e-5)
+    y2 = x2 * inv * w2
+    y2.pow(2).sum().backward()
+    torch.testing.assert_close(y.float(), y2, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(x.grad.float(), x2.grad, rtol=5e-2, atol=5e-1)
+    torch.testing.assert_close(w.grad.float(), w2.grad, rtol=5e-2, atol=5e-1)
+
+
+def test_rope_gpu_matches_ref():
+    from dlrover_amd.ops import rope_rotate
+    from dlrover_amd.ops.api import build_rope_cache, rope_ref
+
+    torch.manual_seed(0)
+    cos, sin = build_rope_cache(256, 128, device=_dev())
+    x = torch.randn(2, 64, 8, 128, device=_dev(), dtype=torch.bfloat16)
+    pos = torch.arange(64, device=_dev(), dtype=torch.int32)
+    y = rope_rotate(x, pos, cos, sin)
+    y_ref = rope_ref(x.float(), pos, cos, sin)
+    torch.testing.assert_close(y.float(), y_ref, rtol=2e-2, atol=2e-2)
+
+
+def test_swiglu_gpu_matches_ref():
+    from dlrover_amd.ops import swiglu
+
+    torch.manual_seed(0)
+    gu = torch.randn(1024, 2048, device=_dev(), dtype=torch.bfloat16, requires_grad=True)
+    out = swiglu(gu)
+    out.float().sum().backward()
+    g2 = gu.detach().float().requires_grad_(True)
+    g, u = g2.chunk(2, -1)
+    ref = torch.nn.functional.silu(g) * u
+    ref.sum().backward()
+    torch.testing.assert_close(out.float(), ref, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(gu.grad.float(), g2.grad, rtol=3e-2, atol=3e-2)
+
+
+def test_causal_softmax_gpu_matches_ref():
+    from dlrover_amd.ops import causal_softmax
+    from dlrover_amd.ops.api import causal_softmax_ref
+
+    torch.manual_seed(0)
+    s = torch.randn(4, 8, 512, 512, device=_dev(), dtype=torch.bfloat16)
+    s2 = s.clone().requires_grad_(True)
+    sf = s.float().requires_grad_(True)
+    p = causal_softmax(s2, scale=1 / math.sqrt(128))
+    p_ref = causal_softmax_ref(sf, 1 / math.sqrt(128))
+    torch.testing.assert_close(p.float(), p_ref.float(), rtol=2e-2, atol=2e-3)
+    p.float().pow(2).sum().backward()
+    p_ref.pow(2).sum().backward()
+    torch.testing.assert_close(s2.grad.float(), sf.grad, rtol=5e-2, atol=1e-3)
+
+
+def test_cross_entropy_gpu_matches_ref():
+    from dlrover_amd.ops import cross_entropy_loss
+
+    torch.manual_seed(0)
+    V = 32000
+    logits = torch.randn(128, V, device=_dev(), dtype=torch.bfloat16, requires_grad=True)
+    targets = torch.randint(0, V, (128,), device=_dev(), dtype=torch.int32)
+    targets[5] = -100
+    loss = cross_entropy_loss(logits, targets)
+    loss.backward()
+    l2 = logits.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(l2, targets.long(), ignore_index=-100)
+    ref.backward()
+    torch.testing.assert_close(loss.float(), ref, rtol=1e-2, atol=1e-3)
+    torch.testing.assert_close(logits.grad.float(), l2.grad, rtol=5e-2, atol=1e-4)
+
+
+def test_fused_adamw_gpu_matches_torch():
+    from dlrover_amd.ops.api import fused_adamw_step
+
+    torch.manual_seed(0)
+    n = 1 << 20 | 3  # odd tail exercises the scalar path
+    p = torch.randn(n, device=_dev())
+    pb = p.bfloat16()
+    m = torch.zeros(n, device=_dev())
+    v = torch.zeros(n, device=_dev())
+    ref = p.clone().requires_grad_(True)
+    opt = torch.optim.AdamW([ref], lr=1e-2, betas=(0.9, 0.95), eps=1e-8, weight_decay=0.1)
+    for step in range(1, 5):
+        g = torch.randn(n, device=_dev())
+        ref.grad = g.clone()
+        opt.step()
+        fused_adamw_step(p, g, m, v, pb, 1e-2, 0.9, 0.95, 1e-8, 0.1, step)
+    torch.testing.assert_close(p, ref.detach(), rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(pb.float(), ref.detach().bfloat16().float())
+
+
+def test_llama_tiny_gpu_step():
+    from dlrover_amd.models import LlamaConfig, LlamaForCausalLM
+    from dlrover_amd.ops import FusedAdamW
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig.tiny()
+    model = LlamaForCausalLM(cfg).to(_dev()).bfloat16()
+    model.rope_cos = model.rope_cos.float()
+    model.rope_sin = model.rope_sin.float()
+    opt = FusedAdamW(model.parameters(), lr=3e-3, weight_decay=0.0)
+    ids = torch.randint(0, cfg.vocab_size, (2, 32), device=_dev())
+    first = last = None
+    for _ in range(10):
+        loss = model(ids, ids.clone())
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        first = first or loss.item()
+        last = loss.item()
+    assert last < first * 0.9, (first, last)
