@@ -13,8 +13,8 @@ void rmsnorm_fwd_launch(const void*, const void*, void*, void*, int, int,
                         float, void*);
 void rmsnorm_bwd_launch(const void*, const void*, const void*, const void*,
                         void*, void*, void*, int, int, int, void*);
-void rope_launch(void*, const void*, const void*, const void*, long long, int,
-                 int, int, void*);
+void rope_launch(void*, const void*, const void*, const void*, long long,
+                 long long, int, int, int, void*);
 void swiglu_fwd_launch(const void*, void*, long long, int, void*);
 void swiglu_bwd_launch(const void*, const void*, void*, long long, int, void*);
 void adamw_launch(void*, const void*, void*, void*, void*, long long, float,
@@ -85,9 +85,12 @@ void rope_apply(at::Tensor& x, const at::Tensor& pos, const at::Tensor& cos,
   const int n_heads = (int)x.size(-2);
   TORCH_CHECK(head_dim % 8 == 0, "head_dim must be a multiple of 8");
   const long long n_tokens = x.numel() / ((long long)n_heads * head_dim);
-  TORCH_CHECK(pos.numel() == n_tokens, "pos must have one entry per token");
+  TORCH_CHECK(pos.numel() > 0 && n_tokens % pos.numel() == 0,
+              "pos length must divide the token count (one entry per "
+              "sequence position)");
   rope_launch(x.data_ptr(), pos.data_ptr(), cos.data_ptr(), sin.data_ptr(),
-              n_tokens, n_heads, head_dim, backward ? 1 : 0, cur_stream());
+              n_tokens, (long long)pos.numel(), n_heads, head_dim,
+              backward ? 1 : 0, cur_stream());
 }
 
 at::Tensor swiglu_fwd(const at::Tensor& gate_up) {

@@ -14,7 +14,8 @@ extern "C" {
 __global__ void rope_kernel(
     short* __restrict__ x, const int* __restrict__ pos,
     const float* __restrict__ cos_tab, const float* __restrict__ sin_tab,
-    long long n_tokens, int n_heads, int head_dim, float sin_sign) {
+    long long n_tokens, long long n_pos, int n_heads, int head_dim,
+    float sin_sign) {
   const int half = head_dim >> 1;
   const int vecs_per_head = half >> 2;  // 4 pairs per thread
   const long long total = n_tokens * n_heads * vecs_per_head;
@@ -28,8 +29,11 @@ __global__ void rope_kernel(
     short* base = x + (t * n_heads + h) * (long long)head_dim;
     short4_t v1 = *reinterpret_cast<short4_t*>(base + d0);
     short4_t v2 = *reinterpret_cast<short4_t*>(base + half + d0);
-    const float* cr = cos_tab + (long long)pos[t] * half + d0;
-    const float* sr = sin_tab + (long long)pos[t] * half + d0;
+    // pos holds one entry per sequence position; token t of a [B, T, ...]
+    // batch sits at sequence position t % n_pos
+    const int p = pos[t % n_pos];
+    const float* cr = cos_tab + (long long)p * half + d0;
+    const float* sr = sin_tab + (long long)p * half + d0;
     float4_t c = *reinterpret_cast<const float4_t*>(cr);
     float4_t s = *reinterpret_cast<const float4_t*>(sr);
     short4_t o1, o2;
@@ -47,13 +51,13 @@ __global__ void rope_kernel(
 }
 
 void rope_launch(void* x, const void* pos, const void* cos_tab,
-                 const void* sin_tab, long long n_tokens, int n_heads,
-                 int head_dim, int backward, hipStream_t stream) {
+                 const void* sin_tab, long long n_tokens, long long n_pos,
+                 int n_heads, int head_dim, int backward, hipStream_t stream) {
   const long long total = n_tokens * n_heads * (head_dim >> 3);
   hipLaunchKernelGGL(rope_kernel, dim3(grid_capped(total, 256)), dim3(256), 0,
                      stream, (short*)x, (const int*)pos,
                      (const float*)cos_tab, (const float*)sin_tab, n_tokens,
-                     n_heads, head_dim, backward ? -1.f : 1.f);
+                     n_pos, n_heads, head_dim, backward ? -1.f : 1.f);
 }
 
 }  // extern "C"
