@@ -1,0 +1,64 @@
+"""DDP checkpointer (ref: flash_checkpoint/ddp.py:25 DdpCheckpointer).
+
+Model/optimizer state is replicated under DDP, so rank 0 snapshots the full
+state into its shm segment; restore broadcasts nothing — every rank reads the
+same full checkpoint (shm on node 0's ranks, storage elsewhere).
+"""
+
+from typing import Optional
+
+import torch.distributed as dist
+
+from dlrover_amd.common.log import logger
+from dlrover_amd.trainer.flash_checkpoint.checkpointer import (
+    Checkpointer,
+    StorageType,
+)
+from dlrover_amd.trainer.flash_checkpoint.engine import FullCheckpointEngine
+
+
+class DdpCheckpointer(Checkpointer):
+    def __init__(self, checkpoint_dir: str, model=None, optimizer=None, storage=None):
+        self.checkpoint_dir = checkpoint_dir
+        self.model = model
+        self.optimizer = optimizer
+        self.engine = FullCheckpointEngine(checkpoint_dir, storage=storage)
+
+    def _state_dict(self, step: int):
+        mod = self.model.module if hasattr(self.model, "module") else self.model
+        sd = {"step": step, "model": mod.state_dict()}
+        if self.optimizer is not None:
+            sd["optimizer"] = self.optimizer.state_dict()
+        return sd
+
+    def save_checkpoint(
+        self,
+        step: int,
+        state_dict: Optional[dict] = None,
+        path: str = "",
+        storage_type: int = StorageType.DISK,
+    ) -> float:
+        sd = state_dict if state_dict is not None else self._state_dict(step)
+        if storage_type == StorageType.MEMORY:
+            return self.engine.save_to_memory(step, sd, path=path)
+        return self.engine.save_to_storage(step, sd, path=path)
+
+    def load_checkpoint(self, resume_path: str = "") -> Optional[dict]:
+        sd = self.engine.load(resume_path)
+        if sd is None:
+            return None
+        if self.model is not None:
+            mod = self.model.module if hasattr(self.model, "module") else self.model
+            mod.load_state_dict(sd["model"])
+        if self.optimizer is not None and "optimizer" in sd:
+            self.optimizer.load_state_dict(sd["optimizer"])
+        if dist.is_available() and dist.is_initialized():
+            dist.barrier()
+        logger.info("restored checkpoint step=%s", sd.get("step"))
+        return sd
+
+    def wait_latest_checkpoint(self, timeout: int = 600):
+        self.engine.wait_saving()
+
+    def close(self):
+        self.engine.close()

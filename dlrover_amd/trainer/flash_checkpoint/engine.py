@@ -1,0 +1,429 @@
+"""Flash-checkpoint engines: coordinate shm snapshots in the TRAINING process
+and hand persistence to the agent's AsyncCheckpointSaver.
+
+Parity target: ref dlrover/trainer/torch/flash_checkpoint/engine.py:60-560
+(CheckpointEngine, readiness all-reduce, save_state_dict_to_memory,
+save_to_storage event queue) and full_ckpt_engine.py / fsdp_engine.py.
+
+Operating modes:
+  - agent mode (default under dlrover-run): SharedLock/SharedQueue served by
+    the elastic agent's IPCServer coordinate with AsyncCheckpointSaver;
+  - standalone mode (no agent socket, e.g. bench.py / notebooks): an
+    in-process saver thread provides the same behavior.
+"""
+
+import os
+import threading
+import time
+from dataclasses import dataclass
+from typing import Any, Optional
+
+import torch
+import torch.distributed as dist
+
+from dlrover_amd.common.constants import CheckpointConstant
+from dlrover_amd.common.log import logger
+from dlrover_amd.common.multi_process import (
+    SharedDict,
+    SharedLock,
+    SharedQueue,
+    ipc_socket_path,
+)
+from dlrover_amd.common.storage import (
+    PosixDiskStorage,
+    read_tracker_step,
+)
+from dlrover_amd.trainer.flash_checkpoint.shm_handler import (
+    SharedMemoryHandler,
+    shm_segment_name,
+)
+
+CKPT_EVENT_QUEUE = "flash_ckpt_events"
+CKPT_META_DICT = "flash_ckpt_meta"
+CKPT_LOCK_PREFIX = "flash_ckpt_shm_"
+
+
+@dataclass
+class CheckpointEvent:
+    SAVE = "SAVE"
+    EXIT = "EXIT"
+    type: str = SAVE
+    step: int = 0
+    path: str = ""
+    local_rank: int = 0
+    global_rank: int = 0
+    world_size: int = 1
+
+
+def _local_rank() -> int:
+    return int(os.getenv("LOCAL_RANK", "0"))
+
+
+def _global_rank() -> int:
+    if dist.is_available() and dist.is_initialized():
+        return dist.get_rank()
+    return int(os.getenv("RANK", "0"))
+
+
+def _world_size() -> int:
+    if dist.is_available() and dist.is_initialized():
+        return dist.get_world_size()
+    return int(os.getenv("WORLD_SIZE", "1"))
+
+
+class CheckpointEngine:
+    """Base engine: one shm segment per local rank.
+
+    save_to_memory(step, state_dict)  -> blocking seconds (training stall)
+    save_to_storage(step, state_dict, path) -> same + async persist event
+    load(path=None) -> state_dict from shm (hit) or storage
+    """
+
+    def __init__(
+        self,
+        checkpoint_dir: str,
+        storage=None,
+        comm_backend: str = "",
+        save_timeout: int = CheckpointConstant.SAVE_TIMEOUT,
+    ):
+        self.checkpoint_dir = checkpoint_dir
+        self.storage = storage or PosixDiskStorage()
+        self._save_timeout = save_timeout
+        self._local_rank = _local_rank()
+        self._job = os.getenv("ELASTIC_JOB_NAME", "default")
+        self.shm_handler = SharedMemoryHandler(
+            shm_segment_name(self._job, self._local_rank)
+        )
+        self._agent_mode = os.path.exists(ipc_socket_path())
+        if self._agent_mode:
+            self._shm_lock = SharedLock(f"{CKPT_LOCK_PREFIX}{self._local_rank}")
+            self._event_queue = SharedQueue(CKPT_EVENT_QUEUE)
+            self._meta_dict = SharedDict(CKPT_META_DICT)
+            self._standalone_saver = None
+        else:
+            self._shm_lock = threading.Lock()
+            self._event_queue = None
+            self._meta_dict = None
+            self._standalone_saver = _StandaloneSaver(self)
+        # gloo side group for readiness checks (ref: engine.py:60-75); only
+        # when a process group exists
+        self._sync_group = None
+        if dist.is_available() and dist.is_initialized():
+            backend = comm_backend or "gloo"
+            try:
+                self._sync_group = dist.new_group(backend=backend)
+            except (RuntimeError, ValueError):
+                self._sync_group = None
+        self._last_save_step = -1
+
+    # -- readiness -------------------------------------------------------------
+
+    def _check_all_ranks_ready(self, ready: bool) -> bool:
+        """All-reduce a readiness flag so a rank that crashed mid-iteration
+        can't leave peers writing inconsistent steps (ref: engine.py:60)."""
+        if self._sync_group is None:
+            return ready
+        t = torch.tensor([0.0 if ready else 1.0])
+        dist.all_reduce(t, group=self._sync_group)
+        return t.item() == 0.0
+
+    def _check_step_consistent(self, step: int) -> bool:
+        if self._sync_group is None:
+            return True
+        ws = dist.get_world_size(self._sync_group)
+        out = [torch.zeros(1, dtype=torch.long) for _ in range(ws)]
+        dist.all_gather(out, torch.tensor([step], dtype=torch.long), group=self._sync_group)
+        return all(int(o.item()) == step for o in out)
+
+    # -- state-dict hooks (framework engines override) ---------------------------
+
+    def gather_state_dict(self, model, optimizer) -> Any:
+        raise NotImplementedError
+
+    def load_into(self, model, optimizer, state_dict):
+        raise NotImplementedError
+
+    def rank_saves(self) -> bool:
+        """Does THIS rank write a shard? (DDP full: rank0 only; FSDP: all)."""
+        return True
+
+    # -- save ------------------------------------------------------------------
+
+    def save_to_memory(
+        self, step: int, state_dict: Any, path: str = "", block: bool = True
+    ) -> float:
+        """Snapshot into shm. Returns blocking seconds."""
+        if not self._check_step_consistent(step):
+            raise RuntimeError(f"checkpoint step {step} differs across ranks")
+        if not self._check_all_ranks_ready(True):
+            logger.warning("skip checkpoint@%s: a peer rank is not ready", step)
+            return 0.0
+        blocking = 0.0
+        if self.rank_saves():
+            acquired = self._shm_lock.acquire(timeout=self._save_timeout) if isinstance(
+                self._shm_lock, SharedLock
+            ) else self._shm_lock.acquire(timeout=self._save_timeout)
+            if not acquired:
+                logger.warning("shm lock not acquired; skip checkpoint@%s", step)
+                return 0.0
+            try:
+                extra = {
+                    "path": path,
+                    "global_rank": _global_rank(),
+                    "world_size": _world_size(),
+                }
+                blocking = self.shm_handler.save_state_dict(
+                    step, state_dict, extra=extra, block=block
+                )
+            finally:
+                self._shm_lock.release()
+        self._last_save_step = step
+        return blocking
+
+    def save_to_storage(self, step: int, state_dict: Any, path: str = "") -> float:
+        path = path or os.path.join(self.checkpoint_dir, str(step))
+        blocking = self.save_to_memory(step, state_dict, path=path)
+        event = CheckpointEvent(
+            type=CheckpointEvent.SAVE,
+            step=step,
+            path=path,
+            local_rank=self._local_rank,
+            global_rank=_global_rank(),
+            world_size=_world_size(),
+        )
+        if self._event_queue is not None:
+            if self.rank_saves():
+                self._event_queue.put(event)
+        elif self._standalone_saver is not None and self.rank_saves():
+            self._standalone_saver.submit(event)
+        return blocking
+
+    # -- load ------------------------------------------------------------------
+
+    def load(self, path: str = "", device: Optional[torch.device] = None) -> Optional[Any]:
+        """shm hit first (ref: engine.py load :534), else storage."""
+        sd = self.shm_handler.load_state_dict(device=device)
+        if sd is not None:
+            return sd
+        return self.load_from_storage(path, device=device)
+
+    def load_from_storage(
+        self, path: str = "", device: Optional[torch.device] = None
+    ) -> Optional[Any]:
+        if not path:
+            step = read_tracker_step(self.checkpoint_dir)
+            if step < 0:
+                return None
+            path = os.path.join(self.checkpoint_dir, str(step))
+        shard = os.path.join(path, self._shard_file_name(_global_rank()))
+        if not os.path.exists(shard):
+            return None
+        return torch.load(shard, map_location=device or "cpu", weights_only=False)
+
+    @staticmethod
+    def _shard_file_name(global_rank: int) -> str:
+        return f"rank_{global_rank:05d}.pt"
+
+    def latest_step(self) -> int:
+        """Max of shm step and storage tracker step."""
+        return max(self.shm_handler.committed_step(), read_tracker_step(self.checkpoint_dir))
+
+    def wait_saving(self):
+        """Block until async persist (if any) finished."""
+        self.shm_handler.wait_drained()
+        if self._standalone_saver is not None:
+            self._standalone_saver.wait_idle()
+
+    def close(self):
+        if self._standalone_saver is not None:
+            self._standalone_saver.stop()
+        self.shm_handler.close()
+
+
+class _StandaloneSaver:
+    """In-process persistence thread for agent-less runs. Mirrors the agent's
+    AsyncCheckpointSaver commit protocol (done files + tracker)."""
+
+    def __init__(self, engine: CheckpointEngine):
+        self._engine = engine
+        self._queue: "list[CheckpointEvent]" = []
+        self._cv = threading.Condition()
+        self._stop = False
+        self._busy = False
+        self._thread = threading.Thread(target=self._run, daemon=True)
+        self._thread.start()
+
+    def submit(self, event: CheckpointEvent):
+        with self._cv:
+            self._queue.append(event)
+            self._cv.notify()
+
+    def wait_idle(self):
+        while True:
+            with self._cv:
+                if not self._queue and not self._busy:
+                    return
+            time.sleep(0.02)
+
+    def stop(self):
+        with self._cv:
+            self._stop = True
+            self._cv.notify()
+        self._thread.join(timeout=10)
+
+    def _run(self):
+        from dlrover_amd.agent.ckpt_saver import persist_shm_to_storage
+
+        while True:
+            with self._cv:
+                while not self._queue and not self._stop:
+                    self._cv.wait(timeout=1.0)
+                if self._stop and not self._queue:
+                    return
+                event = self._queue.pop(0)
+                self._busy = True
+            try:
+                self._engine.shm_handler.wait_drained()
+                persist_shm_to_storage(
+                    self._engine.shm_handler,
+                    event,
+                    self._engine.storage,
+                    self._engine.checkpoint_dir,
+                    expected_shards=1,
+                )
+            except Exception:  # noqa: BLE001
+                logger.exception("standalone checkpoint persist failed")
+            finally:
+                with self._cv:
+                    self._busy = False
+
+
+class FullCheckpointEngine(CheckpointEngine):
+    """DDP-style: the model/optimizer state is replicated, so only global
+    rank 0 snapshots and persists (ref: full_ckpt_engine.py:33)."""
+
+    def rank_saves(self) -> bool:
+        return _global_rank() == 0
+
+    def gather_state_dict(self, model, optimizer):
+        mod = model.module if hasattr(model, "module") else model
+        sd = {"model": mod.state_dict(), "step": self._last_save_step}
+        if optimizer is not None:
+            sd["optimizer"] = optimizer.state_dict()
+        return sd
+
+    def load_into(self, model, optimizer, state_dict):
+        mod = model.module if hasattr(model, "module") else model
+        mod.load_state_dict(state_dict["model"])
+        if optimizer is not None and "optimizer" in state_dict:
+            optimizer.load_state_dict(state_dict["optimizer"])
+
+
+class ShardedCheckpointEngine(CheckpointEngine):
+    """FSDP-style: every rank snapshots its own local shard
+    (ref: fsdp_engine.py:447 — ours stores local shards directly instead of
+    DCP StorageWriter plumbing; resharding load is handled by the UCP hook)."""
+
+    def rank_saves(self) -> bool:
+        return True
+
+    def gather_state_dict(self, model, optimizer):
+        from torch.distributed.checkpoint.state_dict import (
+            StateDictOptions,
+            get_state_dict,
+        )
+
+        opts = StateDictOptions(full_state_dict=False, cpu_offload=False)
+        if optimizer is not None:
+            msd, osd = get_state_dict(model, optimizer, options=opts)
+        else:
+            from torch.distributed.checkpoint.state_dict import get_model_state_dict
+
+            msd = get_model_state_dict(model, options=opts)
+            osd = {}
+        return {"model": _localize(msd), "optimizer": _localize(osd)}
+
+    def load_into(self, model, optimizer, state_dict):
+        from torch.distributed.checkpoint.state_dict import (
+            StateDictOptions,
+            get_state_dict,
+            set_state_dict,
+        )
+
+        opts = StateDictOptions(full_state_dict=False, cpu_offload=False)
+        # rebuild DTensor structure from current model, then fill local shards
+        if optimizer is not None:
+            msd, osd = get_state_dict(model, optimizer, options=opts)
+        else:
+            from torch.distributed.checkpoint.state_dict import get_model_state_dict
+
+            msd = get_model_state_dict(model, options=opts)
+            osd = {}
+        _fill_local(msd, state_dict["model"])
+        _fill_local(osd, state_dict.get("optimizer", {}))
+        if optimizer is not None:
+            set_state_dict(
+                model, optimizer, model_state_dict=msd, optim_state_dict=osd,
+                options=opts,
+            )
+        else:
+            from torch.distributed.checkpoint.state_dict import set_model_state_dict
+
+            set_model_state_dict(model, msd, options=opts)
+
+
+def _localize(obj):
+    """Replace DTensors with their local shards for shm serialization."""
+    from dlrover_amd.trainer.flash_checkpoint.shm_handler import traverse_state_dict
+
+    try:
+        from torch.distributed.tensor import DTensor
+    except ImportError:  # pragma: no cover
+        DTensor = ()
+
+    def conv(v):
+        if DTensor and isinstance(v, DTensor):
+            return v.to_local()
+        return v
+
+    return _map_leaves(obj, conv)
+
+
+def _map_leaves(obj, fn):
+    if isinstance(obj, dict):
+        return {k: _map_leaves(v, fn) for k, v in obj.items()}
+    if isinstance(obj, list):
+        return [_map_leaves(v, fn) for v in obj]
+    if isinstance(obj, tuple):
+        return tuple(_map_leaves(v, fn) for v in obj)
+    return fn(obj)
+
+
+def _fill_local(dst, src):
+    """Copy loaded local-shard values into the live (possibly DTensor) state
+    dict structure, in place."""
+    try:
+        from torch.distributed.tensor import DTensor
+    except ImportError:  # pragma: no cover
+        DTensor = ()
+    if isinstance(dst, dict):
+        for k in dst:
+            if k in src:
+                v = dst[k]
+                if isinstance(v, (dict, list)):
+                    _fill_local(v, src[k])
+                elif DTensor and isinstance(v, DTensor):
+                    v.to_local().copy_(src[k].to(v.device))
+                elif isinstance(v, torch.Tensor):
+                    v.copy_(src[k].to(v.device))
+                else:
+                    dst[k] = src[k]
+    elif isinstance(dst, list):
+        for i, v in enumerate(dst):
+            if i < len(src):
+                if isinstance(v, (dict, list)):
+                    _fill_local(v, src[i])
+                elif isinstance(v, torch.Tensor):
+                    v.copy_(src[i].to(v.device))
+                else:
+                    dst[i] = src[i]

@@ -66,6 +66,11 @@ class SegmentMeta:
     extra: Dict[str, Any] = field(default_factory=dict)
 
 
+class ListIdx(int):
+    """Path element marking a LIST index — optimizer state dicts use plain
+    int keys in dicts, so bare ints cannot distinguish the two."""
+
+
 def traverse_state_dict(value: Any, path: Tuple = ()):
     """Yield (path, leaf) pairs; dicts and lists/tuples are traversed."""
     if isinstance(value, dict):
@@ -73,16 +78,25 @@ def traverse_state_dict(value: Any, path: Tuple = ()):
             yield from traverse_state_dict(v, path + (k,))
     elif isinstance(value, (list, tuple)):
         for i, v in enumerate(value):
-            yield from traverse_state_dict(v, path + (i,))
+            yield from traverse_state_dict(v, path + (ListIdx(i),))
     else:
         yield path, value
 
 
-def _set_by_path(root: Any, path: Tuple, value: Any):
-    node = root
-    for key in path[:-1]:
-        node = node[key]
-    node[path[-1]] = value
+def _child_container(next_key) -> Any:
+    return [] if isinstance(next_key, ListIdx) else {}
+
+
+def _descend(node: Any, key, next_key):
+    if isinstance(node, list):
+        while len(node) <= key:
+            node.append(None)
+        if node[key] is None:
+            node[key] = _child_container(next_key)
+        return node[key]
+    if key not in node:
+        node[key] = _child_container(next_key)
+    return node[key]
 
 
 def _build_skeleton(paths_values):
@@ -91,19 +105,7 @@ def _build_skeleton(paths_values):
     for path, value in paths_values:
         node = root
         for i, key in enumerate(path[:-1]):
-            nxt_key = path[i + 1]
-            if key not in node if isinstance(node, dict) else key >= len(node):
-                container = [] if isinstance(nxt_key, int) else {}
-                if isinstance(node, dict):
-                    node = node.setdefault(key, container)
-                else:
-                    while len(node) <= key:
-                        node.append(None)
-                    if node[key] is None:
-                        node[key] = container
-                    node = node[key]
-            else:
-                node = node[key]
+            node = _descend(node, key, path[i + 1])
         last = path[-1]
         if isinstance(node, list):
             while len(node) <= last:

@@ -92,11 +92,14 @@ def test_cross_entropy_gpu_matches_ref():
     torch.manual_seed(0)
     V = 32000
     logits = torch.randn(128, V, device=_dev(), dtype=torch.bfloat16, requires_grad=True)
+    # the fused op consumes the logits buffer (overwrites it with dlogits) —
+    # snapshot the original values for the reference first
+    logits_orig = logits.detach().clone()
     targets = torch.randint(0, V, (128,), device=_dev(), dtype=torch.int32)
     targets[5] = -100
     loss = cross_entropy_loss(logits, targets)
     loss.backward()
-    l2 = logits.detach().float().requires_grad_(True)
+    l2 = logits_orig.float().requires_grad_(True)
     ref = torch.nn.functional.cross_entropy(l2, targets.long(), ignore_index=-100)
     ref.backward()
     torch.testing.assert_close(loss.float(), ref, rtol=1e-2, atol=1e-3)
@@ -120,7 +123,9 @@ def test_fused_adamw_gpu_matches_torch():
         opt.step()
         fused_adamw_step(p, g, m, v, pb, 1e-2, 0.9, 0.95, 1e-8, 0.1, step)
     torch.testing.assert_close(p, ref.detach(), rtol=1e-4, atol=1e-5)
-    torch.testing.assert_close(pb.float(), ref.detach().bfloat16().float())
+    # bf16 copy must equal the kernel's own master rounded to bf16 (comparing
+    # against torch's master flips 1-ulp rounding at ties)
+    torch.testing.assert_close(pb.float(), p.bfloat16().float())
 
 
 def test_llama_tiny_gpu_step():
