@@ -23,6 +23,19 @@ import torch
 
 from dlrover_amd.ops.api import fused_adamw_step
 
+try:
+    from torch.distributed.tensor import DTensor
+except ImportError:  # pragma: no cover
+    DTensor = ()
+
+
+def _local(t: torch.Tensor) -> torch.Tensor:
+    """FSDP2 params/grads are DTensors; the kernel operates on the local
+    shard (each rank owns its shard's optimizer state — ZeRO-3 style)."""
+    if DTensor and isinstance(t, DTensor):
+        return t.to_local()
+    return t
+
 
 class FusedAdamW(torch.optim.Optimizer):
     def __init__(
@@ -44,7 +57,7 @@ class FusedAdamW(torch.optim.Optimizer):
     def _init_state(self, p: torch.Tensor):
         state = self.state[p]
         state["step"] = 0
-        master = p.detach().float().clone()
+        master = _local(p.detach()).float().clone()
         state["master_param"] = master
         state["exp_avg"] = torch.zeros_like(master)
         state["exp_avg_sq"] = torch.zeros_like(master)
@@ -60,12 +73,13 @@ class FusedAdamW(torch.optim.Optimizer):
                     self._init_state(p)
                 state["step"] += 1
                 is_bf16 = p.dtype == torch.bfloat16
+                p_data = _local(p.data)
                 fused_adamw_step(
                     state["master_param"],
-                    p.grad,
+                    _local(p.grad).contiguous(),
                     state["exp_avg"],
                     state["exp_avg_sq"],
-                    p.data if is_bf16 else None,
+                    p_data if is_bf16 else None,
                     group["lr"],
                     beta1,
                     beta2,
@@ -75,7 +89,7 @@ class FusedAdamW(torch.optim.Optimizer):
                 )
                 if not is_bf16:
                     # fp32 params: master IS the param storage
-                    p.data.copy_(state["master_param"])
+                    p_data.copy_(state["master_param"])
 
     def _grad_ptrs(self):
         return tuple(
@@ -126,6 +140,32 @@ class FusedAdamW(torch.optim.Optimizer):
                     self.state[p]["step"] += 1
         self._graph.replay()
         return loss
+
+    def load_state_dict(self, state_dict):
+        """Override: torch's default casts every floating state tensor to the
+        PARAM dtype — which would narrow our fp32 master weights to bf16.
+        Restore state by position with device moves only."""
+        groups = self.param_groups
+        saved_groups = state_dict["param_groups"]
+        params = [p for g in groups for p in g["params"]]
+        saved_ids = [pid for g in saved_groups for pid in g["params"]]
+        if len(params) != len(saved_ids):
+            raise ValueError(
+                f"optimizer param count mismatch: {len(params)} vs {len(saved_ids)}"
+            )
+        new_state = {}
+        for sid, p in zip(saved_ids, params):
+            if sid in state_dict["state"]:
+                dev = _local(p).device
+                new_state[p] = {
+                    k: v.to(dev) if torch.is_tensor(v) else v
+                    for k, v in state_dict["state"][sid].items()
+                }
+        self.state.clear()
+        self.state.update(new_state)
+        for g, sg in zip(groups, saved_groups):
+            g.update({k: v for k, v in sg.items() if k != "params"})
+        self._graph = None  # state tensors moved: any captured graph is stale
 
     def zero_grad(self, set_to_none: bool = False):
         # graph capture needs stable grad storage: zero in place by default

@@ -320,73 +320,53 @@ class FullCheckpointEngine(CheckpointEngine):
 
 
 class ShardedCheckpointEngine(CheckpointEngine):
-    """FSDP-style: every rank snapshots its own local shard
-    (ref: fsdp_engine.py:447 — ours stores local shards directly instead of
-    DCP StorageWriter plumbing; resharding load is handled by the UCP hook)."""
+    """FSDP-style: every rank snapshots its own local shards
+    (ref: fsdp_engine.py:447 — ours iterates params/buffers directly and
+    stores DTensor LOCAL shards, instead of DCP StorageWriter plumbing:
+    the shm writer then D2H-copies live param storage with zero staging
+    surprises, and restore is an in-place copy into the same storage.
+    Same-world-size restore only; cross-world resharding is the UCP hook)."""
 
     def rank_saves(self) -> bool:
         return True
 
+    @staticmethod
+    def _named_tensors(model):
+        import itertools
+
+        return itertools.chain(model.named_parameters(), model.named_buffers())
+
     def gather_state_dict(self, model, optimizer):
-        from torch.distributed.checkpoint.state_dict import (
-            StateDictOptions,
-            get_state_dict,
-        )
-
-        opts = StateDictOptions(full_state_dict=False, cpu_offload=False)
-        if optimizer is not None:
-            msd, osd = get_state_dict(model, optimizer, options=opts)
-        else:
-            from torch.distributed.checkpoint.state_dict import get_model_state_dict
-
-            msd = get_model_state_dict(model, options=opts)
-            osd = {}
-        return {"model": _localize(msd), "optimizer": _localize(osd)}
+        model_sd = {}
+        for name, t in self._named_tensors(model):
+            model_sd[name] = _to_local(t.detach())
+        opt_sd = _localize(optimizer.state_dict()) if optimizer is not None else {}
+        return {"model": model_sd, "optimizer": opt_sd}
 
     def load_into(self, model, optimizer, state_dict):
-        from torch.distributed.checkpoint.state_dict import (
-            StateDictOptions,
-            get_state_dict,
-            set_state_dict,
-        )
+        live = dict(self._named_tensors(model))
+        with torch.no_grad():
+            for name, saved in state_dict["model"].items():
+                if name not in live:
+                    logger.warning("checkpoint key %s not in model — skipped", name)
+                    continue
+                dst = _to_local(live[name].data)
+                dst.copy_(saved.to(dst.device))
+        if optimizer is not None and state_dict.get("optimizer"):
+            optimizer.load_state_dict(state_dict["optimizer"])
 
-        opts = StateDictOptions(full_state_dict=False, cpu_offload=False)
-        # rebuild DTensor structure from current model, then fill local shards
-        if optimizer is not None:
-            msd, osd = get_state_dict(model, optimizer, options=opts)
-        else:
-            from torch.distributed.checkpoint.state_dict import get_model_state_dict
 
-            msd = get_model_state_dict(model, options=opts)
-            osd = {}
-        _fill_local(msd, state_dict["model"])
-        _fill_local(osd, state_dict.get("optimizer", {}))
-        if optimizer is not None:
-            set_state_dict(
-                model, optimizer, model_state_dict=msd, optim_state_dict=osd,
-                options=opts,
-            )
-        else:
-            from torch.distributed.checkpoint.state_dict import set_model_state_dict
-
-            set_model_state_dict(model, msd, options=opts)
+def _to_local(t):
+    try:
+        from torch.distributed.tensor import DTensor
+    except ImportError:  # pragma: no cover
+        return t
+    return t.to_local() if isinstance(t, DTensor) else t
 
 
 def _localize(obj):
     """Replace DTensors with their local shards for shm serialization."""
-    from dlrover_amd.trainer.flash_checkpoint.shm_handler import traverse_state_dict
-
-    try:
-        from torch.distributed.tensor import DTensor
-    except ImportError:  # pragma: no cover
-        DTensor = ()
-
-    def conv(v):
-        if DTensor and isinstance(v, DTensor):
-            return v.to_local()
-        return v
-
-    return _map_leaves(obj, conv)
+    return _map_leaves(obj, _to_local)
 
 
 def _map_leaves(obj, fn):
@@ -397,33 +377,3 @@ def _map_leaves(obj, fn):
     if isinstance(obj, tuple):
         return tuple(_map_leaves(v, fn) for v in obj)
     return fn(obj)
-
-
-def _fill_local(dst, src):
-    """Copy loaded local-shard values into the live (possibly DTensor) state
-    dict structure, in place."""
-    try:
-        from torch.distributed.tensor import DTensor
-    except ImportError:  # pragma: no cover
-        DTensor = ()
-    if isinstance(dst, dict):
-        for k in dst:
-            if k in src:
-                v = dst[k]
-                if isinstance(v, (dict, list)):
-                    _fill_local(v, src[k])
-                elif DTensor and isinstance(v, DTensor):
-                    v.to_local().copy_(src[k].to(v.device))
-                elif isinstance(v, torch.Tensor):
-                    v.copy_(src[k].to(v.device))
-                else:
-                    dst[k] = src[k]
-    elif isinstance(dst, list):
-        for i, v in enumerate(dst):
-            if i < len(src):
-                if isinstance(v, (dict, list)):
-                    _fill_local(v, src[i])
-                elif isinstance(v, torch.Tensor):
-                    v.copy_(src[i].to(v.device))
-                else:
-                    dst[i] = src[i]
