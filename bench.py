@@ -167,9 +167,10 @@ def main():
             # simulated failure recovery: reload model+optimizer from shm
             cp.engine.shm_handler.wait_drained()
             t0 = time.perf_counter()
-            sd = cp.engine.load(device=device)
+            sd = cp.engine.restore_into(
+                model, None if args.ckpt_scope == "model" else opt
+            )
             assert sd is not None, "no checkpoint in shm to restore from"
-            cp.engine.load_into(model, None if args.ckpt_scope == "model" else opt, sd)
             sync()
             restore_s = time.perf_counter() - t0
     if cp is not None:

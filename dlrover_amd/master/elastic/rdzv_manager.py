@@ -1,0 +1,299 @@
+"""Master-side rendezvous managers.
+
+Parity target: ref dlrover/python/master/elastic_training/rdzv_manager.py
+(RendezvousManager :69, ElasticTrainingRendezvousManager :497,
+NetworkCheckRendezvousManager :599). Semantics preserved:
+
+  - nodes join a round; the round COMPLETES when every alive node has joined,
+    or when >= min_nodes are waiting and the last-call timer expires — the
+    world is then truncated down to a multiple of node_unit;
+  - agents poll get_comm_world until their round completes; the comm world
+    maps node_rank -> local_world_size (GPUs contributed);
+  - num_nodes_waiting > 0 signals running agents that membership changed and
+    they should gracefully restart workers into a new rendezvous
+    (ref: training.py:1687);
+  - the NETWORK_CHECK plane groups nodes in probe pairs: round 0 adjacent
+    pairs, round 1 re-pairs fastest with slowest so a fault node is isolated
+    in two rounds (BASELINE.md straggler localization).
+
+MI355X note: the comm world feeds RCCL process-group formation over xGMI;
+rank ordering is the sorted node-rank order (single-node boxes are fully
+connected via 7 xGMI links/GPU, so no switch-topology sort is needed until
+multi-node — DpTopologySorter hooks in here when node topology is reported).
+"""
+
+import time
+from threading import Lock
+from typing import Dict, List, Optional, Tuple
+
+from dlrover_amd.common.constants import NetworkFailureReason, RendezvousName
+from dlrover_amd.common.log import logger
+
+
+class RendezvousParameters:
+    def __init__(
+        self,
+        min_nodes: int = 1,
+        max_nodes: int = 1,
+        waiting_timeout: float = 60.0,
+        node_unit: int = 1,
+        joint_timeout: float = 600.0,
+    ):
+        self.min_nodes = min_nodes
+        self.max_nodes = max_nodes
+        self.waiting_timeout = waiting_timeout
+        self.node_unit = max(1, node_unit)
+        self.joint_timeout = joint_timeout
+
+
+class RendezvousManager:
+    def __init__(self, name: str = RendezvousName.TRAINING):
+        self.name = name
+        self._lock = Lock()
+        self._params = RendezvousParameters()
+        self._waiting_nodes: Dict[int, int] = {}  # node_rank -> local world
+        self._rdzv_nodes: Dict[int, int] = {}  # the completed world
+        self._alive_nodes: set = set()
+        self._rdzv_round = 0
+        self._lastcall_time = 0.0
+        self._start_round_time = 0.0
+        self._node_unit = 1
+        self._latest_join_time: Dict[int, float] = {}
+
+    # -- configuration ---------------------------------------------------------
+
+    def update_rdzv_params(
+        self, min_nodes: int, max_nodes: int, waiting_timeout: float, node_unit: int
+    ):
+        with self._lock:
+            self._params = RendezvousParameters(
+                min_nodes, max_nodes, waiting_timeout, node_unit
+            )
+            logger.info(
+                "[%s] rdzv params: min=%s max=%s timeout=%s unit=%s",
+                self.name,
+                min_nodes,
+                max_nodes,
+                waiting_timeout,
+                node_unit,
+            )
+
+    @property
+    def min_nodes(self) -> int:
+        return self._params.min_nodes
+
+    @property
+    def max_nodes(self) -> int:
+        return self._params.max_nodes
+
+    # -- node liveness (driven by the job manager) -------------------------------
+
+    def add_alive_node(self, node_rank: int):
+        with self._lock:
+            self._alive_nodes.add(node_rank)
+
+    def remove_alive_node(self, node_rank: int):
+        with self._lock:
+            self._alive_nodes.discard(node_rank)
+            # a dead node can no longer hold up or belong to a pending round
+            self._waiting_nodes.pop(node_rank, None)
+
+    # -- join / completion --------------------------------------------------------
+
+    def join_rendezvous(self, node_rank: int, local_world_size: int) -> int:
+        with self._lock:
+            if node_rank not in self._waiting_nodes:
+                self._waiting_nodes[node_rank] = local_world_size
+                self._latest_join_time[node_rank] = time.time()
+                self._alive_nodes.add(node_rank)
+                if self._lastcall_time == 0.0:
+                    self._lastcall_time = time.time()
+                logger.info(
+                    "[%s] node %s joined (%s waiting, %s alive)",
+                    self.name,
+                    node_rank,
+                    len(self._waiting_nodes),
+                    len(self._alive_nodes),
+                )
+            return self._rdzv_round
+
+    def _check_rdzv_completed_locked(self) -> bool:
+        """ref: rdzv_manager.py:183 — complete when every alive node joined,
+        or on last-call timeout with >= min_nodes waiting (truncated to a
+        node_unit multiple)."""
+        waiting = len(self._waiting_nodes)
+        if waiting == 0:
+            return False
+        p = self._params
+        alive = max(len(self._alive_nodes), 1)
+        target = min(alive, p.max_nodes)
+        completed = False
+        if waiting >= target and waiting >= p.min_nodes:
+            completed = True
+        elif (
+            waiting >= p.min_nodes
+            and self._lastcall_time > 0
+            and time.time() - self._lastcall_time > p.waiting_timeout
+            and waiting % p.node_unit == 0
+        ):
+            completed = True
+        if not completed:
+            return False
+        # truncate to a multiple of node_unit, dropping the highest ranks
+        keep = (waiting // p.node_unit) * p.node_unit
+        ranks = sorted(self._waiting_nodes)[:keep]
+        self._rdzv_nodes = {r: self._waiting_nodes[r] for r in ranks}
+        dropped = [r for r in self._waiting_nodes if r not in self._rdzv_nodes]
+        self._waiting_nodes = {
+            r: w for r, w in self._waiting_nodes.items() if r in dropped
+        }
+        self._rdzv_round += 1
+        self._lastcall_time = 0.0
+        logger.info(
+            "[%s] rendezvous round %s completed: world=%s dropped=%s",
+            self.name,
+            self._rdzv_round,
+            self._rdzv_nodes,
+            dropped,
+        )
+        return True
+
+    def get_comm_world(self, node_rank: int) -> Tuple[int, int, Dict[int, int]]:
+        """Returns (round, group, world). world empty while incomplete."""
+        with self._lock:
+            if node_rank in self._waiting_nodes:
+                self._check_rdzv_completed_locked()
+            if node_rank in self._rdzv_nodes:
+                return self._rdzv_round, 0, dict(self._rdzv_nodes)
+            return self._rdzv_round, 0, {}
+
+    def num_nodes_waiting(self) -> int:
+        with self._lock:
+            return len(self._waiting_nodes)
+
+    def current_world(self) -> Dict[int, int]:
+        with self._lock:
+            return dict(self._rdzv_nodes)
+
+    @property
+    def rdzv_round(self) -> int:
+        with self._lock:
+            return self._rdzv_round
+
+
+class ElasticTrainingRendezvousManager(RendezvousManager):
+    def __init__(self):
+        super().__init__(RendezvousName.TRAINING)
+
+
+class NetworkCheckRendezvousManager(RendezvousManager):
+    """Probe-pair grouping + fault/straggler localization
+    (ref: rdzv_manager.py:599-875)."""
+
+    def __init__(self):
+        super().__init__(RendezvousName.NETWORK_CHECK)
+        self._node_status: Dict[int, bool] = {}
+        self._node_elapsed: Dict[int, Dict[int, float]] = {}  # round -> {rank: s}
+        self._check_round = 0
+        self._fault_nodes: set = set()
+        self._straggler_nodes: set = set()
+        self.straggler_ratio = 2.0  # slower than 2x the median => straggler
+
+    def get_comm_world(self, node_rank: int) -> Tuple[int, int, Dict[int, int]]:
+        """Split the completed world into probe groups; returns this node's
+        group world. round 0: adjacent pairs; round 1: fastest<->slowest."""
+        with self._lock:
+            if node_rank in self._waiting_nodes:
+                self._check_rdzv_completed_locked()
+            if node_rank not in self._rdzv_nodes:
+                return self._rdzv_round, 0, {}
+            groups = self._build_groups_locked()
+            for gi, group in enumerate(groups):
+                if node_rank in group:
+                    world = {r: self._rdzv_nodes[r] for r in group}
+                    return self._rdzv_round, gi, world
+            return self._rdzv_round, 0, {}
+
+    def _build_groups_locked(self) -> List[List[int]]:
+        ranks = sorted(self._rdzv_nodes)
+        if len(ranks) <= 2:
+            return [ranks]
+        if self._check_round == 0 or not self._node_elapsed.get(self._check_round - 1):
+            pairs = [ranks[i : i + 2] for i in range(0, len(ranks), 2)]
+        else:
+            # pair fastest with slowest using the previous round's times
+            prev = self._node_elapsed[self._check_round - 1]
+            by_speed = sorted(ranks, key=lambda r: prev.get(r, float("inf")))
+            pairs = []
+            i, j = 0, len(by_speed) - 1
+            while i < j:
+                pairs.append(sorted([by_speed[i], by_speed[j]]))
+                i += 1
+                j -= 1
+            if i == j:
+                pairs.append([by_speed[i]])
+        # a singleton group cannot run a collective probe: merge into previous
+        if pairs and len(pairs[-1]) == 1 and len(pairs) > 1:
+            pairs[-2].extend(pairs.pop())
+        return pairs
+
+    def report_network_check_result(self, node_rank: int, normal: bool, elapsed: float):
+        with self._lock:
+            self._node_status[node_rank] = normal
+            self._node_elapsed.setdefault(self._check_round, {})[node_rank] = elapsed
+            if len(self._node_elapsed[self._check_round]) == len(self._rdzv_nodes):
+                self._analyse_locked()
+                self._check_round += 1
+
+    def _analyse_locked(self):
+        elapsed = self._node_elapsed[self._check_round]
+        failed = {r for r, ok in self._node_status.items() if not ok}
+        if self._check_round == 0:
+            self._fault_nodes = set(failed)
+        else:
+            # a node failing in two different pairings is the fault node; a
+            # node that failed once but passed when re-paired is exonerated
+            self._fault_nodes &= failed
+            self._fault_nodes |= {
+                r for r in failed if r in self._fault_nodes or not self._node_status.get(r, True)
+            } & failed
+        times = sorted(elapsed.values())
+        if times:
+            median = times[len(times) // 2]
+            self._straggler_nodes = {
+                r
+                for r, t in elapsed.items()
+                if median > 0 and t > self.straggler_ratio * median
+            }
+        logger.info(
+            "[network-check] round %s elapsed=%s fault=%s straggler=%s",
+            self._check_round,
+            {k: round(v, 2) for k, v in elapsed.items()},
+            self._fault_nodes,
+            self._straggler_nodes,
+        )
+
+    def check_fault_node(self) -> Tuple[List[int], str]:
+        with self._lock:
+            if not self._node_status:
+                return [], NetworkFailureReason.NO_INIT
+            if len(self._node_status) < len(self._rdzv_nodes):
+                return [], NetworkFailureReason.WAITING_NODE
+            return sorted(self._fault_nodes), (
+                NetworkFailureReason.NODE_FAILURE if self._fault_nodes else ""
+            )
+
+    def get_stragglers(self) -> List[int]:
+        with self._lock:
+            return sorted(self._straggler_nodes)
+
+    def new_check(self):
+        """Reset state for a fresh 2-round check sequence."""
+        with self._lock:
+            self._node_status.clear()
+            self._node_elapsed.clear()
+            self._check_round = 0
+            self._fault_nodes.clear()
+            self._straggler_nodes.clear()
+            self._rdzv_nodes = {}
+            self._waiting_nodes = {}

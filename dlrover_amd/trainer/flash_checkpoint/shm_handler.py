@@ -290,8 +290,16 @@ class SharedMemoryHandler:
 
     # -- read path -------------------------------------------------------------
 
-    def load_state_dict(self, device: Optional[torch.device] = None) -> Optional[Any]:
-        """Reconstruct the state dict from shm (returns None if empty)."""
+    def load_state_dict(
+        self, device: Optional[torch.device] = None, zero_copy: bool = False
+    ) -> Optional[Any]:
+        """Reconstruct the state dict from shm (returns None if empty).
+
+        zero_copy=True returns tensors that VIEW the shm mapping directly —
+        no host-side copy at all. The caller must copy them into its own
+        storage (load_into does) before the next checkpoint overwrites the
+        segment. With a device, each tensor is a single H2D copy from the
+        page-locked mapping (PCIe-rate restore, no staging)."""
         meta = self.read_meta()
         if meta is None:
             return None
@@ -300,14 +308,20 @@ class SharedMemoryHandler:
             offset=self._payload_offset(),
         )
         pairs = list(meta.objects)
+        to_gpu = device is not None and device.type != "cpu"
         for tm in meta.tensors:
-            raw = torch.from_numpy(
-                host[tm.offset : tm.offset + tm.nbytes].copy()
-            )
-            t = raw.view(getattr(torch, tm.dtype)).view(tm.shape)
-            if device is not None and device.type != "cpu":
-                t = t.to(device, non_blocking=True)
+            view = torch.from_numpy(host[tm.offset : tm.offset + tm.nbytes])
+            view = view.view(getattr(torch, tm.dtype)).view(tm.shape)
+            if to_gpu:
+                t = torch.empty_like(view, device=device)
+                t.copy_(view, non_blocking=self._pinned)
+            elif zero_copy:
+                t = view
+            else:
+                t = view.clone()
             pairs.append((tm.path, t))
+        if to_gpu:
+            torch.cuda.synchronize()
         state = _build_skeleton(pairs)
         return state
 
