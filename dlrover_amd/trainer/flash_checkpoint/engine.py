@@ -369,7 +369,9 @@ class ShardedCheckpointEngine(CheckpointEngine):
         if sd is None:
             return None
         self.load_into(model, optimizer, sd)
-        return sd
+        # drop the zero-copy views: live references into shm.buf would make
+        # the mapping unclosable (BufferError: exported pointers exist)
+        return _map_leaves(sd, lambda v: None if isinstance(v, torch.Tensor) else v)
 
 
 def _to_local(t):

@@ -188,7 +188,14 @@ class SharedMemoryHandler:
             if self._pinned:
                 hipmem.host_unregister(self._buf_addr())
                 self._pinned = False
-            self._shm.close()
+            try:
+                self._shm.close()
+            except BufferError:
+                # zero-copy views of the mapping are still alive somewhere;
+                # leave the mapping open — the segment outlives us anyway
+                logger.warning(
+                    "shm %s still has live views; mapping left open", self.name
+                )
             self._shm = None
 
     def unlink(self):
