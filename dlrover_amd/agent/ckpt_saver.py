@@ -206,7 +206,8 @@ class AsyncCheckpointSaver:
             logger.warning("no shm handler for local rank %s", event.local_rank)
             return
         ok = persist_shm_to_storage(
-            handler, event, self.storage, self.checkpoint_dir, self.expected_shards
+            handler, event, self.storage, self.checkpoint_dir,
+            getattr(event, "expected_shards", 0) or self.expected_shards,
         )
         if ok:
             self._persisted_steps[event.local_rank] = event.step

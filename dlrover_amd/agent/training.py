@@ -1,0 +1,396 @@
+"""Node-level elastic agent: master-coordinated rendezvous + worker process
+management + failure handling.
+
+Parity target: ref dlrover/python/elastic_agent/torch/training.py —
+MasterRendezvousHandler :405-646, ElasticTrainingAgent :648 with the
+_invoke_run monitor loop :1247-1447 (FAILED -> persist shm ckpt + diagnose +
+restart/relaunch/abort; membership change -> graceful restart into a new
+rendezvous), launch_agent :1868.
+
+MI355X specifics: workers re-form RCCL process groups over xGMI on every
+restart — the agent guarantees worker processes are fully dead (SIGTERM then
+SIGKILL via torchelastic pcontext) before a new rendezvous so no stale HIP
+contexts or RCCL communicators hold GPU memory.
+"""
+
+import os
+import threading
+import time
+import traceback
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional, Tuple, Union
+
+import torch.distributed.elastic.timer as timer
+from torch.distributed.elastic.agent.server.api import (
+    RunResult,
+    WorkerGroup,
+    WorkerSpec,
+    WorkerState,
+)
+from torch.distributed.elastic.agent.server.local_elastic_agent import (
+    LocalElasticAgent,
+)
+from torch.distributed.elastic.multiprocessing import DefaultLogsSpecs
+from torch.distributed.elastic.rendezvous.api import (
+    RendezvousHandler,
+    RendezvousInfo,
+    RendezvousParameters,
+    RendezvousStoreInfo,
+)
+
+from dlrover_amd.agent.ckpt_saver import AsyncCheckpointSaver
+from dlrover_amd.agent.master_client import MasterClient
+from dlrover_amd.agent.master_kv_store import MasterKVStore
+from dlrover_amd.common.constants import (
+    JobConstant,
+    NodeEnv,
+    NodeEventType,
+    RendezvousName,
+    TrainingExceptionLevel,
+)
+from dlrover_amd.common.log import logger
+from dlrover_amd.common.multi_process import IPCServer
+from dlrover_amd.diagnosis.actions import DiagnosisActionType, action_from_wire
+
+
+@dataclass
+class ElasticLaunchConfig:
+    """Launch configuration (ref: ElasticLaunchConfig, training.py:300-400)."""
+
+    min_nodes: int = 1
+    max_nodes: int = 1
+    nproc_per_node: int = 1
+    node_rank: int = 0
+    max_restarts: int = 3
+    monitor_interval: float = 5.0
+    rdzv_timeout: float = 600.0
+    waiting_timeout: float = 60.0
+    node_unit: int = 1
+    network_check: bool = False
+    comm_perf_test: bool = False
+    run_id: str = "dlrover"
+    checkpoint_dir: str = "/tmp/dlrover_amd_ckpt"
+    log_dir: Optional[str] = None
+    redirects: str = ""
+    numa_affinity: bool = False
+    training_port: int = 0
+
+    def auto_configure(self):
+        """Fill nproc from visible GPUs (ref: auto_configure_params :345)."""
+        if self.nproc_per_node <= 0:
+            try:
+                import torch
+
+                self.nproc_per_node = max(torch.cuda.device_count(), 1)
+            except Exception:  # noqa: BLE001
+                self.nproc_per_node = 1
+
+
+class MasterRendezvousHandler(RendezvousHandler):
+    """torch RendezvousHandler backed by master RPC (ref: training.py:405)."""
+
+    def __init__(
+        self,
+        name: str,
+        node_rank: int,
+        local_world_size: int,
+        client: Optional[MasterClient] = None,
+        rdzv_timeout: float = 600.0,
+        local_addr: str = "",
+    ):
+        self._name = name
+        self._node_rank = node_rank
+        self._local_world_size = local_world_size
+        self._client = client or MasterClient.singleton_instance()
+        self._rdzv_timeout = rdzv_timeout
+        self._local_addr = local_addr or "127.0.0.1"
+        self._closed = False
+        self.join_timeout = rdzv_timeout
+
+    def get_backend(self) -> str:
+        return "dlrover-master"
+
+    def get_run_id(self) -> str:
+        return self._name
+
+    @property
+    def use_agent_store(self) -> bool:
+        # torchelastic reads this as a property (local_elastic_agent.py)
+        return False
+
+    def is_closed(self) -> bool:
+        return self._closed
+
+    def set_closed(self):
+        self._closed = True
+
+    def num_nodes_waiting(self) -> int:
+        return self._client.num_nodes_waiting(self._name)
+
+    def next_rendezvous(self) -> RendezvousInfo:
+        start = time.time()
+        self._client.join_rendezvous(
+            self._node_rank,
+            self._local_world_size,
+            rdzv_name=self._name,
+            node_ip=self._local_addr,
+        )
+        while True:
+            rnd, group, world = self._client.get_comm_world(
+                self._name, self._node_rank
+            )
+            if world and self._node_rank in world:
+                break
+            if time.time() - start > self._rdzv_timeout:
+                raise TimeoutError(
+                    f"rendezvous {self._name} timed out after "
+                    f"{self._rdzv_timeout}s (world={world})"
+                )
+            time.sleep(JobConstant.RENDEZVOUS_DEFAULT_INTERVAL)
+        ranks = sorted(world)
+        group_rank = ranks.index(self._node_rank)
+        group_world_size = len(ranks)
+        store = MasterKVStore(f"rdzv/{self._name}/{rnd}/{group}", self._client)
+        logger.info(
+            "rendezvous %s round=%s: node %s -> group_rank %s/%s",
+            self._name,
+            rnd,
+            self._node_rank,
+            group_rank,
+            group_world_size,
+        )
+        bootstrap = RendezvousStoreInfo.build(
+            group_rank, store, local_addr=self._local_addr
+        )
+        return RendezvousInfo(store, group_rank, group_world_size, bootstrap)
+
+    def shutdown(self) -> bool:
+        self._closed = True
+        return True
+
+
+class ElasticTrainingAgent(LocalElasticAgent):
+    """ref: ElasticTrainingAgent (training.py:648) — adds to torchelastic's
+    LocalElasticAgent: master heartbeats carrying DiagnosisActions, shm
+    checkpoint persistence on failure, membership-change restarts driven by
+    the master rendezvous, and failure reporting."""
+
+    def __init__(
+        self,
+        spec: WorkerSpec,
+        config: ElasticLaunchConfig,
+        client: Optional[MasterClient] = None,
+        start_method: str = "spawn",
+        logs_specs: Optional[DefaultLogsSpecs] = None,
+        exit_barrier_timeout: float = 300,
+    ):
+        logs_specs = logs_specs or DefaultLogsSpecs(log_dir=config.log_dir or None)
+        super().__init__(
+            spec,
+            logs_specs=logs_specs,
+            start_method=start_method,
+            exit_barrier_timeout=exit_barrier_timeout,
+        )
+        self.config = config
+        self.client = client or MasterClient.singleton_instance()
+        self._restart_requested = threading.Event()
+        self._abort_requested = threading.Event()
+        self._hb_stop = threading.Event()
+        self._hb_thread: Optional[threading.Thread] = None
+
+    # -- heartbeats + master-pushed actions (ref: diagnosis_agent.py:286) --------
+
+    def _start_heartbeats(self):
+        def loop():
+            while not self._hb_stop.wait(15.0):
+                try:
+                    resp = self.client.report_heart_beat(self.config.node_rank)
+                    action = action_from_wire(resp.action_cls, resp.action_kwargs)
+                    if action is None or not action.is_needed():
+                        continue
+                    logger.info("master pushed action: %s", action)
+                    if action.action_type == DiagnosisActionType.JOB_ABORT:
+                        self._abort_requested.set()
+                    elif action.action_type in (
+                        DiagnosisActionType.RESTART_WORKER,
+                        DiagnosisActionType.RELAUNCH_WORKER,
+                    ):
+                        self._restart_requested.set()
+                except Exception:  # noqa: BLE001
+                    logger.warning("heartbeat failed:\n%s", traceback.format_exc())
+
+        self._hb_thread = threading.Thread(target=loop, daemon=True, name="agent-hb")
+        self._hb_thread.start()
+
+    # -- failure-path checkpoint persist (ref: training.py:1533) --------------------
+
+    def _save_ckpt_to_storage(self):
+        saver = AsyncCheckpointSaver.get_ckpt_saver()
+        if saver is not None:
+            try:
+                saver.save_shm_to_storage()
+            except Exception:  # noqa: BLE001
+                logger.exception("failure-path checkpoint persist failed")
+
+    # -- the monitor loop (ref: _invoke_run :1247) -----------------------------------
+
+    def _invoke_run(self, role: str = "default") -> RunResult:
+        spec = self._worker_group.spec
+        self._start_heartbeats()
+        self._initialize_workers(self._worker_group)
+        monitor_interval = spec.monitor_interval
+        rdzv_handler = spec.rdzv_handler
+
+        while True:
+            time.sleep(monitor_interval)
+            run_result = self._monitor_workers(self._worker_group)
+            state = run_result.state
+            self._worker_group.state = state
+
+            if self._abort_requested.is_set():
+                logger.error("job abort requested by master")
+                self._save_ckpt_to_storage()
+                self._stop_workers(self._worker_group)
+                self._worker_group.state = WorkerState.FAILED
+                return run_result
+
+            if state == WorkerState.SUCCEEDED:
+                logger.info("workers finished successfully")
+                self._exit_barrier()
+                self._report_event(NodeEventType.SUCCEEDED_EXITED)
+                return run_result
+
+            if state in (WorkerState.UNHEALTHY, WorkerState.FAILED):
+                logger.error(
+                    "worker group %s: persisting checkpoint then deciding", state.name
+                )
+                self._save_ckpt_to_storage()
+                self._report_failures(run_result)
+                if self._remaining_restarts > 0:
+                    self._remaining_restarts -= 1
+                    logger.info(
+                        "restarting workers (%s restarts left)",
+                        self._remaining_restarts,
+                    )
+                    self._restart_workers(self._worker_group)
+                    continue
+                self._stop_workers(self._worker_group)
+                self._worker_group.state = WorkerState.FAILED
+                self._report_event(NodeEventType.FAILED_EXITED, "restarts exhausted")
+                return run_result
+
+            if state == WorkerState.HEALTHY:
+                if self._restart_requested.is_set():
+                    self._restart_requested.clear()
+                    logger.info("restart requested (diagnosis): restarting workers")
+                    self._save_ckpt_to_storage()
+                    self._restart_workers(self._worker_group)
+                    continue
+                # membership change: another node joined/waits for a round
+                num_waiting = rdzv_handler.num_nodes_waiting()
+                if num_waiting > 0:
+                    logger.info(
+                        "%s nodes waiting: restarting into a new rendezvous "
+                        "(does not consume the restart budget)",
+                        num_waiting,
+                    )
+                    self._restart_workers(self._worker_group)
+                continue
+
+            raise RuntimeError(f"unknown worker state {state}")
+
+    def _report_failures(self, run_result: RunResult):
+        try:
+            errs = {
+                rank: f.message if hasattr(f, "message") else str(f)
+                for rank, f in (run_result.failures or {}).items()
+            }
+            self.client.report_failure(
+                str(errs),
+                TrainingExceptionLevel.PROCESS_ERROR,
+                restart_count=self._worker_group.spec.max_restarts
+                - self._remaining_restarts,
+            )
+        except Exception:  # noqa: BLE001
+            logger.warning("failure report to master failed")
+
+    def _report_event(self, event_type: str, reason: str = ""):
+        try:
+            self.client.report_node_event(event_type, reason, self.config.node_rank)
+        except Exception:  # noqa: BLE001
+            logger.warning("node event report failed")
+
+    def run(self, role: str = "default") -> RunResult:
+        try:
+            return super().run(role)
+        finally:
+            self._hb_stop.set()
+
+
+def _build_worker_spec(
+    config: ElasticLaunchConfig,
+    entrypoint: Union[str, Any],
+    args: Tuple,
+    client: MasterClient,
+) -> WorkerSpec:
+    rdzv_handler = MasterRendezvousHandler(
+        RendezvousName.TRAINING,
+        config.node_rank,
+        config.nproc_per_node,
+        client=client,
+        rdzv_timeout=config.rdzv_timeout,
+        local_addr=os.getenv("POD_IP", "127.0.0.1"),
+    )
+    return WorkerSpec(
+        role="default",
+        local_world_size=config.nproc_per_node,
+        entrypoint=entrypoint,
+        args=tuple(args),
+        rdzv_handler=rdzv_handler,
+        max_restarts=config.max_restarts,
+        monitor_interval=config.monitor_interval,
+    )
+
+
+def launch_agent(
+    config: ElasticLaunchConfig,
+    entrypoint: Union[str, Any],
+    args: List[str],
+) -> Dict[int, Any]:
+    """Run the elastic agent on this node (ref: launch_agent :1868):
+    start the IPC server + async ckpt saver, report rendezvous parameters,
+    run the (optional) node check, then the training agent."""
+    config.auto_configure()
+    client = MasterClient.singleton_instance()
+    client.report_rdzv_params(
+        config.min_nodes,
+        config.max_nodes,
+        config.waiting_timeout,
+        config.node_unit,
+    )
+
+    ipc_server = IPCServer().start()
+    saver = AsyncCheckpointSaver.start_async_saving_ckpt(
+        checkpoint_dir=config.checkpoint_dir,
+        local_world_size=config.nproc_per_node,
+        expected_shards=config.max_nodes * config.nproc_per_node,
+    )
+    saver.register_signal_handlers()
+
+    if config.network_check:
+        from dlrover_amd.agent.node_check_agent import run_network_check
+
+        run_network_check(config, client)
+
+    spec = _build_worker_spec(config, entrypoint, args, client)
+    agent = ElasticTrainingAgent(spec, config, client=client)
+    try:
+        result = agent.run()
+        if result.is_failed():
+            raise RuntimeError(f"workers failed: {result.failures}")
+        return result.return_values
+    finally:
+        spec.rdzv_handler.shutdown()
+        saver.save_shm_to_storage()
+        AsyncCheckpointSaver.reset()
+        ipc_server.stop()

@@ -53,6 +53,7 @@ class CheckpointEvent:
     local_rank: int = 0
     global_rank: int = 0
     world_size: int = 1
+    expected_shards: int = 1
 
 
 def _local_rank() -> int:
@@ -147,6 +148,10 @@ class CheckpointEngine:
         """Does THIS rank write a shard? (DDP full: rank0 only; FSDP: all)."""
         return True
 
+    def expected_shards(self) -> int:
+        """How many shard files make a COMPLETE checkpoint (commit gate)."""
+        return _world_size()
+
     # -- save ------------------------------------------------------------------
 
     def save_to_memory(
@@ -190,6 +195,7 @@ class CheckpointEngine:
             local_rank=self._local_rank,
             global_rank=_global_rank(),
             world_size=_world_size(),
+            expected_shards=self.expected_shards(),
         )
         if self._event_queue is not None:
             if self.rank_saves():
@@ -289,7 +295,7 @@ class _StandaloneSaver:
                     event,
                     self._engine.storage,
                     self._engine.checkpoint_dir,
-                    expected_shards=1,
+                    expected_shards=getattr(event, "expected_shards", 1) or 1,
                 )
             except Exception:  # noqa: BLE001
                 logger.exception("standalone checkpoint persist failed")
@@ -304,6 +310,9 @@ class FullCheckpointEngine(CheckpointEngine):
 
     def rank_saves(self) -> bool:
         return _global_rank() == 0
+
+    def expected_shards(self) -> int:
+        return 1
 
     def gather_state_dict(self, model, optimizer):
         mod = model.module if hasattr(model, "module") else model
