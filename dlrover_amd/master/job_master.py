@@ -35,7 +35,7 @@ from dlrover_amd.master.shard.task_manager import TaskManager
 class JobMaster:
     def __init__(
         self,
-        port: int = 0,
+        port: Optional[int] = None,
         service_type: str = "",
         job_manager=None,
         elastic_run_configs: Optional[Dict[str, str]] = None,
@@ -57,7 +57,8 @@ class JobMaster:
         self.job_manager.rdzv_managers = self.rdzv_managers
         self._elastic_run_configs = elastic_run_configs or {}
         self._service_type = service_type or cfg.master_service_type
-        self._port = port or cfg.master_port
+        # port=0 means "bind an ephemeral port" (standalone); None = default
+        self._port = cfg.master_port if port is None else port
         self._server = None
         self._ckpt_sync_nodes: Dict[int, int] = {}
         self._ckpt_sync_lock = threading.Lock()

@@ -138,6 +138,17 @@ def run(args) -> int:
         if not wait_for_server(master_addr, timeout=120, service_type=args.service_type):
             raise SystemExit(f"master {master_addr} unreachable")
 
+    # worker processes must be able to import dlrover_amd regardless of the
+    # script's location (the package runs in-tree, not from site-packages)
+    import dlrover_amd
+
+    pkg_root = os.path.dirname(os.path.dirname(os.path.abspath(dlrover_amd.__file__)))
+    existing = os.environ.get("PYTHONPATH", "")
+    if pkg_root not in existing.split(os.pathsep):
+        os.environ["PYTHONPATH"] = (
+            f"{pkg_root}{os.pathsep}{existing}" if existing else pkg_root
+        )
+
     os.environ[NodeEnv.MASTER_ADDR] = master_addr
     os.environ[NodeEnv.MASTER_SERVICE_TYPE] = args.service_type
     os.environ[NodeEnv.NODE_ID] = str(args.node_rank)

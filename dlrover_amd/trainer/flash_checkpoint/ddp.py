@@ -5,6 +5,7 @@ state into its shm segment; restore broadcasts nothing — every rank reads the
 same full checkpoint (shm on node 0's ranks, storage elsewhere).
 """
 
+import os
 from typing import Optional
 
 import torch.distributed as dist
@@ -45,6 +46,25 @@ class DdpCheckpointer(Checkpointer):
 
     def load_checkpoint(self, resume_path: str = "") -> Optional[dict]:
         sd = self.engine.load(resume_path)
+        # all ranks must resume the SAME step or DDP deadlocks on divergent
+        # loop lengths: agree on the minimum committed step across ranks
+        if dist.is_available() and dist.is_initialized() and dist.get_world_size() > 1:
+            import torch
+
+            my_step = int(sd.get("step", -1)) if sd is not None else -1
+            t = torch.tensor([my_step], dtype=torch.long)
+            dist.all_reduce(t, op=dist.ReduceOp.MIN)
+            agreed = int(t.item())
+            if agreed < 0:
+                return None  # at least one rank has nothing: fresh start
+            if agreed != my_step:
+                sd = self.engine.load_from_storage(
+                    os.path.join(self.checkpoint_dir, str(agreed))
+                )
+                if sd is None:
+                    raise RuntimeError(
+                        f"cannot load agreed checkpoint step {agreed}"
+                    )
         if sd is None:
             return None
         if self.model is not None:

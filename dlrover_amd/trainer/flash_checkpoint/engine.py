@@ -314,6 +314,27 @@ class FullCheckpointEngine(CheckpointEngine):
     def expected_shards(self) -> int:
         return 1
 
+    @staticmethod
+    def _shard_file_name(global_rank: int) -> str:
+        # the full checkpoint is always rank 0's shard — every rank reads it
+        return "rank_00000.pt"
+
+    def load(self, path: str = "", device=None):
+        """Full-checkpoint load order: own shm, local rank 0's shm (the full
+        state lives there on this node), then storage. All ranks MUST resolve
+        the same step or DDP ranks would resume divergent and deadlock."""
+        sd = self.shm_handler.load_state_dict(device=device)
+        if sd is None and self._local_rank != 0:
+            h0 = SharedMemoryHandler(
+                shm_segment_name(self._job, 0), host_pin=False
+            )
+            if h0.attach():
+                sd = h0.load_state_dict(device=device)
+                h0.close()
+        if sd is None:
+            sd = self.load_from_storage(path, device=device)
+        return sd
+
     def gather_state_dict(self, model, optimizer):
         mod = model.module if hasattr(model, "module") else model
         sd = {"model": mod.state_dict(), "step": self._last_save_step}
