@@ -1,0 +1,91 @@
+"""ElasticTrainer: keep the EFFECTIVE global batch size fixed while the
+world size changes (ref: dlrover/trainer/torch/elastic/trainer.py:181-340).
+
+Mechanism: gradient accumulation steps = ceil(max_workers / current_world);
+optimizer.step() fires every `accum` micro-batches, with DDP gradient sync
+suppressed (no_sync) on non-boundary micro-batches. Writes the global step to
+a file the agent's training monitor reports to the master (throughput +
+hang detection input).
+"""
+
+import contextlib
+import json
+import os
+import time
+from typing import Optional
+
+import torch.distributed as dist
+
+from dlrover_amd.common.log import logger
+
+STEP_FILE_DIR = "/tmp/dlrover_amd_monitor"
+
+
+def _world_size() -> int:
+    if dist.is_available() and dist.is_initialized():
+        return dist.get_world_size()
+    return int(os.getenv("WORLD_SIZE", "1"))
+
+
+class ElasticTrainer:
+    def __init__(self, model, dataloader=None, max_workers: Optional[int] = None):
+        self.model = model
+        self.dataloader = dataloader
+        self.max_workers = max_workers or int(
+            os.getenv("DLROVER_MAX_WORKERS", str(_world_size()))
+        )
+        self.gradient_state = _GradState(self._accum_steps())
+        self.global_step = 0
+        self._step_file = os.path.join(
+            STEP_FILE_DIR, f"global_step_{os.getenv('ELASTIC_JOB_NAME', 'job')}.json"
+        )
+        os.makedirs(STEP_FILE_DIR, exist_ok=True)
+
+    def _accum_steps(self) -> int:
+        world = max(_world_size(), 1)
+        return max(1, (self.max_workers + world - 1) // world)
+
+    def reset(self):
+        """Call after a re-rendezvous changed the world size."""
+        self.gradient_state = _GradState(self._accum_steps())
+        logger.info(
+            "elastic trainer: world=%s accumulation=%s",
+            _world_size(),
+            self.gradient_state.accum_steps,
+        )
+
+    @contextlib.contextmanager
+    def step(self):
+        """Context manager around one micro-batch: suppresses DDP allreduce
+        except on accumulation boundaries (ref: trainer.py:271-277)."""
+        gs = self.gradient_state
+        gs.micro_step += 1
+        sync = gs.micro_step % gs.accum_steps == 0
+        ctx = contextlib.nullcontext()
+        if not sync and hasattr(self.model, "no_sync"):
+            ctx = self.model.no_sync()
+        with ctx:
+            yield sync
+        if sync:
+            self.global_step += 1
+            self._report_step()
+
+    @property
+    def step_boundary(self) -> bool:
+        gs = self.gradient_state
+        return gs.micro_step % gs.accum_steps == 0
+
+    def _report_step(self):
+        try:
+            rank = dist.get_rank() if dist.is_initialized() else 0
+            if rank == 0:
+                with open(self._step_file, "w") as f:
+                    json.dump({"step": self.global_step, "ts": time.time()}, f)
+        except OSError:
+            pass
+
+
+class _GradState:
+    def __init__(self, accum_steps: int):
+        self.accum_steps = accum_steps
+        self.micro_step = 0

@@ -1,0 +1,67 @@
+"""ElasticTrainer / sampler / sharding client behaviors (CPU, no dist)."""
+
+import torch
+
+from dlrover_amd.trainer.elastic import ElasticDistributedSampler, ElasticTrainer
+
+
+def test_elastic_trainer_accumulation(monkeypatch):
+    monkeypatch.setenv("DLROVER_MAX_WORKERS", "4")
+    monkeypatch.setenv("WORLD_SIZE", "1")
+    model = torch.nn.Linear(4, 4)
+    tr = ElasticTrainer(model)
+    assert tr.gradient_state.accum_steps == 4
+    boundaries = []
+    for _ in range(8):
+        with tr.step() as sync:
+            boundaries.append(sync)
+    assert boundaries == [False, False, False, True] * 2
+    assert tr.global_step == 2
+
+
+def test_elastic_trainer_reset_on_world_change(monkeypatch):
+    monkeypatch.setenv("DLROVER_MAX_WORKERS", "8")
+    monkeypatch.setenv("WORLD_SIZE", "2")
+    tr = ElasticTrainer(torch.nn.Linear(2, 2))
+    assert tr.gradient_state.accum_steps == 4
+    monkeypatch.setenv("WORLD_SIZE", "8")
+    tr.reset()
+    assert tr.gradient_state.accum_steps == 1
+
+
+def test_sampler_partitions_disjoint_and_complete():
+    data = list(range(20))
+    samplers = [
+        ElasticDistributedSampler(data, num_replicas=4, rank=r, shuffle=False)
+        for r in range(4)
+    ]
+    seen = []
+    for s in samplers:
+        seen += list(iter(s))
+    assert sorted(seen) == sorted(list(range(20)))
+
+
+def test_sampler_resume_skips_consumed():
+    data = list(range(16))
+    s = ElasticDistributedSampler(data, num_replicas=2, rank=0, shuffle=False)
+    state = s.state_dict(step=2, batch_size=2)  # consumed 2*2*2=8 samples
+    s2 = ElasticDistributedSampler(data, num_replicas=2, rank=0, shuffle=False)
+    s2.load_state_dict(state)
+    remaining = list(iter(s2))
+    assert remaining == [8, 10, 12, 14]
+    assert len(s2) == 4
+
+
+def test_sampler_reshard_after_scale():
+    data = list(range(24))
+    state = {"epoch": 0, "completed_num": 8}
+    # resume with a DIFFERENT world size: remaining samples still disjoint
+    samplers = [
+        ElasticDistributedSampler(data, num_replicas=3, rank=r, shuffle=False)
+        for r in range(3)
+    ]
+    out = []
+    for s in samplers:
+        s.load_state_dict(state)
+        out += list(iter(s))
+    assert sorted(out) == list(range(8, 24))
