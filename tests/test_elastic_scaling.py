@@ -1,0 +1,61 @@
+"""Elastic scale up/down through the master rendezvous (config #4 plumbing):
+2 -> 4 -> 2 node worlds with re-rendezvous, as the agents drive it."""
+
+import pytest
+
+from dlrover_amd.agent.master_client import MasterClient
+from dlrover_amd.common.constants import RendezvousName
+from dlrover_amd.master.job_master import LocalJobMaster
+from dlrover_amd.master.node.job_context import JobContext
+
+
+@pytest.fixture()
+def master():
+    JobContext._reset_for_tests()
+    m = LocalJobMaster(port=0).prepare()
+    yield m
+    m.stop()
+    JobContext._reset_for_tests()
+
+
+def _join_all(clients, ranks):
+    for r in ranks:
+        clients[r].join_rendezvous(r, 8)
+
+
+def test_scale_up_then_down(master):
+    clients = {i: MasterClient(f"127.0.0.1:{master.port}", node_id=i) for i in range(4)}
+    clients[0].report_rdzv_params(2, 4, waiting_timeout=600, node_unit=2)
+
+    # phase 1: nodes 0,1 form a world of 2
+    _join_all(clients, [0, 1])
+    _, _, world = clients[0].get_comm_world(RendezvousName.TRAINING, 0)
+    assert set(world) == {0, 1}
+
+    # phase 2: nodes 2,3 join -> running agents observe waiting>0
+    _join_all(clients, [2, 3])
+    assert clients[0].num_nodes_waiting() > 0
+    # agents restart workers and re-join; new world = 4 nodes
+    _join_all(clients, [0, 1])
+    _, _, world = clients[2].get_comm_world(RendezvousName.TRAINING, 2)
+    assert set(world) == {0, 1, 2, 3}
+
+    # phase 3: nodes 2,3 die (scale down); master removes them
+    for r in (2, 3):
+        master.rdzv_managers[RendezvousName.TRAINING].remove_alive_node(r)
+    _join_all(clients, [0, 1])
+    _, _, world = clients[0].get_comm_world(RendezvousName.TRAINING, 0)
+    assert set(world) == {0, 1}
+    for c in clients.values():
+        c.close()
+
+
+def test_scale_respects_node_unit(master):
+    clients = {i: MasterClient(f"127.0.0.1:{master.port}", node_id=i) for i in range(3)}
+    clients[0].report_rdzv_params(2, 4, waiting_timeout=600, node_unit=2)
+    _join_all(clients, [0, 1, 2])
+    _, _, world = clients[0].get_comm_world(RendezvousName.TRAINING, 0)
+    # 3 alive but node_unit=2: world truncated to 2 nodes
+    assert set(world) == {0, 1}
+    for c in clients.values():
+        c.close()
