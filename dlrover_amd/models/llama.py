@@ -107,18 +107,22 @@ class Attention(nn.Module):
         v = v.view(B, S, nkv, hd)
         q = rope_rotate(q, pos, cos, sin)
         k = rope_rotate(k, pos, cos, sin)
-        # [B, H, S, D]
-        q = q.transpose(1, 2)
-        k = k.transpose(1, 2)
+        # grouped-query attention with ZERO kv copies: fold the query heads
+        # of each kv group into the row dimension — q [B, nkv, rep*S, D]
+        # against k/v [B, nkv, S, D]. (repeat_interleave showed up as 11% of
+        # step time as __amd_rocclr_copyBuffer; see profiles/.)
+        rep = nh // nkv
+        q = q.view(B, S, nkv, rep, hd).permute(0, 2, 3, 1, 4)  # [B,nkv,rep,S,D]
+        q = q.reshape(B, nkv, rep * S, hd)
+        k = k.transpose(1, 2)  # [B, nkv, S, D]
         v = v.transpose(1, 2)
-        if nkv != nh:
-            rep = nh // nkv
-            k = k.repeat_interleave(rep, dim=1)
-            v = v.repeat_interleave(rep, dim=1)
-        scores = torch.matmul(q, k.transpose(-1, -2))  # batched MFMA GEMM
-        probs = causal_softmax(scores, scale=1.0 / math.sqrt(hd))
-        out = torch.matmul(probs, v)  # batched MFMA GEMM
-        out = out.transpose(1, 2).reshape(B, S, nh * hd)
+        scores = torch.matmul(q, k.transpose(-1, -2))  # [B,nkv,rep*S,S] MFMA
+        # causal pattern: row r of the rep*S rows is query position r % S —
+        # exactly the kernel's q_len=S row mapping
+        probs = causal_softmax(scores, scale=1.0 / math.sqrt(hd), q_len=S)
+        out = torch.matmul(probs, v)  # [B, nkv, rep*S, D] MFMA
+        out = out.view(B, nkv, rep, S, hd).permute(0, 3, 1, 2, 4)
+        out = out.reshape(B, S, nh * hd)
         return self.o_proj(out)
 
 

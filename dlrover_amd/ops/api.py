@@ -80,11 +80,14 @@ def rope_ref(
 
 
 def causal_softmax_ref(
-    scores: torch.Tensor, scale: float, q_offset: int = 0
+    scores: torch.Tensor, scale: float, q_offset: int = 0, q_len: int = 0
 ) -> torch.Tensor:
     s = scores.float() * scale
-    q_len, k_len = s.shape[-2], s.shape[-1]
-    qpos = torch.arange(q_len, device=s.device).unsqueeze(-1) + q_offset
+    n_rows, k_len = s.shape[-2], s.shape[-1]
+    q_len = q_len or n_rows
+    # row r is query position (r % q_len) — supports the GQA grouped layout
+    # where rep query-head blocks are folded into the row dimension
+    qpos = (torch.arange(n_rows, device=s.device) % q_len).unsqueeze(-1) + q_offset
     kpos = torch.arange(k_len, device=s.device).unsqueeze(0)
     s = s.masked_fill(kpos > qpos, float("-inf"))
     return torch.softmax(s, dim=-1).to(scores.dtype)
@@ -199,14 +202,13 @@ def build_rope_cache(
 
 class _CausalSoftmax(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, scores, scale, q_offset):
+    def forward(ctx, scores, scale, q_offset, q_len):
+        q_len = q_len or scores.shape[-2]
         if _use_hip(scores):
             probs = scores.contiguous()
-            hip_ops().causal_softmax_fwd(
-                probs, scores.shape[-2], q_offset, scale
-            )
+            hip_ops().causal_softmax_fwd(probs, q_len, q_offset, scale)
         else:
-            probs = causal_softmax_ref(scores, scale, q_offset)
+            probs = causal_softmax_ref(scores, scale, q_offset, q_len)
         ctx.save_for_backward(probs)
         ctx.scale = scale
         return probs
@@ -220,17 +222,19 @@ class _CausalSoftmax(torch.autograd.Function):
             return ds, None, None
         pf, df = probs.float(), dprobs.float()
         dot = (pf * df).sum(-1, keepdim=True)
-        return (pf * (df - dot) * ctx.scale).to(probs.dtype), None, None
+        return (pf * (df - dot) * ctx.scale).to(probs.dtype), None, None, None
 
 
 def causal_softmax(
-    scores: torch.Tensor, scale: float, q_offset: int = 0
+    scores: torch.Tensor, scale: float, q_offset: int = 0, q_len: int = 0
 ) -> torch.Tensor:
     """In one fused pass: probs = softmax(scale * scores + causal_mask).
 
+    q_len: the true sequence length when rows fold multiple query-head
+    blocks (GQA grouped layout) — row r is query position r % q_len.
     NOTE (GPU path): consumes ``scores`` in place — do not reuse it.
     """
-    return _CausalSoftmax.apply(scores, scale, q_offset)
+    return _CausalSoftmax.apply(scores, scale, q_offset, q_len)
 
 
 class _CrossEntropy(torch.autograd.Function):

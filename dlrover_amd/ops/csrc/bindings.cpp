@@ -61,6 +61,8 @@ std::tuple<at::Tensor, at::Tensor> rmsnorm_bwd(const at::Tensor& dy,
   check_bf16(dy, "dy");
   check_bf16(x, "x");
   const int hidden = (int)x.size(-1);
+  TORCH_CHECK(hidden <= 16384, "rmsnorm_bwd: hidden > 16384 unsupported "
+              "(register dw accumulator: RMSN_MAX_VPT)");
   const long long n_rows = x.numel() / hidden;
   const int n_partials = 64;
   auto dx = at::empty_like(x);

@@ -45,7 +45,15 @@ __global__ void rmsnorm_fwd_kernel(
 
 // backward:
 //   dx = inv * w * dy - x * inv^3 / H * sum_j(dy_j * w_j * x_j)
-//   dw = sum_rows(dy * x * inv)   (fp32 partials, reduced by a second kernel)
+//   dw = sum_rows(dy * x * inv)
+//
+// dw accumulates in REGISTERS across the block's grid-stride rows (thread t
+// always owns the same column set), flushed ONCE per block into fp32
+// partial slices — profiling showed the per-element global atomicAdd version
+// ran at 0.65 TB/s (profiles/r01_small1b_kernel_stats.txt); this removes
+// n_rows*H atomics down to gridDim*H.
+#define RMSN_MAX_VPT 8  // vecs/thread: supports hidden up to 8*8*256 = 16384
+
 __global__ void rmsnorm_bwd_kernel(
     const short* __restrict__ dy, const short* __restrict__ x,
     const short* __restrict__ w, const float* __restrict__ invrms,
@@ -54,6 +62,12 @@ __global__ void rmsnorm_bwd_kernel(
   __shared__ float scratch[16];
   const int vecs = hidden >> 3;
   float* dwp = dw_partial + (long long)(blockIdx.x % n_partials) * hidden;
+  float dw_acc[RMSN_MAX_VPT][8];
+#pragma unroll
+  for (int i = 0; i < RMSN_MAX_VPT; ++i)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) dw_acc[i][j] = 0.f;
+
   for (int row = blockIdx.x; row < n_rows; row += gridDim.x) {
     const short* dyr = dy + (long long)row * hidden;
     const short* xr = x + (long long)row * hidden;
@@ -71,8 +85,9 @@ __global__ void rmsnorm_bwd_kernel(
     }
     dot = block_reduce_sum(dot, scratch);
     const float k = dot * inv * inv * inv / hidden;
-    // pass 2: dx + dw partial accumulation
-    for (int v = threadIdx.x; v < vecs; v += blockDim.x) {
+    // pass 2: dx + register dw accumulation
+    int vi = 0;
+    for (int v = threadIdx.x; v < vecs; v += blockDim.x, ++vi) {
       float dyv[8], wv[8], xv[8], out[8];
       load8(dyr + v * 8, dyv);
       load8(w + v * 8, wv);
@@ -80,11 +95,17 @@ __global__ void rmsnorm_bwd_kernel(
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         out[j] = dyv[j] * wv[j] * inv - xv[j] * k;
-        atomicAdd(&dwp[v * 8 + j], dyv[j] * xv[j] * inv);
+        dw_acc[vi][j] += dyv[j] * xv[j] * inv;
       }
       store8(dxr + v * 8, out);
     }
     __syncthreads();
+  }
+  // flush: one atomicAdd per column per block into this block's partial slice
+  int vi = 0;
+  for (int v = threadIdx.x; v < vecs; v += blockDim.x, ++vi) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) atomicAdd(&dwp[v * 8 + j], dw_acc[vi][j]);
   }
 }
 
