@@ -377,6 +377,22 @@ def launch_agent(
     )
     saver.register_signal_handlers()
 
+    # hiptimer: LD_PRELOAD profiler + hang detection in workers, collector in
+    # the agent feeding the master's diagnostician (config #5)
+    collector = None
+    if os.getenv("DLROVER_HIPTIMER", "") == "1":
+        from dlrover_amd import xpu_timer
+
+        if xpu_timer.available():
+            metrics_dir = f"/tmp/hiptimer_{os.getenv('ELASTIC_JOB_NAME', 'job')}"
+            os.environ.update(
+                xpu_timer.preload_env(
+                    metrics_dir,
+                    hang_secs=float(os.getenv("DLROVER_HANG_SECS", "60")),
+                )
+            )
+            collector = xpu_timer.HiptimerCollector(metrics_dir, client).start()
+
     if config.network_check:
         from dlrover_amd.agent.node_check_agent import run_network_check
 
@@ -390,6 +406,8 @@ def launch_agent(
             raise RuntimeError(f"workers failed: {result.failures}")
         return result.return_values
     finally:
+        if collector is not None:
+            collector.stop()
         spec.rdzv_handler.shutdown()
         saver.save_shm_to_storage()
         AsyncCheckpointSaver.reset()

@@ -65,6 +65,9 @@ def parse_args(argv: Optional[List[str]] = None):
     p.add_argument("--numa-affinity", action="store_true")
     p.add_argument("--accelerator", default="amd.com/gpu")
     p.add_argument("--save-at-breakpoint", "--save_at_breakpoint", action="store_true")
+    p.add_argument("--hiptimer", action="store_true",
+                   help="LD_PRELOAD the hiptimer profiler into workers "
+                        "(kernel/GEMM/RCCL timing + hang detection)")
     p.add_argument("--checkpoint-dir", default="/tmp/dlrover_amd_ckpt")
     p.add_argument("--log-dir", default=None)
     p.add_argument("--service-type", default=CommServiceType.TCP)
@@ -166,6 +169,9 @@ def run(args) -> int:
                 args.network_check = overrides["network_check"] == "true"
     except Exception:  # noqa: BLE001
         pass
+
+    if args.hiptimer:
+        os.environ["DLROVER_HIPTIMER"] = "1"
 
     wait_pre_check(client)
 

@@ -1,0 +1,125 @@
+"""hiptimer: the MI355X profiling/hang-detection stack (xpu_timer rebuild).
+
+C++ side: csrc/hiptimer.cc -> libhiptimer.so, LD_PRELOADed into workers.
+Python side: env setup, Prometheus-text metrics parsing, and the agent-side
+collector that feeds the master's hang diagnostician.
+"""
+
+import glob
+import json
+import os
+import threading
+import time
+from typing import Dict, Optional
+
+from dlrover_amd.common.log import logger
+
+HANG_METRIC = "XPU_TIMER_COMMON_HANG"
+
+
+def library_path() -> str:
+    return os.path.join(os.path.dirname(os.path.abspath(__file__)), "libhiptimer.so")
+
+
+def available() -> bool:
+    return os.path.exists(library_path())
+
+
+def preload_env(
+    metrics_dir: str = "/tmp/hiptimer",
+    hang_secs: float = 60.0,
+    base_env: Optional[dict] = None,
+) -> dict:
+    """Env additions that enable hiptimer in child processes."""
+    env = dict(base_env) if base_env else {}
+    prev = env.get("LD_PRELOAD", os.environ.get("LD_PRELOAD", ""))
+    lib = library_path()
+    if not os.path.exists(lib):
+        logger.warning("libhiptimer.so not built — hiptimer disabled")
+        return env
+    env["LD_PRELOAD"] = f"{lib}:{prev}" if prev else lib
+    env["HIPTIMER_METRICS_DIR"] = metrics_dir
+    env["HIPTIMER_HANG_SECS"] = str(hang_secs)
+    return env
+
+
+def parse_metrics_file(path: str) -> Dict[str, float]:
+    """Parse the Prometheus-text dump (labels folded into the key)."""
+    out: Dict[str, float] = {}
+    try:
+        with open(path) as f:
+            for line in f:
+                line = line.strip()
+                if not line or line.startswith("#"):
+                    continue
+                key, _, val = line.rpartition(" ")
+                try:
+                    out[key] = float(val)
+                except ValueError:
+                    continue
+    except FileNotFoundError:
+        pass
+    return out
+
+
+class HiptimerCollector:
+    """Agent-side collector (ref: diagnosis/datacollector/
+    xpu_timer_metric_collector.py:28-80): reads every local rank's metrics
+    file and reports the hang state to the master as DiagnosisReportData."""
+
+    def __init__(self, metrics_dir: str, client=None, interval: float = 15.0):
+        self.metrics_dir = metrics_dir
+        self.interval = interval
+        self._client = client
+        self._stop = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+        self._hang_started: float = 0.0
+
+    def start(self):
+        self._thread = threading.Thread(
+            target=self._loop, name="hiptimer-collector", daemon=True
+        )
+        self._thread.start()
+        return self
+
+    def stop(self):
+        self._stop.set()
+        if self._thread is not None:
+            self._thread.join(timeout=3)
+
+    def snapshot(self) -> Dict[int, Dict[str, float]]:
+        out = {}
+        for path in glob.glob(os.path.join(self.metrics_dir, "hiptimer_*.prom")):
+            try:
+                rank = int(os.path.basename(path).split("_")[1].split(".")[0])
+            except (IndexError, ValueError):
+                continue
+            out[rank] = parse_metrics_file(path)
+        return out
+
+    def node_hang_state(self) -> dict:
+        """hang iff EVERY local rank reports hang (ref: training_hang.py:160:
+        a hang is global, one busy rank means the node is not hung)."""
+        snap = self.snapshot()
+        if not snap:
+            return {"hang": False, "since": 0.0}
+        all_hang = all(m.get(HANG_METRIC, 0) >= 1 for m in snap.values())
+        if all_hang:
+            if self._hang_started == 0.0:
+                self._hang_started = time.time()
+            sinces = [m.get("hiptimer_hang_since_seconds", 0) for m in snap.values()]
+            return {"hang": True, "since": self._hang_started, "ranks": len(snap),
+                    "device_since_monotonic": min(sinces)}
+        self._hang_started = 0.0
+        return {"hang": False, "since": 0.0, "ranks": len(snap)}
+
+    def _loop(self):
+        while not self._stop.wait(self.interval):
+            try:
+                state = self.node_hang_state()
+                if self._client is not None:
+                    self._client.report_diagnosis_data("hang", json.dumps(state))
+                if state.get("hang"):
+                    logger.warning("hiptimer: node reports GPU hang: %s", state)
+            except Exception:  # noqa: BLE001
+                logger.exception("hiptimer collector iteration failed")

@@ -115,13 +115,13 @@ class Manager {
     pool_.pop_back();
     *stop = pool_.back();
     pool_.pop_back();
-    hipEventRecord(*start, stream);
+    (void)hipEventRecord(*start, stream);
   }
 
   void record_end(hipStream_t stream, Category cat, double bytes,
                   hipEvent_t start, hipEvent_t stop) {
     if (!enabled_ || start == nullptr) return;
-    hipEventRecord(stop, stream);
+    (void)hipEventRecord(stop, stream);
     std::lock_guard<std::mutex> g(q_mu_);
     pending_.push_back({start, stop, cat, bytes, now()});
     launched_.fetch_add(1);
@@ -321,13 +321,14 @@ hipError_t hipLaunchKernel(const void* function_address, dim3 numBlocks,
   return rc;
 }
 
-hipError_t hipModuleLaunchKernel(void* f, unsigned gx, unsigned gy, unsigned gz,
-                                 unsigned bx, unsigned by, unsigned bz,
-                                 unsigned sharedMemBytes, hipStream_t stream,
-                                 void** params, void** extra) {
-  using fn_t = hipError_t (*)(void*, unsigned, unsigned, unsigned, unsigned,
-                              unsigned, unsigned, unsigned, hipStream_t, void**,
-                              void**);
+hipError_t hipModuleLaunchKernel(hipFunction_t f, unsigned gx, unsigned gy,
+                                 unsigned gz, unsigned bx, unsigned by,
+                                 unsigned bz, unsigned sharedMemBytes,
+                                 hipStream_t stream, void** params,
+                                 void** extra) {
+  using fn_t = hipError_t (*)(hipFunction_t, unsigned, unsigned, unsigned,
+                              unsigned, unsigned, unsigned, unsigned,
+                              hipStream_t, void**, void**);
   static fn_t fn = (fn_t)real("hipModuleLaunchKernel");
   hiptimer::Scoped sc(stream, CAT_KERNEL, 0);
   hipError_t rc =
@@ -336,15 +337,16 @@ hipError_t hipModuleLaunchKernel(void* f, unsigned gx, unsigned gy, unsigned gz,
   return rc;
 }
 
-hipError_t hipExtModuleLaunchKernel(void* f, unsigned gx, unsigned gy,
+hipError_t hipExtModuleLaunchKernel(hipFunction_t f, unsigned gx, unsigned gy,
                                     unsigned gz, unsigned bx, unsigned by,
                                     unsigned bz, size_t sharedMemBytes,
                                     hipStream_t stream, void** params,
                                     void** extra, hipEvent_t startEvent,
                                     hipEvent_t stopEvent, unsigned flags) {
-  using fn_t = hipError_t (*)(void*, unsigned, unsigned, unsigned, unsigned,
-                              unsigned, unsigned, size_t, hipStream_t, void**,
-                              void**, hipEvent_t, hipEvent_t, unsigned);
+  using fn_t = hipError_t (*)(hipFunction_t, unsigned, unsigned, unsigned,
+                              unsigned, unsigned, unsigned, size_t,
+                              hipStream_t, void**, void**, hipEvent_t,
+                              hipEvent_t, unsigned);
   static fn_t fn = (fn_t)real("hipExtModuleLaunchKernel");
   hiptimer::Scoped sc(stream, CAT_KERNEL, 0);
   hipError_t rc = fn(f, gx, gy, gz, bx, by, bz, sharedMemBytes, stream, params,
