@@ -68,7 +68,8 @@ def persist_shm_to_storage(
     global_rank = meta.extra.get("global_rank", event.global_rank)
     state = handler.load_state_dict()
     storage.safe_makedirs(path)
-    shard_file = os.path.join(path, f"rank_{global_rank:05d}.pt")
+    shard_name = meta.extra.get("shard_name") or f"rank_{global_rank:05d}.pt"
+    shard_file = os.path.join(path, shard_name)
     tmp = shard_file + ".tmp"
     t0 = time.perf_counter()
     torch.save(state, tmp)

@@ -176,6 +176,7 @@ class CheckpointEngine:
                     "path": path,
                     "global_rank": _global_rank(),
                     "world_size": _world_size(),
+                    "shard_name": self._shard_file_name(_global_rank()),
                 }
                 blocking = self.shm_handler.save_state_dict(
                     step, state_dict, extra=extra, block=block
