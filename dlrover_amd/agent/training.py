@@ -377,6 +377,11 @@ def launch_agent(
     )
     saver.register_signal_handlers()
 
+    from dlrover_amd.agent.monitor import ResourceMonitor, TorchTrainingMonitor
+
+    resource_monitor = ResourceMonitor(client).start()
+    training_monitor = TorchTrainingMonitor(client).start()
+
     # hiptimer: LD_PRELOAD profiler + hang detection in workers, collector in
     # the agent feeding the master's diagnostician (config #5)
     collector = None
@@ -406,6 +411,8 @@ def launch_agent(
             raise RuntimeError(f"workers failed: {result.failures}")
         return result.return_values
     finally:
+        resource_monitor.stop()
+        training_monitor.stop()
         if collector is not None:
             collector.stop()
         spec.rdzv_handler.shutdown()

@@ -1,0 +1,123 @@
+"""PodScaler + PodWatcher against the fake k8s API (the reference's test
+pattern: tests/test_utils.py mocked k8sClient)."""
+
+import time
+
+import pytest
+
+from dlrover_amd.common.constants import NodeEventType, NodeStatus, NodeType
+from dlrover_amd.common.events import AsyncExporter, EventEmitter
+from dlrover_amd.common.node import Node, NodeResource
+from dlrover_amd.master.scaler.pod_scaler import (
+    FakeK8sApi,
+    PodScaler,
+    build_pod_spec,
+)
+from dlrover_amd.master.watcher.k8s_watcher import (
+    FakeEventSource,
+    PodWatcher,
+    pod_to_node_event,
+)
+
+
+def _wait(cond, timeout=10):
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        if cond():
+            return True
+        time.sleep(0.05)
+    return False
+
+
+def test_pod_spec_has_amd_gpu_and_env():
+    node = Node(NodeType.WORKER, 3, config_resource=NodeResource(gpu_num=8))
+    spec = build_pod_spec("jobx", node, master_addr="10.0.0.1:24666")
+    limits = spec["spec"]["containers"][0]["resources"]["limits"]
+    assert limits["amd.com/gpu"] == "8"
+    env = {e["name"]: e["value"] for e in spec["spec"]["containers"][0]["env"]}
+    assert env["DLROVER_MASTER_ADDR"] == "10.0.0.1:24666"
+    assert env["HSA_ENABLE_IPC_MODE_LEGACY"] == "0"
+    assert spec["metadata"]["labels"]["elasticjob.dlrover/replica-index"] == "3"
+
+
+def test_scaler_launch_and_remove():
+    api = FakeK8sApi()
+    scaler = PodScaler("jobx", api=api)
+    try:
+        node = Node(NodeType.WORKER, 0, config_resource=NodeResource(gpu_num=8))
+        scaler.launch_node(node)
+        assert _wait(lambda: "jobx-worker-0" in api.pods)
+        scaler.remove_node(node)
+        assert "jobx-worker-0" in api.deleted
+    finally:
+        scaler.stop()
+
+
+def test_scaler_scale_to():
+    api = FakeK8sApi()
+    scaler = PodScaler("jobx", api=api)
+    try:
+        nodes = [Node(NodeType.WORKER, i, status=NodeStatus.RUNNING) for i in range(2)]
+        scaler.scale_to(4, nodes)
+        assert _wait(lambda: len(api.created) == 2)
+        # scale down removes highest ranks
+        all_nodes = nodes + [
+            Node(NodeType.WORKER, i, status=NodeStatus.RUNNING) for i in (2, 3)
+        ]
+        scaler.scale_to(2, all_nodes)
+        assert _wait(lambda: len(api.deleted) == 2)
+        assert set(api.deleted) == {"jobx-worker-3", "jobx-worker-2"}
+    finally:
+        scaler.stop()
+
+
+def _pod(idx, phase, reason=""):
+    pod = {
+        "metadata": {
+            "name": f"j-worker-{idx}",
+            "labels": {
+                "elasticjob.dlrover/name": "j",
+                "elasticjob.dlrover/replica-type": "worker",
+                "elasticjob.dlrover/replica-index": str(idx),
+                "elasticjob.dlrover/rank-index": str(idx),
+            },
+        },
+        "status": {"phase": phase, "podIP": "10.1.1.5"},
+    }
+    if reason:
+        pod["status"]["containerStatuses"] = [
+            {"state": {"terminated": {"reason": reason, "exitCode": 137}}}
+        ]
+    return pod
+
+
+def test_pod_event_translation():
+    evt = pod_to_node_event("MODIFIED", _pod(2, "Failed", reason="OOMKilled"))
+    assert evt.node.id == 2 and evt.node.status == NodeStatus.FAILED
+    assert evt.reason == "OOMKilled"
+    evt = pod_to_node_event("DELETED", _pod(1, "Running"))
+    assert evt.event_type == NodeEventType.DELETED
+
+
+def test_watcher_stream():
+    src = FakeEventSource()
+    watcher = PodWatcher("j", source=src)
+    src.push("ADDED", _pod(0, "Pending"))
+    src.push("MODIFIED", _pod(0, "Running"))
+    events = list(watcher.watch())
+    assert [e.node.status for e in events] == [NodeStatus.PENDING, NodeStatus.RUNNING]
+
+
+def test_events_exporter(tmp_path, monkeypatch):
+    monkeypatch.setenv("DLROVER_EVENT_DIR", str(tmp_path))
+    exporter = AsyncExporter(path=str(tmp_path / "ev.jsonl"))
+    em = EventEmitter("test", exporter)
+    em.instant("node_join", {"rank": 1})
+    with em.duration("rendezvous", {"round": 2}):
+        time.sleep(0.01)
+    exporter.close()
+    import json
+
+    rows = [json.loads(l) for l in open(tmp_path / "ev.jsonl")]
+    assert [r["type"] for r in rows] == ["instant", "begin", "end"]
+    assert rows[2]["duration_s"] > 0
