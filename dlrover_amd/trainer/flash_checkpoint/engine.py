@@ -368,11 +368,24 @@ class ShardedCheckpointEngine(CheckpointEngine):
         return itertools.chain(model.named_parameters(), model.named_buffers())
 
     def gather_state_dict(self, model, optimizer):
+        try:
+            from torch.distributed.tensor import DTensor
+        except ImportError:  # pragma: no cover
+            DTensor = ()
         model_sd = {}
+        tags = {}
         for name, t in self._named_tensors(model):
+            tags[name] = (
+                "shard0" if DTensor and isinstance(t, DTensor) else "replicated"
+            )
             model_sd[name] = _to_local(t.detach())
         opt_sd = _localize(optimizer.state_dict()) if optimizer is not None else {}
-        return {"model": model_sd, "optimizer": opt_sd}
+        return {
+            "model": model_sd,
+            "optimizer": opt_sd,
+            "_sharding": tags,
+            "world_size": _world_size(),
+        }
 
     def load_into(self, model, optimizer, state_dict):
         live = dict(self._named_tensors(model))

@@ -42,9 +42,32 @@ class FsdpShardCheckpointer(Checkpointer):
         return self.engine.save_to_storage(step, state_dict, path=path)
 
     def load_checkpoint(self, resume_path: str = "") -> Optional[dict]:
+        import os
+
+        world = dist.get_world_size() if dist.is_initialized() else 1
         sd = self.engine.load(resume_path)
+        if sd is not None and sd.get("world_size", world) != world:
+            sd = None  # shard is from a different world size: must reshard
         if sd is None:
-            return None
+            # UCP path: the checkpoint on disk was written at another world
+            # size (elastic scale) — reshard it (ref: UCP hook,
+            # training.py:1548)
+            from dlrover_amd.common.storage import read_tracker_step
+            from dlrover_amd.trainer.flash_checkpoint import ucp
+
+            path = resume_path
+            if not path:
+                step = read_tracker_step(self.checkpoint_dir)
+                if step < 0:
+                    return None
+                path = os.path.join(self.checkpoint_dir, str(step))
+            meta = ucp.load_resharded(self.engine, self.model, self.optimizer, path)
+            if meta is None:
+                return None
+            if dist.is_available() and dist.is_initialized():
+                dist.barrier()
+            logger.info("UCP-resharded FSDP checkpoint step=%s", meta.get("step"))
+            return meta
         if self.model is not None:
             self.engine.load_into(self.model, self.optimizer, sd)
         if dist.is_available() and dist.is_initialized():
