@@ -1,0 +1,128 @@
+"""Resource optimization + auto-scaling.
+
+Parity targets:
+  - ref master/node/job_auto_scaler.py:71-375 (AllreduceTrainingAutoScaler:
+    scale workers by alive/pending counts; OOM recovery)
+  - ref master/resource/local_optimizer.py:66-400 (PSLocalOptimizer resource
+    plans per job stage)
+  - ref master/hyperparams/simple_strategy_generator.py (ParallelConfig
+    tuning: dataloader batch size / workers from node stats)
+
+The MI355X build keeps the same shape: observe PerfMonitor + node stats ->
+produce a ResourcePlan / ParallelConfig; the scaler executes node-count
+changes, the agents pick up ParallelConfig over RPC.
+"""
+
+import threading
+from dataclasses import dataclass, field
+from typing import Dict, Optional
+
+from dlrover_amd.common import comm
+from dlrover_amd.common.constants import NodeStatus
+from dlrover_amd.common.log import logger
+from dlrover_amd.master.node.job_context import JobContext
+from dlrover_amd.master.perf_monitor import PerfMonitor
+
+
+@dataclass
+class ResourcePlan:
+    """Desired worker group size/resources (ref: comm ResourcePlan)."""
+
+    node_count: int = 0
+    node_resource: Dict[str, float] = field(default_factory=dict)
+    comment: str = ""
+
+
+class LocalResourceOptimizer:
+    """Single-job heuristics (ref: PSLocalOptimizer, reduced to the
+    allreduce/worker case that matters for GPU training)."""
+
+    def __init__(self, perf: PerfMonitor, ctx: Optional[JobContext] = None,
+                 min_nodes: int = 1, max_nodes: int = 1):
+        self.perf = perf
+        self.ctx = ctx or JobContext.singleton_instance()
+        self.min_nodes = min_nodes
+        self.max_nodes = max_nodes
+
+    def generate_plan(self) -> Optional[ResourcePlan]:
+        nodes = self.ctx.job_nodes()
+        alive = [n for n in nodes.values() if n.is_alive()]
+        pending = [n for n in alive if n.status == NodeStatus.PENDING]
+        running = [n for n in alive if n.status == NodeStatus.RUNNING]
+        # ref: AllreduceTrainingAutoScaler :276 — if pods stay pending the
+        # cluster can't satisfy us: shrink to what actually runs (respecting
+        # min_nodes) so training proceeds instead of waiting forever
+        if pending and len(running) >= self.min_nodes:
+            return ResourcePlan(
+                node_count=len(running),
+                comment=f"{len(pending)} nodes pending: shrink to running set",
+            )
+        if not pending and len(alive) < self.max_nodes:
+            return ResourcePlan(
+                node_count=self.max_nodes,
+                comment="capacity available: grow to max_nodes",
+            )
+        return None
+
+
+class JobAutoScaler:
+    """Periodic plan->execute loop (ref: job_auto_scaler.py:71)."""
+
+    def __init__(self, optimizer: LocalResourceOptimizer, scaler,
+                 interval: float = 60.0):
+        self.optimizer = optimizer
+        self.scaler = scaler
+        self.interval = interval
+        self._stop = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+
+    def start(self):
+        self._thread = threading.Thread(
+            target=self._loop, name="auto-scaler", daemon=True
+        )
+        self._thread.start()
+        return self
+
+    def stop(self):
+        self._stop.set()
+
+    def execute_once(self) -> Optional[ResourcePlan]:
+        plan = self.optimizer.generate_plan()
+        if plan is None or self.scaler is None:
+            return plan
+        nodes = list(self.optimizer.ctx.job_nodes().values())
+        logger.info("auto-scale: %s -> %s (%s)",
+                    len([n for n in nodes if n.is_alive()]),
+                    plan.node_count, plan.comment)
+        self.scaler.scale_to(plan.node_count, nodes)
+        return plan
+
+    def _loop(self):
+        while not self._stop.wait(self.interval):
+            try:
+                self.execute_once()
+            except Exception:  # noqa: BLE001
+                logger.exception("auto-scale iteration failed")
+
+
+class SimpleStrategyGenerator:
+    """Dataloader/optimizer hyperparam suggestions from node stats
+    (ref: simple_strategy_generator.py — NOT TP/PP tuning)."""
+
+    def __init__(self, perf: PerfMonitor):
+        self.perf = perf
+
+    def generate_parallel_config(self, node_id: int = 0) -> comm.ParallelConfig:
+        cfg = comm.ParallelConfig()
+        stats = self.perf.node_resource(node_id)
+        if stats is None or not stats.gpu_stats:
+            return cfg
+        # heuristic from the reference: if GPU memory is underused, suggest a
+        # larger dataloader batch (the trainer decides whether to apply it)
+        used = max((g.get("used_mb", 0) for g in stats.gpu_stats), default=0)
+        total = max((g.get("total_mb", 1) for g in stats.gpu_stats), default=1)
+        util = used / max(total, 1)
+        if util < 0.5:
+            cfg.dataloader.batch_size = 0  # 0 = "caller may double"
+            cfg.dataloader.num_workers = 4
+        return cfg

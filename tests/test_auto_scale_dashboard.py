@@ -1,0 +1,104 @@
+"""Auto-scaler plans + dashboard endpoints."""
+
+import json
+import urllib.request
+
+import pytest
+
+from dlrover_amd.common.constants import NodeStatus, NodeType
+from dlrover_amd.common.node import Node
+from dlrover_amd.master.auto_scale import (
+    JobAutoScaler,
+    LocalResourceOptimizer,
+    SimpleStrategyGenerator,
+)
+from dlrover_amd.master.dashboard import Dashboard
+from dlrover_amd.master.job_master import LocalJobMaster
+from dlrover_amd.master.node.job_context import JobContext
+from dlrover_amd.master.perf_monitor import PerfMonitor
+from dlrover_amd.master.scaler.pod_scaler import FakeK8sApi, PodScaler
+
+
+@pytest.fixture()
+def ctx():
+    JobContext._reset_for_tests()
+    yield JobContext.singleton_instance()
+    JobContext._reset_for_tests()
+
+
+def _add_node(ctx, nid, status):
+    n = Node(NodeType.WORKER, nid, status=status)
+    ctx.update_node(n)
+    return n
+
+
+def test_optimizer_shrinks_on_pending(ctx):
+    perf = PerfMonitor()
+    opt = LocalResourceOptimizer(perf, ctx, min_nodes=2, max_nodes=4)
+    for i in range(2):
+        _add_node(ctx, i, NodeStatus.RUNNING)
+    for i in (2, 3):
+        _add_node(ctx, i, NodeStatus.PENDING)
+    plan = opt.generate_plan()
+    assert plan is not None and plan.node_count == 2
+
+
+def test_optimizer_grows_to_max(ctx):
+    perf = PerfMonitor()
+    opt = LocalResourceOptimizer(perf, ctx, min_nodes=1, max_nodes=4)
+    _add_node(ctx, 0, NodeStatus.RUNNING)
+    plan = opt.generate_plan()
+    assert plan is not None and plan.node_count == 4
+
+
+def test_auto_scaler_executes_via_pod_scaler(ctx):
+    perf = PerfMonitor()
+    opt = LocalResourceOptimizer(perf, ctx, min_nodes=1, max_nodes=3)
+    _add_node(ctx, 0, NodeStatus.RUNNING)
+    api = FakeK8sApi()
+    scaler = PodScaler("j", api=api)
+    try:
+        auto = JobAutoScaler(opt, scaler)
+        plan = auto.execute_once()
+        assert plan.node_count == 3
+        import time
+
+        t0 = time.time()
+        while len(api.created) < 2 and time.time() - t0 < 10:
+            time.sleep(0.05)
+        assert len(api.created) == 2  # two new pods to reach 3
+    finally:
+        scaler.stop()
+
+
+def test_strategy_generator():
+    import dlrover_amd.common.comm as comm
+
+    perf = PerfMonitor()
+    gen = SimpleStrategyGenerator(perf)
+    cfg = gen.generate_parallel_config(0)
+    assert cfg.dataloader.num_workers == 0  # no stats: defaults
+    perf.report_resource(
+        0,
+        comm.ResourceStats(
+            node_id=0, gpu_stats=[{"index": 0, "used_mb": 1000, "total_mb": 100000}]
+        ),
+    )
+    cfg = gen.generate_parallel_config(0)
+    assert cfg.dataloader.num_workers == 4
+
+
+def test_dashboard_endpoints(ctx):
+    master = LocalJobMaster(port=0).prepare()
+    dash = Dashboard(master, port=0).start()
+    try:
+        base = f"http://127.0.0.1:{dash.port}"
+        job = json.loads(urllib.request.urlopen(f"{base}/api/job").read())
+        assert "stage" in job and "rdzv_round" in job
+        nodes = json.loads(urllib.request.urlopen(f"{base}/api/nodes").read())
+        assert nodes == []
+        page = urllib.request.urlopen(f"{base}/").read().decode()
+        assert "dlrover_amd job master" in page
+    finally:
+        dash.stop()
+        master.stop()
