@@ -1,0 +1,100 @@
+"""Failure classification + hang diagnostician behaviors."""
+
+import json
+import time
+
+import pytest
+
+from dlrover_amd.common import comm
+from dlrover_amd.common.global_context import Context
+from dlrover_amd.diagnosis.diagnosis_agent import (
+    ABORT_JOB,
+    RELAUNCH_NODE,
+    RESTART_WORKER,
+    TrainingLogCollector,
+    WorkerDiagnosisAgent,
+    classify_error,
+)
+from dlrover_amd.master.diagnosis_master import DiagnosisMaster
+from dlrover_amd.master.node.job_context import JobContext
+from dlrover_amd.master.perf_monitor import PerfMonitor
+
+
+def test_classify_software_error():
+    assert classify_error("RuntimeError: shape mismatch [2,3] vs [4]") == RESTART_WORKER
+
+
+def test_classify_hardware_error():
+    assert classify_error("Memory access fault by GPU node-2") == RELAUNCH_NODE
+    assert classify_error("RCCL WARN unhandled system error, NCCL version ...") == (
+        RELAUNCH_NODE
+    )
+    assert classify_error("HSA_STATUS_ERROR_OUT_OF_RESOURCES") == RELAUNCH_NODE
+
+
+def test_classify_fatal_user_error():
+    assert classify_error("ModuleNotFoundError: No module named 'foo'") == ABORT_JOB
+    assert classify_error("torch.OutOfMemoryError: HIP out of memory") == ABORT_JOB
+
+
+def test_restart_budget_escalates():
+    agent = WorkerDiagnosisAgent()
+    assert agent.diagnose_training_failure("ValueError: x", 0, 3) == RESTART_WORKER
+    assert agent.diagnose_training_failure("ValueError: x", 3, 3) == RELAUNCH_NODE
+
+
+def test_log_collector(tmp_path):
+    p = tmp_path / "worker.log"
+    p.write_text("\n".join(f"line{i}" for i in range(500)))
+    log = TrainingLogCollector(str(p), n_lines=100).collect()
+    assert len(log.lines) == 100 and log.lines[-1] == "line499"
+
+
+def test_hang_diagnostician_metric_path(monkeypatch):
+    JobContext._reset_for_tests()
+    Context._reset_for_tests()
+    ctx = JobContext.singleton_instance()
+    cfg = Context.singleton_instance()
+    cfg.hang_downtime = 1  # 1s for the test
+    perf = PerfMonitor()
+    dm = DiagnosisMaster(perf, ctx)
+    old = time.time() - 10
+    for nid in (0, 1):
+        dm.collect_data(
+            comm.DiagnosisReportData(
+                data_cls="hang",
+                data_content=json.dumps({"hang": True, "since": old}),
+                node_id=nid,
+            )
+        )
+    dm._check_hang()
+    action = ctx.next_action(-1)
+    assert action is not None and action.action_type == "restart_worker"
+    # one rank NOT hung -> no further action
+    dm._last_hang_action = 0
+    dm.collect_data(
+        comm.DiagnosisReportData(
+            data_cls="hang",
+            data_content=json.dumps({"hang": False, "since": 0}),
+            node_id=1,
+        )
+    )
+    dm._check_hang()
+    assert ctx.next_action(-1) is None
+    JobContext._reset_for_tests()
+    Context._reset_for_tests()
+
+
+def test_hang_diagnostician_step_path():
+    JobContext._reset_for_tests()
+    Context._reset_for_tests()
+    ctx = JobContext.singleton_instance()
+    Context.singleton_instance().hang_downtime = 1
+    perf = PerfMonitor()
+    dm = DiagnosisMaster(perf, ctx)
+    perf.report_global_step(10, time.time() - 30)  # stale progress
+    dm._check_hang()
+    action = ctx.next_action(-1)
+    assert action is not None and "hang" in action.reason
+    JobContext._reset_for_tests()
+    Context._reset_for_tests()
