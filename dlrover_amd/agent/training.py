@@ -265,7 +265,36 @@ class ElasticTrainingAgent(LocalElasticAgent):
                     "worker group %s: persisting checkpoint then deciding", state.name
                 )
                 self._save_ckpt_to_storage()
-                self._report_failures(run_result)
+                failure_text = self._report_failures(run_result)
+                # restart (software) vs relaunch (hardware) vs abort ladder
+                # (ref: diagnosis_agent.diagnose_training_failure :153)
+                from dlrover_amd.diagnosis.diagnosis_agent import (
+                    ABORT_JOB,
+                    RELAUNCH_NODE,
+                    WorkerDiagnosisAgent,
+                )
+
+                verdict = WorkerDiagnosisAgent(self.client).diagnose_training_failure(
+                    failure_text,
+                    spec.max_restarts - self._remaining_restarts,
+                    spec.max_restarts,
+                )
+                if verdict == ABORT_JOB:
+                    self._stop_workers(self._worker_group)
+                    self._worker_group.state = WorkerState.FAILED
+                    self._report_event(
+                        NodeEventType.FAILED_EXITED, f"unrecoverable: {failure_text[:200]}"
+                    )
+                    return run_result
+                if verdict == RELAUNCH_NODE:
+                    # this node is suspect: exit FAILED so the master
+                    # relaunches the pod elsewhere (ref: _relaunch_node)
+                    self._stop_workers(self._worker_group)
+                    self._worker_group.state = WorkerState.FAILED
+                    self._report_event(
+                        NodeEventType.FAILED_EXITED, "hardware error suspected"
+                    )
+                    return run_result
                 if self._remaining_restarts > 0:
                     self._remaining_restarts -= 1
                     logger.info(
@@ -299,7 +328,8 @@ class ElasticTrainingAgent(LocalElasticAgent):
 
             raise RuntimeError(f"unknown worker state {state}")
 
-    def _report_failures(self, run_result: RunResult):
+    def _report_failures(self, run_result: RunResult) -> str:
+        errs = {}
         try:
             errs = {
                 rank: f.message if hasattr(f, "message") else str(f)
@@ -313,6 +343,7 @@ class ElasticTrainingAgent(LocalElasticAgent):
             )
         except Exception:  # noqa: BLE001
             logger.warning("failure report to master failed")
+        return str(errs)
 
     def _report_event(self, event_type: str, reason: str = ""):
         try:
