@@ -24,6 +24,7 @@ void causal_softmax_bwd_launch(void*, const void*, long long, int, float,
                                void*);
 void cross_entropy_launch(void*, const void*, void*, long long, int, int,
                           float, int, void*);
+void mfma16_probe_launch(const void*, const void*, void*, void*);
 }
 
 namespace {
@@ -162,6 +163,17 @@ void causal_softmax_bwd(at::Tensor& dscores, const at::Tensor& probs,
                             row_len, (float)scale, cur_stream());
 }
 
+at::Tensor mfma16_probe(const at::Tensor& A, const at::Tensor& B) {
+  check_bf16(A, "A");
+  check_bf16(B, "B");
+  TORCH_CHECK(A.sizes() == at::IntArrayRef({16, 32}) &&
+              B.sizes() == at::IntArrayRef({32, 16}),
+              "probe wants A[16,32], B[32,16]");
+  auto C = at::empty({16, 16}, A.options().dtype(at::kFloat));
+  mfma16_probe_launch(A.data_ptr(), B.data_ptr(), C.data_ptr(), cur_stream());
+  return C;
+}
+
 at::Tensor cross_entropy_fwd_bwd(at::Tensor& logits, const at::Tensor& targets,
                                  int64_t ignore_index, double grad_scale,
                                  bool compute_grad) {
@@ -194,4 +206,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "in-place softmax backward (bf16)");
   m.def("cross_entropy_fwd_bwd", &cross_entropy_fwd_bwd,
         "fused CE loss + in-place dlogits (bf16)");
+  m.def("mfma16_probe", &mfma16_probe, "MFMA 16x16x32 bf16 layout self-test");
 }

@@ -25,6 +25,7 @@ sources = [
     os.path.join(CSRC, "adamw.hip"),
     os.path.join(CSRC, "softmax.hip"),
     os.path.join(CSRC, "cross_entropy.hip"),
+    os.path.join(CSRC, "mfma_probe.hip"),
 ]
 
 setup(
