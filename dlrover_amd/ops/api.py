@@ -219,7 +219,7 @@ class _CausalSoftmax(torch.autograd.Function):
         if _use_hip(probs):
             ds = dprobs.contiguous().clone()
             hip_ops().causal_softmax_bwd(ds, probs, ctx.scale)
-            return ds, None, None
+            return ds, None, None, None
         pf, df = probs.float(), dprobs.float()
         dot = (pf * df).sum(-1, keepdim=True)
         return (pf * (df - dot) * ctx.scale).to(probs.dtype), None, None, None

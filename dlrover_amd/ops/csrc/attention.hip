@@ -1,0 +1,258 @@
+// Flash attention (forward) for MI355X: MFMA 16x16x32 bf16, online softmax,
+// causal, GQA, head_dim=128.
+//
+// The reference has no attention kernel of its own (SURVEY.md §2.3 — it only
+// OBSERVES FlashAttention launches in xpu_timer); this is part of the
+// MI355X-native hot path. Design per the CDNA4 guide (§B fused attention):
+// never materialize the S x S score matrix; per Q-tile iterate K/V-tiles with
+// running (m, l) rescaling.
+//
+// Geometry: block = 256 threads = 4 waves; one block owns one
+// (batch, q-head, 64-row q-tile); wave w owns q rows [16w, 16w+16).
+// Per KV tile (64 rows):
+//   K staged in LDS row-major [64][128] with the guide's XOR swizzle
+//   (byte ^= (row&7)<<4) so B-fragment ds_read_b128 is conflict-light;
+//   V staged TRANSPOSED [128][64] (so the PV B-fragment reads are contiguous)
+//   with the same swizzle;
+//   S_band[16,64] = Q_band @ K^T  (16 MFMA per wave),
+//   online-softmax update, P staged through LDS (swizzled) to re-layout for
+//   the A operand, O_band[16,128] += P_band @ V (16 MFMA per wave).
+// Saves per-row logsumexp L for the backward.
+//
+// MFMA operand layouts (hardware-verified by tests/test_mfma_gpu.py):
+//   A[16,32]: lane l -> row l%16, k = (l/16)*8 + [0,8)
+//   B[32,16]: lane l -> k = (l/16)*8 + [0,8), col l%16
+//   C[16,16]: lane l, reg r -> row (l>>4)*4 + r, col l&15
+#include "kern_common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4_t;
+
+#define FA_D 128
+#define FA_QT 64  // q rows per block
+#define FA_KT 64  // kv rows per tile
+
+// XOR swizzle on a byte offset within a 128-byte LDS row group (guide G4)
+__device__ __forceinline__ int swz(int row, int byte_col) {
+  return (row * 0 + byte_col) ^ ((row & 7) << 4);
+}
+
+// wave-local reduction across the 16 lanes that share a C-row quarter
+__device__ __forceinline__ float qmax16(float v) {
+#pragma unroll
+  for (int s = 8; s > 0; s >>= 1) v = fmaxf(v, __shfl_xor(v, s, 64));
+  return v;
+}
+__device__ __forceinline__ float qsum16(float v) {
+#pragma unroll
+  for (int s = 8; s > 0; s >>= 1) v += __shfl_xor(v, s, 64);
+  return v;
+}
+
+__device__ __forceinline__ bf16x8 ld_frag_b128(const char* lds_base, int row,
+                                               int byte_col) {
+  const int off = row * 256 + swz(row, byte_col);
+  return *reinterpret_cast<const bf16x8*>(lds_base + off);
+}
+
+__global__ __launch_bounds__(256) void flash_attn_fwd_kernel(
+    const short* __restrict__ q, const short* __restrict__ k,
+    const short* __restrict__ v, short* __restrict__ out,
+    float* __restrict__ lse, int B, int H, int HKV, int S, float scale) {
+  // LDS: K [64][128] bf16 (rows padded to 256B as-is), V^T [128][64] bf16,
+  // P [4 waves][16][64] bf16 — all swizzled.
+  __shared__ char k_lds[FA_KT * FA_D * 2];        // 16 KB, row stride 256 B
+  __shared__ char vt_lds[FA_D * FA_KT * 2];       // 16 KB, row stride 128 B
+  __shared__ char p_lds[4 * 16 * FA_KT * 2];      // 8 KB, row stride 128 B
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int quarter = lane >> 4;  // 0..3
+  const int sub = lane & 15;      // 0..15
+
+  const int qt = blockIdx.x;
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int g = h / (H / HKV);  // GQA kv head
+
+  const long long q_base = (((long long)b * H + h) * S + qt * FA_QT) * FA_D;
+  const long long kv_base = (((long long)b * HKV + g) * S) * FA_D;
+
+  // ---- load this wave's Q band [16,128] into A fragments (persistent) ----
+  // frag ks (k-slice of 32): lane holds Q[qrow0 + sub][ks*32 + quarter*8 + i]
+  const int qrow_w = wave * 16;
+  bf16x8 aq[4];
+#pragma unroll
+  for (int ks = 0; ks < 4; ++ks) {
+    const short* src =
+        q + q_base + (long long)(qrow_w + sub) * FA_D + ks * 32 + quarter * 8;
+    aq[ks] = *reinterpret_cast<const bf16x8*>(src);
+  }
+
+  float m_run[4], l_run[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    m_run[r] = -1e30f;
+    l_run[r] = 0.f;
+  }
+  f32x4_t acc_o[8];
+#pragma unroll
+  for (int n = 0; n < 8; ++n) acc_o[n] = {0.f, 0.f, 0.f, 0.f};
+
+  char* p_wave = p_lds + wave * 16 * FA_KT * 2;
+
+  for (int kt = 0; kt <= qt; ++kt) {
+    // ---- stage K tile (row-major, swizzled) + V tile (transposed) ----
+    // 64*128 elems, 256 threads -> 32 elems (4x b128 chunks) per thread
+    {
+      const short* ksrc = k + kv_base + (long long)kt * FA_KT * FA_D;
+      const short* vsrc = v + kv_base + (long long)kt * FA_KT * FA_D;
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+        const int linear = (tid * 4 + c) * 8;  // 8-elem chunk start
+        const int row = linear / FA_D;
+        const int col = linear % FA_D;
+        // K: contiguous global read, swizzled b128 LDS write
+        bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(ksrc + linear);
+        *reinterpret_cast<bf16x8*>(k_lds + row * 256 + swz(row, col * 2)) = kv8;
+        // V: same read pattern, transposed scalar writes (row<->col)
+        bf16x8 vv8 = *reinterpret_cast<const bf16x8*>(vsrc + linear);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          const int trow = col + i;  // head-dim index
+          const int tcol = row;      // kv position
+          *reinterpret_cast<__bf16*>(
+              vt_lds + trow * 128 + swz(trow, tcol * 2)) = vv8[i];
+        }
+      }
+    }
+    __syncthreads();
+
+    // ---- S_band = Q_band @ K^T ----
+    f32x4_t acc_s[4];
+#pragma unroll
+    for (int n = 0; n < 4; ++n) acc_s[n] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        // B[k][col] = K[n*16 + sub][ks*32 + quarter*8 + i] (K^T fragment)
+        bf16x8 bk;
+        {
+          const int row = n * 16 + sub;
+          const int byte_col = (ks * 32 + quarter * 8) * 2;
+          bk = *reinterpret_cast<const bf16x8*>(
+              k_lds + row * 256 + swz(row, byte_col));
+        }
+        acc_s[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[ks], bk,
+                                                           acc_s[n], 0, 0, 0);
+      }
+    }
+
+    // ---- causal mask + online softmax ----
+    const int row_glob = qt * FA_QT + qrow_w + quarter * 4;  // + r
+    float pmax[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) pmax[r] = -1e30f;
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      const int col_glob = kt * FA_KT + n * 16 + sub;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float s = acc_s[n][r] * scale;
+        if (col_glob > row_glob + r) s = -1e30f;
+        acc_s[n][r] = s;
+        pmax[r] = fmaxf(pmax[r], s);
+      }
+    }
+    float alpha[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const float m_new = fmaxf(m_run[r], qmax16(pmax[r]));
+      alpha[r] = __expf(m_run[r] - m_new);
+      m_run[r] = m_new;
+    }
+    // P = exp(s - m), row sums, stage P^T-layout into LDS
+    float psum[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const float p = __expf(acc_s[n][r] - m_run[r]);
+        psum[r] += p;
+        // P row = quarter*4 + r, col = n*16 + sub
+        const int prow = quarter * 4 + r;
+        const int pcol = n * 16 + sub;
+        *reinterpret_cast<__bf16*>(p_wave + prow * 128 + swz(prow, pcol * 2)) =
+            (__bf16)p;
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      l_run[r] = l_run[r] * alpha[r] + qsum16(psum[r]);
+#pragma unroll
+      for (int n = 0; n < 8; ++n) {
+        // rescale only the rows this (quarter, r) owns — all acc rows share
+        // the lane's quarter, so alpha[r] applies to element r of every frag
+        acc_o[n][r] *= alpha[r];
+      }
+    }
+    __syncthreads();  // P visible wave-locally; K/V reuse next iter needs all
+
+    // ---- O_band += P_band @ V ----
+#pragma unroll
+    for (int n = 0; n < 8; ++n) {
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        // A fragment: P[row sub][ks*32 + quarter*8 + i]
+        bf16x8 ap;
+        {
+          const int row = sub;
+          const int byte_col = (ks * 32 + quarter * 8) * 2;
+          ap = *reinterpret_cast<const bf16x8*>(
+              p_wave + row * 128 + swz(row, byte_col));
+        }
+        // B fragment: V[k][n*16+sub] = VT[n*16+sub][k], k = ks*32+quarter*8+i
+        bf16x8 bv;
+        {
+          const int row = n * 16 + sub;
+          const int byte_col = (ks * 32 + quarter * 8) * 2;
+          bv = *reinterpret_cast<const bf16x8*>(
+              vt_lds + row * 128 + swz(row, byte_col));
+        }
+        acc_o[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, acc_o[n],
+                                                           0, 0, 0);
+      }
+    }
+    __syncthreads();  // done with this tile's K/V/P
+  }
+
+  // ---- epilogue: normalize, store O and logsumexp ----
+  const long long o_base = q_base + (long long)qrow_w * FA_D;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const float inv = l_run[r] > 0.f ? 1.f / l_run[r] : 0.f;
+    const int orow = quarter * 4 + r;
+#pragma unroll
+    for (int n = 0; n < 8; ++n) {
+      out[o_base + (long long)orow * FA_D + n * 16 + sub] =
+          f2bf(acc_o[n][r] * inv);
+    }
+    if (sub == 0) {
+      const long long lrow =
+          ((long long)b * H + h) * S + qt * FA_QT + qrow_w + orow;
+      lse[lrow] = m_run[r] + __logf(fmaxf(l_run[r], 1e-30f));
+    }
+  }
+}
+
+extern "C" void flash_attn_fwd_launch(const void* q, const void* k,
+                                      const void* v, void* out, void* lse,
+                                      int B, int H, int HKV, int S,
+                                      float scale, hipStream_t stream) {
+  dim3 grid(S / FA_QT, H, B);
+  hipLaunchKernelGGL(flash_attn_fwd_kernel, grid, dim3(256), 0, stream,
+                     (const short*)q, (const short*)k, (const short*)v,
+                     (short*)out, (float*)lse, B, H, HKV, S, scale);
+}

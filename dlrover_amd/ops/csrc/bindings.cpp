@@ -25,6 +25,8 @@ void causal_softmax_bwd_launch(void*, const void*, long long, int, float,
 void cross_entropy_launch(void*, const void*, void*, long long, int, int,
                           float, int, void*);
 void mfma16_probe_launch(const void*, const void*, void*, void*);
+void flash_attn_fwd_launch(const void*, const void*, const void*, void*,
+                           void*, int, int, int, int, float, void*);
 }
 
 namespace {
@@ -163,6 +165,27 @@ void causal_softmax_bwd(at::Tensor& dscores, const at::Tensor& probs,
                             row_len, (float)scale, cur_stream());
 }
 
+std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
+                                                  const at::Tensor& k,
+                                                  const at::Tensor& v,
+                                                  double scale) {
+  check_bf16(q, "q");
+  check_bf16(k, "k");
+  check_bf16(v, "v");
+  TORCH_CHECK(q.dim() == 4 && k.dim() == 4, "q/k/v must be [B,H,S,D]");
+  const int B = (int)q.size(0), H = (int)q.size(1), S = (int)q.size(2);
+  const int D = (int)q.size(3), HKV = (int)k.size(1);
+  TORCH_CHECK(D == 128, "flash_attn: head_dim must be 128");
+  TORCH_CHECK(S % 64 == 0, "flash_attn: seq len must be a multiple of 64");
+  TORCH_CHECK(H % HKV == 0, "flash_attn: H % HKV != 0");
+  auto out = at::empty_like(q);
+  auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
+  flash_attn_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                        out.data_ptr(), lse.data_ptr(), B, H, HKV, S,
+                        (float)scale, cur_stream());
+  return {out, lse};
+}
+
 at::Tensor mfma16_probe(const at::Tensor& A, const at::Tensor& B) {
   check_bf16(A, "A");
   check_bf16(B, "B");
@@ -207,4 +230,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_fwd_bwd", &cross_entropy_fwd_bwd,
         "fused CE loss + in-place dlogits (bf16)");
   m.def("mfma16_probe", &mfma16_probe, "MFMA 16x16x32 bf16 layout self-test");
+  m.def("flash_attn_fwd", &flash_attn_fwd,
+        "flash attention forward (bf16, causal, GQA, D=128) -> (out, lse)");
 }

@@ -26,6 +26,7 @@ sources = [
     os.path.join(CSRC, "softmax.hip"),
     os.path.join(CSRC, "cross_entropy.hip"),
     os.path.join(CSRC, "mfma_probe.hip"),
+    os.path.join(CSRC, "attention.hip"),
 ]
 
 setup(
