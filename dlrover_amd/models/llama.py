@@ -40,6 +40,9 @@ class LlamaConfig:
     rope_base: float = 500000.0
     norm_eps: float = 1e-5
     tie_embeddings: bool = False
+    # "auto": hand-written flash kernel on GPU when D==128 and S%64==0,
+    # composite (hipBLASLt GEMM + fused softmax) otherwise
+    attn_impl: str = "auto"
 
     @property
     def head_dim(self) -> int:
@@ -107,6 +110,23 @@ class Attention(nn.Module):
         v = v.view(B, S, nkv, hd)
         q = rope_rotate(q, pos, cos, sin)
         k = rope_rotate(k, pos, cos, sin)
+        use_flash = (
+            cfg.attn_impl in ("auto", "flash")
+            and x.is_cuda
+            and hd == 128
+            and S % 64 == 0
+        )
+        if use_flash:
+            from dlrover_amd.ops import flash_attention
+
+            out = flash_attention(
+                q.transpose(1, 2).contiguous(),
+                k.transpose(1, 2).contiguous(),
+                v.transpose(1, 2).contiguous(),
+                1.0 / math.sqrt(hd),
+            )
+            out = out.transpose(1, 2).reshape(B, S, nh * hd)
+            return self.o_proj(out)
         # grouped-query attention with ZERO kv copies: fold the query heads
         # of each kv group into the row dimension — q [B, nkv, rep*S, D]
         # against k/v [B, nkv, S, D]. (repeat_interleave showed up as 11% of

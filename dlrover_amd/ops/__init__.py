@@ -12,6 +12,7 @@ Loading policy (the framework's hot path must actually run native code):
 from dlrover_amd.ops.api import (  # noqa: F401
     ExtensionMissingError,
     causal_softmax,
+    flash_attention,
     cross_entropy_loss,
     fused_adamw_step,
     hip_ops,

@@ -237,6 +237,62 @@ def causal_softmax(
     return _CausalSoftmax.apply(scores, scale, q_offset, q_len)
 
 
+def flash_attention_ref(
+    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, scale: float
+) -> torch.Tensor:
+    """fp32 composite reference: q [B,H,S,D], k/v [B,HKV,S,D], causal."""
+    B, H, S, D = q.shape
+    rep = H // k.shape[1]
+    kf = k.float().repeat_interleave(rep, 1)
+    vf = v.float().repeat_interleave(rep, 1)
+    s_ = torch.matmul(q.float(), kf.transpose(-1, -2)) * scale
+    mask = torch.triu(torch.ones(S, S, dtype=torch.bool, device=q.device), 1)
+    s_ = s_.masked_fill(mask, float("-inf"))
+    p = torch.softmax(s_, -1)
+    return torch.matmul(p, vf).to(q.dtype)
+
+
+class _FlashAttention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        if _use_hip(q):
+            q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+            out, lse = hip_ops().flash_attn_fwd(q, k, v, scale)
+            ctx.save_for_backward(q, k, v, out, lse)
+            ctx.scale = scale
+            ctx.use_hip = True
+            return out
+        ctx.use_hip = False
+        q32 = q.detach().clone().requires_grad_(True)
+        k32 = k.detach().clone().requires_grad_(True)
+        v32 = v.detach().clone().requires_grad_(True)
+        with torch.enable_grad():
+            out = flash_attention_ref(q32, k32, v32, scale)
+        ctx.saved_ref = (q32, k32, v32, out)
+        return out.detach()
+
+    @staticmethod
+    def backward(ctx, dout):
+        if ctx.use_hip:
+            q, k, v, out, lse = ctx.saved_tensors
+            dq, dk, dv = hip_ops().flash_attn_bwd(
+                q, k, v, out, dout.contiguous(), lse, ctx.scale
+            )
+            return dq, dk, dv, None
+        q32, k32, v32, out = ctx.saved_ref
+        torch.autograd.backward(out, dout)
+        return q32.grad, k32.grad, v32.grad, None
+
+
+def flash_attention(
+    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, scale: float
+) -> torch.Tensor:
+    """Fused causal GQA attention. q [B,H,S,D], k/v [B,HKV,S,D], D=128,
+    S % 64 == 0. Never materializes the S^2 score matrix (hand-written MFMA
+    kernels, ops/csrc/attention*.hip)."""
+    return _FlashAttention.apply(q, k, v, scale)
+
+
 class _CrossEntropy(torch.autograd.Function):
     @staticmethod
     def forward(ctx, logits, targets, ignore_index):

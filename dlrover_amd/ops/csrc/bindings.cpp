@@ -27,6 +27,14 @@ void cross_entropy_launch(void*, const void*, void*, long long, int, int,
 void mfma16_probe_launch(const void*, const void*, void*, void*);
 void flash_attn_fwd_launch(const void*, const void*, const void*, void*,
                            void*, int, int, int, int, float, void*);
+void fa_bwd_pre_launch(const void*, const void*, void*, long long, void*);
+void fa_bwd_dq_launch(const void*, const void*, const void*, const void*,
+                      const void*, const void*, void*, int, int, int, int,
+                      float, void*);
+void fa_bwd_dkv_launch(const void*, const void*, const void*, const void*,
+                       const void*, const void*, void*, void*, int, int, int,
+                       int, float, void*);
+void f32_to_bf16_launch(const void*, void*, long long, void*);
 }
 
 namespace {
@@ -186,6 +194,34 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
   return {out, lse};
 }
 
+std::tuple<at::Tensor, at::Tensor, at::Tensor> flash_attn_bwd(
+    const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
+    const at::Tensor& out, const at::Tensor& dout, const at::Tensor& lse,
+    double scale) {
+  check_bf16(q, "q");
+  check_bf16(dout, "dout");
+  const int B = (int)q.size(0), H = (int)q.size(1), S = (int)q.size(2);
+  const int HKV = (int)k.size(1);
+  const long long n_rows = (long long)B * H * S;
+  auto dvec = at::empty({B, H, S}, q.options().dtype(at::kFloat));
+  fa_bwd_pre_launch(dout.data_ptr(), out.data_ptr(), dvec.data_ptr(), n_rows,
+                    cur_stream());
+  auto dq = at::empty_like(q);
+  fa_bwd_dq_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), dout.data_ptr(),
+                   lse.data_ptr(), dvec.data_ptr(), dq.data_ptr(), B, H, HKV,
+                   S, (float)scale, cur_stream());
+  auto dk32 = at::zeros(k.sizes(), k.options().dtype(at::kFloat));
+  auto dv32 = at::zeros(v.sizes(), v.options().dtype(at::kFloat));
+  fa_bwd_dkv_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), dout.data_ptr(),
+                    lse.data_ptr(), dvec.data_ptr(), dk32.data_ptr(),
+                    dv32.data_ptr(), B, H, HKV, S, (float)scale, cur_stream());
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  f32_to_bf16_launch(dk32.data_ptr(), dk.data_ptr(), dk32.numel(), cur_stream());
+  f32_to_bf16_launch(dv32.data_ptr(), dv.data_ptr(), dv32.numel(), cur_stream());
+  return {dq, dk, dv};
+}
+
 at::Tensor mfma16_probe(const at::Tensor& A, const at::Tensor& B) {
   check_bf16(A, "A");
   check_bf16(B, "B");
@@ -232,4 +268,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma16_probe", &mfma16_probe, "MFMA 16x16x32 bf16 layout self-test");
   m.def("flash_attn_fwd", &flash_attn_fwd,
         "flash attention forward (bf16, causal, GQA, D=128) -> (out, lse)");
+  m.def("flash_attn_bwd", &flash_attn_bwd,
+        "flash attention backward -> (dq, dk, dv)");
 }

@@ -60,6 +60,10 @@ static void* real(const char* name) {
   auto it = cache.find(name);
   if (it != cache.end()) return it->second;
   void* fn = dlsym(RTLD_NEXT, name);
+  if (getenv("HIPTIMER_DEBUG")) {
+    fprintf(stderr, "[hiptimer] resolve %s -> %p\n", name, fn);
+    fflush(stderr);
+  }
   cache[name] = fn;
   return fn;
 }
@@ -102,7 +106,7 @@ class Manager {
                     hipEvent_t* start, hipEvent_t* stop) {
     *start = nullptr;
     *stop = nullptr;
-    if (!enabled_) return;
+    if (!enabled_ || no_events_) return;
     std::lock_guard<std::mutex> g(pool_mu_);
     if (pool_.size() < 2) {
       for (int i = 0; i < 16; ++i) {
@@ -141,6 +145,12 @@ class Manager {
   Manager() {
     const char* dis = getenv("HIPTIMER_DISABLE");
     enabled_ = !(dis && dis[0] == '1');
+    const char* noev = getenv("HIPTIMER_NO_EVENTS");
+    no_events_ = noev && noev[0] == '1';
+    const char* nopoll = getenv("HIPTIMER_NO_POLLER");
+    no_poller_ = nopoll && nopoll[0] == '1';
+    const char* dbg = getenv("HIPTIMER_DEBUG");
+    debug_ = dbg && dbg[0] == '1';
     hang_secs_ = getenv("HIPTIMER_HANG_SECS")
                      ? atof(getenv("HIPTIMER_HANG_SECS"))
                      : 60.0;
@@ -155,7 +165,8 @@ class Manager {
     snprintf(cmd, sizeof(cmd), "mkdir -p %s", metrics_dir_.c_str());
     if (system(cmd) != 0) enabled_ = false;
     last_completion_ = now();
-    if (enabled_) pthread_create(&poller_, nullptr, &Manager::poll_entry, this);
+    if (enabled_ && !no_poller_)
+      pthread_create(&poller_, nullptr, &Manager::poll_entry, this);
   }
 
   static void* poll_entry(void* self) {
@@ -267,6 +278,9 @@ class Manager {
   }
 
   bool enabled_ = false;
+  bool no_events_ = false;
+  bool no_poller_ = false;
+  bool debug_ = false;
   double hang_secs_ = 60.0;
   double dump_interval_ = 5.0;
   std::string metrics_dir_;
