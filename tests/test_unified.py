@@ -1,6 +1,7 @@
 """Unified MPMD API: builder validation, local execution, role failover."""
 
 import os
+import uuid
 
 import pytest
 
@@ -35,7 +36,7 @@ def test_builder_validates():
 
 @pytest.mark.timeout(120)
 def test_two_role_job_runs(tmp_path):
-    tag = os.path.basename(str(tmp_path))
+    tag = uuid.uuid4().hex[:8]
     job = (
         DLJobBuilder(f"j{tag}")
         .train(total=2).run(_ok_worker, tag)
@@ -51,7 +52,7 @@ def test_two_role_job_runs(tmp_path):
 
 @pytest.mark.timeout(120)
 def test_role_failover(tmp_path):
-    tag = os.path.basename(str(tmp_path))
+    tag = uuid.uuid4().hex[:8]
     job = (
         DLJobBuilder(f"f{tag}")
         .train(total=2).run(_flaky_worker, tag).max_restarts(2)
