@@ -157,10 +157,29 @@ class FusedAdamW(torch.optim.Optimizer):
         for sid, p in zip(saved_ids, params):
             if sid in state_dict["state"]:
                 dev = _local(p).device
-                new_state[p] = {
-                    k: v.to(dev) if torch.is_tensor(v) else v
-                    for k, v in state_dict["state"][sid].items()
-                }
+                cur = self.state.get(p, {})
+                entry = {}
+                for k, v in state_dict["state"][sid].items():
+                    if torch.is_tensor(v):
+                        # copy INTO existing state storage when shapes match:
+                        # a fresh .to(dev) would double-allocate ~96 GB of
+                        # fp32 optimizer state mid-restore (OOM on 8B @ N=1)
+                        old = cur.get(k)
+                        if (
+                            torch.is_tensor(old)
+                            and old.shape == v.shape
+                            and old.dtype == v.dtype
+                            and old.device == dev
+                        ):
+                            old.copy_(v, non_blocking=True)
+                            entry[k] = old
+                        else:
+                            entry[k] = v.to(dev)
+                    else:
+                        entry[k] = v
+                new_state[p] = entry
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
         self.state.clear()
         self.state.update(new_state)
         for g, sg in zip(groups, saved_groups):

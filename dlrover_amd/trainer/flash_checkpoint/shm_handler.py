@@ -370,7 +370,9 @@ class _DeviceStager:
         if self._buf is not None and self._buf.numel() >= payload and not self._chunked:
             return
         free, _total = torch.cuda.mem_get_info()
-        if payload + (2 << 30) < free:
+        # generous headroom: restore/rescale paths need transient allocations;
+        # a staging buffer that "just fits" starves them (8B @ N=1 OOM'd)
+        if payload + (24 << 30) < free:
             self._chunked = False
             self._buf = torch.empty(payload, dtype=torch.uint8, device="cuda")
         else:

@@ -53,6 +53,17 @@ namespace hiptimer {
 
 using real_fn_t = void*;
 
+// Map an interposed symbol to the library that actually provides it. Needed
+// because torch dlopen()s its backend libs WITHOUT RTLD_GLOBAL, so
+// dlsym(RTLD_NEXT, ...) from this preloaded lib cannot see them — calling the
+// nil result was a segfault. dlsym on an explicit dlopen handle bypasses the
+// search-order problem (and cannot find OUR interposer, unlike RTLD_DEFAULT).
+static const char* provider_of(const char* name) {
+  if (strncmp(name, "nccl", 4) == 0) return "librccl.so";
+  if (strncmp(name, "hipblasLt", 9) == 0) return "libhipblaslt.so";
+  return "libamdhip64.so";
+}
+
 static void* real(const char* name) {
   static std::mutex m;
   static std::map<std::string, void*> cache;
@@ -60,6 +71,13 @@ static void* real(const char* name) {
   auto it = cache.find(name);
   if (it != cache.end()) return it->second;
   void* fn = dlsym(RTLD_NEXT, name);
+  if (fn == nullptr) {
+    static std::map<std::string, void*> handles;
+    const char* lib = provider_of(name);
+    void*& h = handles[lib];
+    if (h == nullptr) h = dlopen(lib, RTLD_LAZY | RTLD_LOCAL);
+    if (h != nullptr) fn = dlsym(h, name);
+  }
   if (getenv("HIPTIMER_DEBUG")) {
     fprintf(stderr, "[hiptimer] resolve %s -> %p\n", name, fn);
     fflush(stderr);
