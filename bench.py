@@ -115,14 +115,19 @@ def main():
     if not args.no_ckpt:
         cp = FsdpShardCheckpointer(ckpt_dir, model, opt)
 
-    ids = torch.randint(0, cfg.vocab_size, (args.batch, args.seq), device=device)
-    labels = torch.randint(0, cfg.vocab_size, (args.batch, args.seq), device=device)
+    def fresh_batch():
+        # new synthetic tokens every step: a fixed batch lets an 8B model
+        # memorize to ~0 loss within the bench window, which distorts timing
+        ids = torch.randint(0, cfg.vocab_size, (args.batch, args.seq), device=device)
+        labels = torch.randint(0, cfg.vocab_size, (args.batch, args.seq), device=device)
+        return ids, labels
 
     def sync():
         if on_gpu:
             torch.cuda.synchronize()
 
     def train_step():
+        ids, labels = fresh_batch()
         loss = model(ids, labels)
         loss.backward()
         opt.step()

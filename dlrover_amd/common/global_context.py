@@ -34,8 +34,12 @@ class Context:
         self.job_name = os.getenv("ELASTIC_JOB_NAME", "dlrover-job")
         self.relaunch_error_max = DefaultValues.RELAUNCH_ERROR_MAX
         self.seconds_to_wait_pending = DefaultValues.SECONDS_TO_WAIT_PENDING
-        self.heartbeat_timeout = DefaultValues.SECONDS_HEARTBEAT_TIMEOUT
-        self.hang_downtime = DefaultValues.HANG_DOWNTIME_SECS
+        self.heartbeat_timeout = float(
+            os.getenv("DLROVER_HEARTBEAT_TIMEOUT", DefaultValues.SECONDS_HEARTBEAT_TIMEOUT)
+        )
+        self.hang_downtime = float(
+            os.getenv("DLROVER_HANG_DOWNTIME", DefaultValues.HANG_DOWNTIME_SECS)
+        )
         self.seconds_interval_collect = DefaultValues.SECONDS_INTERVAL_COLLECT
         self.auto_worker_enabled = False
         self.auto_ps_enabled = False

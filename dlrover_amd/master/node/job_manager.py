@@ -156,7 +156,28 @@ class JobManager:
     def _handle_node_failure(self, node: Node, reason: str = ""):
         for mgr in self.rdzv_managers.values():
             mgr.remove_alive_node(node.id)
+        if node.exit_reason == NodeExitReason.NO_HEARTBEAT:
+            # the node's AGENT is gone (scale-down, preemption, host death):
+            # nobody is listening for a restart action. Shrink the job — the
+            # survivors re-rendezvous; abort only if nobody is left.
+            node.relaunchable = False
+            self.ctx.update_node(node)
+            if not self.ctx.alive_nodes():
+                logger.error("all nodes gone: stopping job")
+                self.ctx.request_stop(JobExitReason.WORKER_ERROR, code=1)
+            return
         if node.is_unrecoverable_failure():
+            others = [n for n in self.ctx.alive_nodes() if n.id != node.id]
+            if others:
+                # eliminate just this node; the rest of the job continues
+                logger.error(
+                    "node %s unrecoverable (%s): eliminating it, job continues "
+                    "with %s nodes", node, reason, len(others),
+                )
+                node.eliminated = True
+                node.relaunchable = False
+                self.ctx.update_node(node)
+                return
             logger.error("node %s unrecoverable (%s): aborting job", node, reason)
             self.ctx.enqueue_action(
                 JobAbortAction(node_id=-1, reason=f"node {node.id}: {reason}")

@@ -38,7 +38,14 @@ def main():
     local_rank = int(os.getenv("LOCAL_RANK", "0"))
     if use_gpu:
         torch.cuda.set_device(local_rank)
-    dist.init_process_group(backend="nccl" if use_gpu else "gloo")
+    # short collective timeout so a dead peer surfaces as a worker failure
+    # quickly (elastic scale-down path); DLROVER_PG_TIMEOUT in seconds
+    from datetime import timedelta
+
+    pg_timeout = timedelta(seconds=int(os.getenv("DLROVER_PG_TIMEOUT", "1800")))
+    dist.init_process_group(
+        backend="nccl" if use_gpu else "gloo", timeout=pg_timeout
+    )
     rank = dist.get_rank()
     device = torch.device(f"cuda:{local_rank}" if use_gpu else "cpu")
 
@@ -79,7 +86,8 @@ def main():
                 f.write(
                     json.dumps(
                         {"step": step, "loss": round(loss.item(), 4),
-                         "incarnation": incarnation, "resumed_from": start_step}
+                         "incarnation": incarnation, "resumed_from": start_step,
+                         "world": dist.get_world_size()}
                     )
                     + "\n"
                 )

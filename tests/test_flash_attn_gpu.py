@@ -83,3 +83,29 @@ def test_flash_attn_in_model_trains():
         first = first or loss.item()
         last = loss.item()
     assert last < first * 0.9, (first, last)
+
+
+def test_model_grads_flash_vs_composite():
+    """Full tiny-model (D=128) step: flash and composite attention must give
+    the same loss and parameter gradients."""
+    from dlrover_amd.models import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(0)
+    base = dict(
+        vocab_size=512, hidden_size=512, intermediate_size=1024, n_layers=2,
+        n_heads=4, n_kv_heads=2, max_seq_len=256, rope_base=10000.0,
+    )
+    m1 = LlamaForCausalLM(LlamaConfig(**base, attn_impl="flash")).cuda().bfloat16()
+    m2 = LlamaForCausalLM(LlamaConfig(**base, attn_impl="composite")).cuda().bfloat16()
+    m2.load_state_dict(m1.state_dict())
+    ids = torch.randint(0, 512, (2, 128), device="cuda")
+    l1 = m1(ids, ids.clone())
+    l1.backward()
+    l2 = m2(ids, ids.clone())
+    l2.backward()
+    torch.testing.assert_close(l1, l2, rtol=2e-2, atol=2e-2)
+    for (n1, p1), (n2, p2) in zip(m1.named_parameters(), m2.named_parameters()):
+        g1, g2 = p1.grad.float(), p2.grad.float()
+        denom = g2.norm().clamp_min(1e-6)
+        rel = (g1 - g2).norm() / denom
+        assert rel < 0.05, (n1, rel.item())
