@@ -311,8 +311,23 @@ class ElasticTrainingAgent(LocalElasticAgent):
             if state == WorkerState.HEALTHY:
                 if self._restart_requested.is_set():
                     self._restart_requested.clear()
-                    logger.info("restart requested (diagnosis): restarting workers")
                     self._save_ckpt_to_storage()
+                    # a diagnosis-driven restart is a real failure recovery:
+                    # it consumes the restart budget (and bumps
+                    # TORCHELASTIC_RESTART_COUNT for the workers)
+                    if self._remaining_restarts <= 0:
+                        logger.error("restart requested but budget exhausted")
+                        self._stop_workers(self._worker_group)
+                        self._worker_group.state = WorkerState.FAILED
+                        self._report_event(
+                            NodeEventType.FAILED_EXITED, "restarts exhausted"
+                        )
+                        return run_result
+                    self._remaining_restarts -= 1
+                    logger.info(
+                        "restart requested (diagnosis): restarting workers "
+                        "(%s restarts left)", self._remaining_restarts,
+                    )
                     self._restart_workers(self._worker_group)
                     continue
                 # membership change: another node joined/waits for a round

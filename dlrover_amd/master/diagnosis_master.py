@@ -112,8 +112,9 @@ class DiagnosisMaster:
         if not (self._is_step_hang() or self._is_metric_hang()):
             return
         now = time.time()
-        if now - self._last_hang_action < self._config.hang_downtime:
-            return  # already acted on this hang; give recovery time
+        # give the restart time to re-rendezvous AND produce a first step
+        if now - self._last_hang_action < max(3 * self._config.hang_downtime, 60):
+            return
         self._last_hang_action = now
         self._hang_restarts += 1
         if self._hang_restarts > 2:
@@ -132,3 +133,5 @@ class DiagnosisMaster:
                 reason="training hang",
             )
         )
+        # restarting IS activity: don't re-diagnose the same stall
+        self.perf.mark_activity()

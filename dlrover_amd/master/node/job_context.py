@@ -60,6 +60,19 @@ class JobContext:
 
     def enqueue_action(self, action: DiagnosisAction):
         with self._mutex:
+            if action.node_id == -1:
+                # broadcast: fan out one copy per alive node so EVERY agent
+                # receives it on its own heartbeat (a single -1 queue would be
+                # consumed by whichever agent polls first)
+                import copy as _copy
+
+                targets = [n.id for t in self._nodes.values() for n in t.values()
+                           if n.is_alive()]
+                for nid in targets or [-1]:
+                    a = _copy.copy(action)
+                    a.node_id = nid
+                    self._actions.setdefault(nid, []).append(a)
+                return
             self._actions.setdefault(action.node_id, []).append(action)
 
     def next_action(self, node_id: int) -> Optional[DiagnosisAction]:

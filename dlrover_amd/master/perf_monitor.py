@@ -46,6 +46,12 @@ class PerfMonitor:
         with self._lock:
             return self._last_step
 
+    def mark_activity(self):
+        """Reset the progress timer (e.g. when a recovery action was just
+        issued — the job is legitimately not stepping while it restarts)."""
+        with self._lock:
+            self._last_step_time = time.time()
+
     def seconds_since_last_step(self) -> float:
         with self._lock:
             if self._last_step_time == 0.0:

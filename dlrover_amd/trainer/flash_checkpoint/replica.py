@@ -1,0 +1,152 @@
+"""In-memory checkpoint replicas: back up each rank's shm snapshot on a peer
+rank so a relaunched node restores from RAM instead of storage.
+
+Parity target: ref dlrover/trainer/torch/flash_checkpoint/replica.py:28-352
+(FullCkptReplicaManager: gloo all_gather of shm bytes within backup groups
+:149-169; locate-owner + broadcast restore after relaunch :303-350).
+
+Mechanics here: ranks form pairs (r, r^1) over a gloo group. backup() ships
+the RAW shm segment bytes (commit word + meta + payload — the segment layout
+is self-describing) to the partner, which stores them in its own
+``<segment>_backup`` shm. gather() is the inverse: a rank whose segment is
+empty (fresh pod) asks the world who holds its backup and receives the bytes
+by broadcast, then writes them into its own segment — after which the normal
+shm-first load path works untouched.
+"""
+
+import struct
+from typing import Optional
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+from dlrover_amd.common.log import logger
+from dlrover_amd.common.multi_process import (
+    attach_shared_memory,
+    create_shared_memory,
+)
+from dlrover_amd.trainer.flash_checkpoint.shm_handler import SharedMemoryHandler
+
+
+def backup_peer(rank: int, world: int) -> int:
+    peer = rank ^ 1
+    return peer if peer < world else rank
+
+
+class ReplicaManager:
+    def __init__(self, shm_handler: SharedMemoryHandler, group=None):
+        self.handler = shm_handler
+        self.group = group  # gloo group (collectives carry host bytes)
+        self._backup_shm = None
+
+    # -- helpers ---------------------------------------------------------------
+
+    def _segment_bytes(self) -> Optional[bytes]:
+        """Raw self-describing segment prefix (header + meta + payload)."""
+        meta = self.handler.read_meta()
+        if meta is None:
+            return None
+        used = self.handler._payload_offset() + meta.payload_bytes
+        return bytes(self.handler._shm.buf[:used])
+
+    def _write_segment(self, raw: bytes):
+        # ensure sized, then splat the raw bytes; the commit word rides along
+        payload = len(raw) - self.handler._payload_offset()
+        self.handler.ensure_size(max(payload, 1))
+        self.handler._shm.buf[: len(raw)] = raw
+
+    def _backup_name(self) -> str:
+        return f"{self.handler.name}_backup"
+
+    # -- backup after save (ref: replica.py:149) ---------------------------------
+
+    def backup(self) -> bool:
+        if not (dist.is_available() and dist.is_initialized()):
+            return False
+        rank = dist.get_rank(self.group)
+        world = dist.get_world_size(self.group)
+        if world < 2:
+            return False
+        peer = backup_peer(rank, world)
+        raw = self._segment_bytes() or b""
+        # exchange sizes, then bytes, within the pair (gloo all_gather)
+        sizes = [torch.zeros(1, dtype=torch.long) for _ in range(world)]
+        dist.all_gather(sizes, torch.tensor([len(raw)], dtype=torch.long),
+                        group=self.group)
+        max_size = int(max(s.item() for s in sizes))
+        if max_size == 0:
+            return False
+        buf = torch.zeros(max_size, dtype=torch.uint8)
+        if raw:
+            buf[: len(raw)] = torch.frombuffer(bytearray(raw), dtype=torch.uint8)
+        gathered = [torch.zeros(max_size, dtype=torch.uint8) for _ in range(world)]
+        dist.all_gather(gathered, buf, group=self.group)
+        peer_size = int(sizes[peer].item())
+        if peer == rank or peer_size == 0:
+            return False
+        peer_raw = gathered[peer][:peer_size]
+        shm = create_shared_memory(self._backup_name(), peer_size + 8)
+        struct.pack_into("<q", shm.buf, 0, peer_size)
+        shm.buf[8 : 8 + peer_size] = peer_raw.numpy().tobytes()
+        self._backup_shm = shm
+        logger.info(
+            "replica: stored rank %s's snapshot (%.1f MB) as local backup",
+            peer, peer_size / (1 << 20),
+        )
+        return True
+
+    def _local_backup_bytes(self) -> Optional[bytes]:
+        shm = self._backup_shm or attach_shared_memory(self._backup_name())
+        if shm is None:
+            return None
+        (size,) = struct.unpack_from("<q", shm.buf, 0)
+        if size <= 0:
+            return None
+        return bytes(shm.buf[8 : 8 + size])
+
+    # -- restore after relaunch (ref: replica.py:303) ------------------------------
+
+    def gather(self) -> bool:
+        """If this rank's segment is empty, fetch its backup from whichever
+        peer holds it. Collective: ALL ranks must call this together."""
+        if not (dist.is_available() and dist.is_initialized()):
+            return False
+        rank = dist.get_rank(self.group)
+        world = dist.get_world_size(self.group)
+        if world < 2:
+            return False
+        have_own = self.handler.committed_step() > 0
+        # whose backup do I hold, and is it valid?
+        peer = backup_peer(rank, world)
+        backup = self._local_backup_bytes()
+        holdings = [torch.zeros(2, dtype=torch.long) for _ in range(world)]
+        mine = torch.tensor(
+            [peer if backup else -1, len(backup) if backup else 0],
+            dtype=torch.long,
+        )
+        dist.all_gather(holdings, mine, group=self.group)
+        restored = False
+        for owner_rank in range(world):
+            backed_for, size = int(holdings[owner_rank][0]), int(holdings[owner_rank][1])
+            if backed_for < 0 or size == 0:
+                continue
+            # does the target rank actually need it? (its shm is empty)
+            need = torch.tensor(
+                [0 if (backed_for != rank or have_own) else 1], dtype=torch.long
+            )
+            dist.all_reduce(need, group=self.group)
+            if int(need.item()) == 0:
+                continue
+            buf = torch.zeros(size, dtype=torch.uint8)
+            if owner_rank == rank and backup:
+                buf[:] = torch.frombuffer(bytearray(backup), dtype=torch.uint8)
+            dist.broadcast(buf, src=owner_rank, group=self.group)
+            if backed_for == rank and not have_own:
+                self._write_segment(buf.numpy().tobytes())
+                restored = True
+                logger.info(
+                    "replica: restored my snapshot from rank %s (%.1f MB)",
+                    owner_rank, size / (1 << 20),
+                )
+        return restored

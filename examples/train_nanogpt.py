@@ -21,6 +21,7 @@ import torch.distributed as dist
 from torch.nn.parallel import DistributedDataParallel as DDP
 
 from dlrover_amd.models import GPTConfig, NanoGPT
+from dlrover_amd.trainer.elastic import ElasticTrainer
 from dlrover_amd.trainer.flash_checkpoint import DdpCheckpointer, StorageType
 
 
@@ -55,6 +56,7 @@ def main():
     model = DDP(model)
     opt = torch.optim.AdamW(model.parameters(), lr=args.lr)
 
+    trainer = ElasticTrainer(model)  # reports global step for the monitor
     cp = DdpCheckpointer(args.ckpt_dir, model, opt)
     start_step = 0
     restored = cp.load_checkpoint()
@@ -64,16 +66,26 @@ def main():
             print(f"[train] resumed from checkpoint step {start_step}", flush=True)
 
     kill_at = int(os.getenv("DLROVER_TEST_KILL_AT_STEP", "0"))
+    hang_at = int(os.getenv("DLROVER_TEST_HANG_AT_STEP", "0"))
     incarnation = int(os.getenv("TORCHELASTIC_RESTART_COUNT", "0"))
 
     torch.manual_seed(1 + rank)
     ids = torch.randint(0, cfg.vocab_size, (args.batch, cfg.block_size), device=device)
 
     for step in range(start_step + 1, args.steps + 1):
+        if hang_at and step == hang_at and incarnation == 0 and rank == 0:
+            # simulate a wedged collective: this rank stops participating
+            # (config #5 — the master's hang diagnostician must recover us)
+            print(f"[train] injecting hang at step {step}", flush=True)
+            import time as _t
+
+            _t.sleep(3600)
         loss = model(ids, ids.clone())
         opt.zero_grad()
         loss.backward()
         opt.step()
+        trainer.global_step = step
+        trainer._report_step()
         if step % args.ckpt_interval == 0:
             sd = {
                 "step": step,
