@@ -92,6 +92,9 @@ def apply_fsdp(model, world):
 def main():
     args = parse_args()
     rank, world, local_rank, on_gpu = setup_dist(args)
+    from dlrover_amd.utils.numa import maybe_bind_from_env
+
+    maybe_bind_from_env()
     device = torch.device(f"cuda:{local_rank}" if on_gpu else "cpu")
     torch.manual_seed(1234 + rank)
 

@@ -407,6 +407,9 @@ def launch_agent(
     start the IPC server + async ckpt saver, report rendezvous parameters,
     run the (optional) node check, then the training agent."""
     config.auto_configure()
+    if config.numa_affinity:
+        # workers self-bind to their GPU's NUMA node at startup
+        os.environ["DLROVER_NUMA_BIND"] = "1"
     client = MasterClient.singleton_instance()
     client.report_rdzv_params(
         config.min_nodes,
