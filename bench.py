@@ -55,10 +55,15 @@ def setup_dist(args):
         os.environ.setdefault("WORLD_SIZE", "1")
         os.environ.setdefault("LOCAL_RANK", "0")
     on_gpu = torch.cuda.is_available()
+    backend = "nccl" if on_gpu else "gloo"  # nccl == RCCL on ROCm
     if on_gpu:
         torch.cuda.set_device(local_rank)
-    backend = "nccl" if on_gpu else "gloo"  # nccl == RCCL on ROCm
-    dist.init_process_group(backend=backend, rank=rank, world_size=world)
+        dist.init_process_group(
+            backend=backend, rank=rank, world_size=world,
+            device_id=torch.device(f"cuda:{local_rank}"),
+        )
+    else:
+        dist.init_process_group(backend=backend, rank=rank, world_size=world)
     return rank, world, local_rank, on_gpu
 
 
