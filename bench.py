@@ -116,6 +116,11 @@ def main():
     use_fsdp = on_gpu  # DTensor fully_shard needs a device mesh on GPU
     if use_fsdp:
         model = apply_fsdp(model, world)
+    elif world > 1:
+        # CPU plumbing lane: plain DDP over gloo so grads still sync
+        from torch.nn.parallel import DistributedDataParallel as DDP
+
+        model = DDP(model)
     opt = FusedAdamW(model.parameters(), lr=args.lr, weight_decay=0.1)
 
     ckpt_dir = os.path.join(os.getcwd(), "gpurun_out", "bench_ckpt")
