@@ -29,6 +29,7 @@
 #include <string.h>
 #include <unistd.h>
 
+#include <algorithm>
 #include <atomic>
 #include <chrono>
 #include <deque>
@@ -102,6 +103,7 @@ struct PendingOp {
   Category cat;
   double bytes;
   double enqueue_ts;
+  const char* name;  // owned by the HIP runtime (kernel symbol) or static
 };
 
 struct CatStats {
@@ -141,11 +143,12 @@ class Manager {
   }
 
   void record_end(hipStream_t stream, Category cat, double bytes,
-                  hipEvent_t start, hipEvent_t stop) {
+                  hipEvent_t start, hipEvent_t stop,
+                  const char* name = nullptr) {
     if (!enabled_ || start == nullptr) return;
     (void)hipEventRecord(stop, stream);
     std::lock_guard<std::mutex> g(q_mu_);
-    pending_.push_back({start, stop, cat, bytes, now()});
+    pending_.push_back({start, stop, cat, bytes, now(), name});
     launched_.fetch_add(1);
   }
 
@@ -225,6 +228,12 @@ class Manager {
         atomic_add(s.total_ms, (double)ms);
         atomic_max(s.max_ms, (double)ms);
         atomic_add(s.bytes, op.bytes);
+        if (op.name != nullptr) {
+          // per-kernel attribution (poller thread only: no lock needed)
+          auto& e = kernel_stats_[op.name];
+          e.first += 1;
+          e.second += (double)ms;
+        }
       }
       last_completion_ = now();
       {
@@ -287,6 +296,25 @@ class Manager {
       fprintf(f, "hiptimer_op_bytes_total{cat=\"%s\"} %.0f\n", kCatNames[c],
               s.bytes.load());
     }
+    // top kernels by total time (reference exposes per-kernel latency too)
+    {
+      std::vector<std::pair<std::string, std::pair<long, double>>> top(
+          kernel_stats_.begin(), kernel_stats_.end());
+      std::sort(top.begin(), top.end(), [](const auto& a, const auto& b) {
+        return a.second.second > b.second.second;
+      });
+      int n = 0;
+      for (auto& kv : top) {
+        if (++n > 20) break;
+        std::string nm = kv.first.substr(0, 120);
+        for (auto& c : nm)
+          if (c == '"' || c == '\\' || c == '\n') c = '_';
+        fprintf(f, "hiptimer_kernel_count{name=\"%s\"} %ld\n", nm.c_str(),
+                kv.second.first);
+        fprintf(f, "hiptimer_kernel_ms_total{name=\"%s\"} %.3f\n",
+                nm.c_str(), kv.second.second);
+      }
+    }
     fprintf(f, "hiptimer_device_alloc_bytes %.0f\n", (double)alloc_bytes_.load());
     fprintf(f, "hiptimer_device_free_total %ld\n", free_count_.load());
     fprintf(f, "hiptimer_host_alloc_bytes %.0f\n",
@@ -309,6 +337,7 @@ class Manager {
   std::mutex q_mu_;
   std::deque<PendingOp> pending_;
   CatStats stats_[CAT_COUNT];
+  std::map<std::string, std::pair<long, double>> kernel_stats_;
   std::atomic<long> launched_{0};
   std::atomic<long> alloc_bytes_{0};
   std::atomic<long> free_count_{0};
@@ -321,10 +350,14 @@ struct Scoped {
   hipStream_t stream;
   Category cat;
   double bytes;
-  Scoped(hipStream_t s, Category c, double b) : stream(s), cat(c), bytes(b) {
+  const char* name = nullptr;
+  Scoped(hipStream_t s, Category c, double b, const char* n = nullptr)
+      : stream(s), cat(c), bytes(b), name(n) {
     Manager::inst().record_begin(s, c, b, &start, &stop);
   }
-  void finish() { Manager::inst().record_end(stream, cat, bytes, start, stop); }
+  void finish() {
+    Manager::inst().record_end(stream, cat, bytes, start, stop, name);
+  }
 };
 
 }  // namespace hiptimer
@@ -346,7 +379,11 @@ hipError_t hipLaunchKernel(const void* function_address, dim3 numBlocks,
   using fn_t = hipError_t (*)(const void*, dim3, dim3, void**, size_t,
                               hipStream_t);
   static fn_t fn = (fn_t)real("hipLaunchKernel");
-  hiptimer::Scoped sc(stream, CAT_KERNEL, 0);
+  using name_fn_t = const char* (*)(const void*, hipStream_t);
+  static name_fn_t name_fn = (name_fn_t)real("hipKernelNameRefByPtr");
+  const char* kname =
+      name_fn != nullptr ? name_fn(function_address, stream) : nullptr;
+  hiptimer::Scoped sc(stream, CAT_KERNEL, 0, kname);
   hipError_t rc =
       fn(function_address, numBlocks, dimBlocks, args, sharedMemBytes, stream);
   sc.finish();
