@@ -119,10 +119,12 @@ class Attention(nn.Module):
         if use_flash:
             from dlrover_amd.ops import flash_attention
 
+            # permuted VIEWS: the stride-aware kernels read [B,S,H,D] storage
+            # directly — zero transpose copies in or out
             out = flash_attention(
-                q.transpose(1, 2).contiguous(),
-                k.transpose(1, 2).contiguous(),
-                v.transpose(1, 2).contiguous(),
+                q.transpose(1, 2),
+                k.transpose(1, 2),
+                v.transpose(1, 2),
                 1.0 / math.sqrt(hd),
             )
             out = out.transpose(1, 2).reshape(B, S, nh * hd)

@@ -164,7 +164,11 @@ class _RoPE(torch.autograd.Function):
     def forward(ctx, x, pos, cos, sin):
         ctx.save_for_backward(pos, cos, sin)
         if _use_hip(x):
-            out = x.contiguous().clone()
+            # one copy, not two: contiguous() of a strided view already
+            # produces a fresh tensor the kernel may mutate in place
+            out = x.contiguous()
+            if out is x:
+                out = x.clone()
             hip_ops().rope_apply(out, pos.int(), cos, sin, False)
             return out
         return rope_ref(x, pos, cos, sin)
@@ -173,7 +177,9 @@ class _RoPE(torch.autograd.Function):
     def backward(ctx, dy):
         pos, cos, sin = ctx.saved_tensors
         if _use_hip(dy):
-            dx = dy.contiguous().clone()
+            dx = dy.contiguous()
+            if dx is dy:
+                dx = dy.clone()
             hip_ops().rope_apply(dx, pos.int(), cos, sin, True)
             return dx, None, None, None
         # inverse rotation
@@ -256,7 +262,14 @@ class _FlashAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale):
         if _use_hip(q):
-            q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+            # strided [B,H,S,D] views are fine (kernels are stride-aware);
+            # only the head dim must be contiguous
+            if q.stride(-1) != 1:
+                q = q.contiguous()
+            if k.stride(-1) != 1:
+                k = k.contiguous()
+            if v.stride(-1) != 1:
+                v = v.contiguous()
             out, lse = hip_ops().flash_attn_fwd(q, k, v, scale)
             ctx.save_for_backward(q, k, v, out, lse)
             ctx.scale = scale

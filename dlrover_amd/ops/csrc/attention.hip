@@ -55,10 +55,17 @@ __device__ __forceinline__ bf16x8 ld_frag_b128(const char* lds_base, int row,
   return *reinterpret_cast<const bf16x8*>(lds_base + off);
 }
 
+// All tensors are logically [B, H, S, D] with EXPLICIT strides (element
+// units, D contiguous) — callers pass permuted views, so [B,S,H,D] storage
+// needs no transpose copies.
 __global__ __launch_bounds__(256) void flash_attn_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ out,
-    float* __restrict__ lse, int B, int H, int HKV, int S, float scale) {
+    float* __restrict__ lse, int B, int H, int HKV, int S, float scale,
+    long long qs_b, long long qs_h, long long qs_s,
+    long long ks_b, long long ks_h, long long ks_s,
+    long long vs_b, long long vs_h, long long vs_s,
+    long long os_b, long long os_h, long long os_s) {
   // LDS: K [64][128] bf16 (rows padded to 256B as-is), V^T [128][64] bf16,
   // P [4 waves][16][64] bf16 — all swizzled.
   __shared__ char k_lds[FA_KT * FA_D * 2];        // 16 KB, row stride 256 B
@@ -76,8 +83,10 @@ __global__ __launch_bounds__(256) void flash_attn_fwd_kernel(
   const int b = blockIdx.z;
   const int g = h / (H / HKV);  // GQA kv head
 
-  const long long q_base = (((long long)b * H + h) * S + qt * FA_QT) * FA_D;
-  const long long kv_base = (((long long)b * HKV + g) * S) * FA_D;
+  const short* q_blk = q + (long long)b * qs_b + (long long)h * qs_h +
+                       (long long)(qt * FA_QT) * qs_s;
+  const short* k_head = k + (long long)b * ks_b + (long long)g * ks_h;
+  const short* v_head = v + (long long)b * vs_b + (long long)g * vs_h;
 
   // ---- load this wave's Q band [16,128] into A fragments (persistent) ----
   // frag ks (k-slice of 32): lane holds Q[qrow0 + sub][ks*32 + quarter*8 + i]
@@ -86,7 +95,7 @@ __global__ __launch_bounds__(256) void flash_attn_fwd_kernel(
 #pragma unroll
   for (int ks = 0; ks < 4; ++ks) {
     const short* src =
-        q + q_base + (long long)(qrow_w + sub) * FA_D + ks * 32 + quarter * 8;
+        q_blk + (long long)(qrow_w + sub) * qs_s + ks * 32 + quarter * 8;
     aq[ks] = *reinterpret_cast<const bf16x8*>(src);
   }
 
@@ -106,18 +115,20 @@ __global__ __launch_bounds__(256) void flash_attn_fwd_kernel(
     // ---- stage K tile (row-major, swizzled) + V tile (transposed) ----
     // 64*128 elems, 256 threads -> 32 elems (4x b128 chunks) per thread
     {
-      const short* ksrc = k + kv_base + (long long)kt * FA_KT * FA_D;
-      const short* vsrc = v + kv_base + (long long)kt * FA_KT * FA_D;
+      const short* ksrc = k_head + (long long)(kt * FA_KT) * ks_s;
+      const short* vsrc = v_head + (long long)(kt * FA_KT) * vs_s;
 #pragma unroll
       for (int c = 0; c < 4; ++c) {
         const int linear = (tid * 4 + c) * 8;  // 8-elem chunk start
         const int row = linear / FA_D;
         const int col = linear % FA_D;
-        // K: contiguous global read, swizzled b128 LDS write
-        bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(ksrc + linear);
+        // K: strided-row global read, swizzled b128 LDS write
+        bf16x8 kv8 =
+            *reinterpret_cast<const bf16x8*>(ksrc + (long long)row * ks_s + col);
         *reinterpret_cast<bf16x8*>(k_lds + row * 256 + swz(row, col * 2)) = kv8;
         // V: same read pattern, transposed scalar writes (row<->col)
-        bf16x8 vv8 = *reinterpret_cast<const bf16x8*>(vsrc + linear);
+        bf16x8 vv8 =
+            *reinterpret_cast<const bf16x8*>(vsrc + (long long)row * vs_s + col);
 #pragma unroll
         for (int i = 0; i < 8; ++i) {
           const int trow = col + i;  // head-dim index
@@ -229,15 +240,15 @@ __global__ __launch_bounds__(256) void flash_attn_fwd_kernel(
   }
 
   // ---- epilogue: normalize, store O and logsumexp ----
-  const long long o_base = q_base + (long long)qrow_w * FA_D;
+  short* o_blk = out + (long long)b * os_b + (long long)h * os_h +
+                 (long long)(qt * FA_QT + qrow_w) * os_s;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const float inv = l_run[r] > 0.f ? 1.f / l_run[r] : 0.f;
     const int orow = quarter * 4 + r;
 #pragma unroll
     for (int n = 0; n < 8; ++n) {
-      out[o_base + (long long)orow * FA_D + n * 16 + sub] =
-          f2bf(acc_o[n][r] * inv);
+      o_blk[(long long)orow * os_s + n * 16 + sub] = f2bf(acc_o[n][r] * inv);
     }
     if (sub == 0) {
       const long long lrow =
@@ -250,9 +261,13 @@ __global__ __launch_bounds__(256) void flash_attn_fwd_kernel(
 extern "C" void flash_attn_fwd_launch(const void* q, const void* k,
                                       const void* v, void* out, void* lse,
                                       int B, int H, int HKV, int S,
-                                      float scale, hipStream_t stream) {
+                                      float scale, const long long* strides,
+                                      hipStream_t stream) {
   dim3 grid(S / FA_QT, H, B);
   hipLaunchKernelGGL(flash_attn_fwd_kernel, grid, dim3(256), 0, stream,
                      (const short*)q, (const short*)k, (const short*)v,
-                     (short*)out, (float*)lse, B, H, HKV, S, scale);
+                     (short*)out, (float*)lse, B, H, HKV, S, scale,
+                     strides[0], strides[1], strides[2], strides[3],
+                     strides[4], strides[5], strides[6], strides[7],
+                     strides[8], strides[9], strides[10], strides[11]);
 }

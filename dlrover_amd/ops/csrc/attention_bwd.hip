@@ -43,14 +43,15 @@ __device__ __forceinline__ bf16x8 ld_a(const char* lds, int sub, int k0,
 // stage a [64][128] bf16 global tile into LDS row-major (swizzled), and
 // optionally also transposed into a [128][64] buffer.
 __device__ __forceinline__ void stage_tile(const short* __restrict__ src,
-                                           char* row_lds, char* tr_lds,
-                                           int tid) {
+                                           long long row_stride, char* row_lds,
+                                           char* tr_lds, int tid) {
 #pragma unroll
   for (int c = 0; c < 4; ++c) {
     const int linear = (tid * 4 + c) * 8;
     const int row = linear / FA_D;
     const int col = linear % FA_D;
-    bf16x8 v8 = *reinterpret_cast<const bf16x8*>(src + linear);
+    bf16x8 v8 =
+        *reinterpret_cast<const bf16x8*>(src + (long long)row * row_stride + col);
     if (row_lds)
       *reinterpret_cast<bf16x8*>(row_lds + row * 256 + swz2(row, col * 2)) = v8;
     if (tr_lds) {
@@ -96,7 +97,11 @@ __global__ __launch_bounds__(256) void fa_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ dvec,
-    short* __restrict__ dq, int B, int H, int HKV, int S, float scale) {
+    short* __restrict__ dq, int B, int H, int HKV, int S, float scale,
+    long long qs_b, long long qs_h, long long qs_s,
+    long long ks_b, long long ks_h, long long ks_s,
+    long long vs_b, long long vs_h, long long vs_s,
+    long long ds_b, long long ds_h, long long ds_s) {
   __shared__ char k_lds[FA_T * 256];    // K row-major [64][128]
   __shared__ char kt_lds[FA_D * 128];   // K^T [128][64]
   __shared__ char v_lds[FA_T * 256];    // V row-major [64][128]
@@ -110,8 +115,14 @@ __global__ __launch_bounds__(256) void fa_bwd_dq_kernel(
 
   const int qt = blockIdx.x, h = blockIdx.y, b = blockIdx.z;
   const int g = h / (H / HKV);
-  const long long q_base = (((long long)b * H + h) * S + qt * FA_T) * FA_D;
-  const long long kv_base = (((long long)b * HKV + g) * S) * FA_D;
+  // dout shares q's layout (torch grads of a permuted view may differ, the
+  // binding normalizes); q/dout/dq all use the q-strides triple here
+  const short* q_blk = q + (long long)b * qs_b + (long long)h * qs_h +
+                       (long long)(qt * FA_T) * qs_s;
+  const short* do_blk = dout + (long long)b * ds_b + (long long)h * ds_h +
+                        (long long)(qt * FA_T) * ds_s;
+  const short* k_head = k + (long long)b * ks_b + (long long)g * ks_h;
+  const short* v_head = v + (long long)b * vs_b + (long long)g * vs_h;
   const long long row_base = ((long long)b * H + h) * S + qt * FA_T;
   const int qrow_w = wave * 16;
 
@@ -119,10 +130,11 @@ __global__ __launch_bounds__(256) void fa_bwd_dq_kernel(
   bf16x8 aq[4], ado[4];
 #pragma unroll
   for (int ks = 0; ks < 4; ++ks) {
-    const long long off = q_base + (long long)(qrow_w + sub) * FA_D +
-                          ks * 32 + quarter * 8;
-    aq[ks] = *reinterpret_cast<const bf16x8*>(q + off);
-    ado[ks] = *reinterpret_cast<const bf16x8*>(dout + off);
+    const int ko = ks * 32 + quarter * 8;
+    aq[ks] = *reinterpret_cast<const bf16x8*>(
+        q_blk + (long long)(qrow_w + sub) * qs_s + ko);
+    ado[ks] = *reinterpret_cast<const bf16x8*>(
+        do_blk + (long long)(qrow_w + sub) * ds_s + ko);
   }
   // per-row stats for this lane's 4 C-rows
   float lse_r[4], dv_r[4];
@@ -139,8 +151,8 @@ __global__ __launch_bounds__(256) void fa_bwd_dq_kernel(
   char* ds_wave = ds_lds + wave * 16 * 128;
 
   for (int kt = 0; kt <= qt; ++kt) {
-    stage_tile(k + kv_base + (long long)kt * FA_T * FA_D, k_lds, kt_lds, tid);
-    stage_tile(v + kv_base + (long long)kt * FA_T * FA_D, v_lds, nullptr, tid);
+    stage_tile(k_head + (long long)(kt * FA_T) * ks_s, ks_s, k_lds, kt_lds, tid);
+    stage_tile(v_head + (long long)(kt * FA_T) * vs_s, vs_s, v_lds, nullptr, tid);
     __syncthreads();
 
     // S = Q @ K^T ; dP = dO @ V^T  (both [16,64])
@@ -190,12 +202,14 @@ __global__ __launch_bounds__(256) void fa_bwd_dq_kernel(
     __syncthreads();
   }
 
+  short* dq_blk = dq + (long long)b * qs_b + (long long)h * qs_h +
+                  (long long)(qt * FA_T) * qs_s;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int orow = quarter * 4 + r;
 #pragma unroll
     for (int n = 0; n < 8; ++n)
-      dq[q_base + ((long long)qrow_w + orow) * FA_D + n * 16 + sub] =
+      dq_blk[((long long)qrow_w + orow) * qs_s + n * 16 + sub] =
           f2bf(acc_dq[n][r]);
   }
 }
@@ -209,7 +223,11 @@ __global__ __launch_bounds__(256) void fa_bwd_dkv_kernel(
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ dvec,
     float* __restrict__ dk32, float* __restrict__ dv32, int B, int H,
-    int HKV, int S, float scale) {
+    int HKV, int S, float scale,
+    long long qs_b, long long qs_h, long long qs_s,
+    long long ks_b, long long ks_h, long long ks_s,
+    long long vs_b, long long vs_h, long long vs_s,
+    long long ds_b, long long ds_h, long long ds_s) {
   __shared__ char q_lds[FA_T * 256];     // Q row-major
   __shared__ char qt_lds[FA_D * 128];    // Q^T
   __shared__ char do_lds[FA_T * 256];    // dO row-major
@@ -225,9 +243,12 @@ __global__ __launch_bounds__(256) void fa_bwd_dkv_kernel(
 
   const int kt = blockIdx.x, h = blockIdx.y, b = blockIdx.z;
   const int g = h / (H / HKV);
-  const long long kv_base =
-      (((long long)b * HKV + g) * S + kt * FA_T) * FA_D;
-  const long long qh_base = ((long long)b * H + h) * S * FA_D;
+  const short* k_blk = k + (long long)b * ks_b + (long long)g * ks_h +
+                       (long long)(kt * FA_T) * ks_s;
+  const short* v_blk = v + (long long)b * vs_b + (long long)g * vs_h +
+                       (long long)(kt * FA_T) * vs_s;
+  const short* q_head = q + (long long)b * qs_b + (long long)h * qs_h;
+  const short* do_head = dout + (long long)b * ds_b + (long long)h * ds_h;
   const long long row_base = ((long long)b * H + h) * S;
   const int kvrow_w = wave * 16;
 
@@ -235,10 +256,11 @@ __global__ __launch_bounds__(256) void fa_bwd_dkv_kernel(
   bf16x8 ak[4], av[4];
 #pragma unroll
   for (int ks = 0; ks < 4; ++ks) {
-    const long long off =
-        kv_base + (long long)(kvrow_w + sub) * FA_D + ks * 32 + quarter * 8;
-    ak[ks] = *reinterpret_cast<const bf16x8*>(k + off);
-    av[ks] = *reinterpret_cast<const bf16x8*>(v + off);
+    const int ko = ks * 32 + quarter * 8;
+    ak[ks] = *reinterpret_cast<const bf16x8*>(
+        k_blk + (long long)(kvrow_w + sub) * ks_s + ko);
+    av[ks] = *reinterpret_cast<const bf16x8*>(
+        v_blk + (long long)(kvrow_w + sub) * vs_s + ko);
   }
 
   f32x4_t acc_dk[8], acc_dv[8];
@@ -252,8 +274,8 @@ __global__ __launch_bounds__(256) void fa_bwd_dkv_kernel(
 
   const int n_qt = S / FA_T;
   for (int qt = kt; qt < n_qt; ++qt) {
-    stage_tile(q + qh_base + (long long)qt * FA_T * FA_D, q_lds, qt_lds, tid);
-    stage_tile(dout + qh_base + (long long)qt * FA_T * FA_D, do_lds, dot_lds,
+    stage_tile(q_head + (long long)(qt * FA_T) * qs_s, qs_s, q_lds, qt_lds, tid);
+    stage_tile(do_head + (long long)(qt * FA_T) * ds_s, ds_s, do_lds, dot_lds,
                tid);
     __syncthreads();
 
@@ -314,12 +336,13 @@ __global__ __launch_bounds__(256) void fa_bwd_dkv_kernel(
     __syncthreads();
   }
 
-  // accumulate into fp32 dK/dV (GQA head groups collide -> device atomics)
+  // accumulate into fp32 dK/dV (GQA head groups collide -> device atomics).
+  // dk32/dv32 are CONTIGUOUS [B,HKV,S,D] buffers allocated by the binding.
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int krow = quarter * 4 + r;
     const long long out_off =
-        kv_base + ((long long)kvrow_w + krow) * FA_D;
+        (((long long)b * HKV + g) * S + kt * FA_T + kvrow_w + krow) * FA_D;
 #pragma unroll
     for (int n = 0; n < 8; ++n) {
       atomicAdd(&dk32[out_off + n * 16 + sub], acc_dk[n][r]);
@@ -348,24 +371,28 @@ void fa_bwd_pre_launch(const void* dout, const void* out, void* dvec,
 void fa_bwd_dq_launch(const void* q, const void* k, const void* v,
                       const void* dout, const void* lse, const void* dvec,
                       void* dq, int B, int H, int HKV, int S, float scale,
-                      hipStream_t stream) {
+                      const long long* st, hipStream_t stream) {
   dim3 grid(S / FA_T, H, B);
   hipLaunchKernelGGL(fa_bwd_dq_kernel, grid, dim3(256), 0, stream,
                      (const short*)q, (const short*)k, (const short*)v,
                      (const short*)dout, (const float*)lse,
-                     (const float*)dvec, (short*)dq, B, H, HKV, S, scale);
+                     (const float*)dvec, (short*)dq, B, H, HKV, S, scale,
+                     st[0], st[1], st[2], st[3], st[4], st[5], st[6], st[7],
+                     st[8], st[9], st[10], st[11]);
 }
 
 void fa_bwd_dkv_launch(const void* q, const void* k, const void* v,
                        const void* dout, const void* lse, const void* dvec,
                        void* dk32, void* dv32, int B, int H, int HKV, int S,
-                       float scale, hipStream_t stream) {
+                       float scale, const long long* st, hipStream_t stream) {
   dim3 grid(S / FA_T, H, B);
   hipLaunchKernelGGL(fa_bwd_dkv_kernel, grid, dim3(256), 0, stream,
                      (const short*)q, (const short*)k, (const short*)v,
                      (const short*)dout, (const float*)lse,
                      (const float*)dvec, (float*)dk32, (float*)dv32, B, H,
-                     HKV, S, scale);
+                     HKV, S, scale,
+                     st[0], st[1], st[2], st[3], st[4], st[5], st[6], st[7],
+                     st[8], st[9], st[10], st[11]);
 }
 
 void f32_to_bf16_launch(const void* src, void* dst, long long n,

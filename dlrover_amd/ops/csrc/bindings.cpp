@@ -26,14 +26,15 @@ void cross_entropy_launch(void*, const void*, void*, long long, int, int,
                           float, int, void*);
 void mfma16_probe_launch(const void*, const void*, void*, void*);
 void flash_attn_fwd_launch(const void*, const void*, const void*, void*,
-                           void*, int, int, int, int, float, void*);
+                           void*, int, int, int, int, float,
+                           const long long*, void*);
 void fa_bwd_pre_launch(const void*, const void*, void*, long long, void*);
 void fa_bwd_dq_launch(const void*, const void*, const void*, const void*,
                       const void*, const void*, void*, int, int, int, int,
-                      float, void*);
+                      float, const long long*, void*);
 void fa_bwd_dkv_launch(const void*, const void*, const void*, const void*,
                        const void*, const void*, void*, void*, int, int, int,
-                       int, float, void*);
+                       int, float, const long long*, void*);
 void f32_to_bf16_launch(const void*, void*, long long, void*);
 }
 
@@ -173,53 +174,119 @@ void causal_softmax_bwd(at::Tensor& dscores, const at::Tensor& probs,
                             row_len, (float)scale, cur_stream());
 }
 
+// q/k/v/out are LOGICALLY [B,H,S,D]; strided permuted views are fine as long
+// as the last (D) dim is contiguous — no transpose copies needed.
+static void check_attn_view(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == at::kBFloat16, name,
+              " must be bf16 GPU");
+  TORCH_CHECK(t.dim() == 4 && t.stride(3) == 1, name,
+              " must be [B,H,S,D] with contiguous D");
+}
+
+static void pack_strides(long long* st, const at::Tensor& t, int at) {
+  st[at + 0] = (long long)t.stride(0);
+  st[at + 1] = (long long)t.stride(1);
+  st[at + 2] = (long long)t.stride(2);
+}
+
 std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
                                                   const at::Tensor& k,
                                                   const at::Tensor& v,
                                                   double scale) {
-  check_bf16(q, "q");
-  check_bf16(k, "k");
-  check_bf16(v, "v");
-  TORCH_CHECK(q.dim() == 4 && k.dim() == 4, "q/k/v must be [B,H,S,D]");
+  check_attn_view(q, "q");
+  check_attn_view(k, "k");
+  check_attn_view(v, "v");
   const int B = (int)q.size(0), H = (int)q.size(1), S = (int)q.size(2);
   const int D = (int)q.size(3), HKV = (int)k.size(1);
   TORCH_CHECK(D == 128, "flash_attn: head_dim must be 128");
   TORCH_CHECK(S % 64 == 0, "flash_attn: seq len must be a multiple of 64");
   TORCH_CHECK(H % HKV == 0, "flash_attn: H % HKV != 0");
-  auto out = at::empty_like(q);
+  // output in [B,S,H,D] storage (what the model consumes — avoids the
+  // transpose-back copy); returned as a [B,H,S,D] view
+  auto out_bshd = at::empty({B, S, H, D}, q.options());
+  auto out = out_bshd.permute({0, 2, 1, 3});
   auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
+  long long st[12];
+  pack_strides(st, q, 0);
+  pack_strides(st, k, 3);
+  pack_strides(st, v, 6);
+  pack_strides(st, out, 9);
   flash_attn_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(),
-                        out.data_ptr(), lse.data_ptr(), B, H, HKV, S,
-                        (float)scale, cur_stream());
+                        out_bshd.data_ptr(), lse.data_ptr(), B, H, HKV, S,
+                        (float)scale, st, cur_stream());
   return {out, lse};
 }
 
 std::tuple<at::Tensor, at::Tensor, at::Tensor> flash_attn_bwd(
     const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
-    const at::Tensor& out, const at::Tensor& dout, const at::Tensor& lse,
+    const at::Tensor& out, const at::Tensor& dout_in, const at::Tensor& lse,
     double scale) {
-  check_bf16(q, "q");
-  check_bf16(dout, "dout");
+  check_attn_view(q, "q");
+  check_attn_view(k, "k");
+  check_attn_view(v, "v");
   const int B = (int)q.size(0), H = (int)q.size(1), S = (int)q.size(2);
-  const int HKV = (int)k.size(1);
+  const int D = (int)q.size(3), HKV = (int)k.size(1);
+  // the preprocess + dkv stage read dO/O with [B,S,H,D]-style strides; accept
+  // any incoming grad layout by normalizing dO into out's layout
+  at::Tensor dout = dout_in;
+  if (dout.strides() != out.strides()) {
+    dout = at::empty_like(out.permute({0, 2, 1, 3}).contiguous())
+               .view({B, S, H, D});
+    dout = dout.permute({0, 2, 1, 3});
+    dout.copy_(dout_in);
+  }
+  TORCH_CHECK(dout.stride(3) == 1, "dout needs contiguous D");
   const long long n_rows = (long long)B * H * S;
   auto dvec = at::empty({B, H, S}, q.options().dtype(at::kFloat));
-  fa_bwd_pre_launch(dout.data_ptr(), out.data_ptr(), dvec.data_ptr(), n_rows,
-                    cur_stream());
-  auto dq = at::empty_like(q);
-  fa_bwd_dq_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), dout.data_ptr(),
-                   lse.data_ptr(), dvec.data_ptr(), dq.data_ptr(), B, H, HKV,
-                   S, (float)scale, cur_stream());
-  auto dk32 = at::zeros(k.sizes(), k.options().dtype(at::kFloat));
-  auto dv32 = at::zeros(v.sizes(), v.options().dtype(at::kFloat));
-  fa_bwd_dkv_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), dout.data_ptr(),
-                    lse.data_ptr(), dvec.data_ptr(), dk32.data_ptr(),
-                    dv32.data_ptr(), B, H, HKV, S, (float)scale, cur_stream());
-  auto dk = at::empty_like(k);
-  auto dv = at::empty_like(v);
-  f32_to_bf16_launch(dk32.data_ptr(), dk.data_ptr(), dk32.numel(), cur_stream());
-  f32_to_bf16_launch(dv32.data_ptr(), dv.data_ptr(), dv32.numel(), cur_stream());
-  return {dq, dk, dv};
+  {
+    // the pre kernel wants row-contiguous [N, D]: build contiguous views
+    auto dof = dout.is_contiguous() ? dout : dout.contiguous();
+    auto of = out.is_contiguous() ? out : out.contiguous();
+    fa_bwd_pre_launch(dof.data_ptr(), of.data_ptr(), dvec.data_ptr(), n_rows,
+                      cur_stream());
+    long long st[12];
+    pack_strides(st, q, 0);
+    pack_strides(st, k, 3);
+    pack_strides(st, v, 6);
+    pack_strides(st, dout, 9);
+    // dq mirrors q's layout: allocate [B,S,H,D] storage, return as view
+    auto dq_bshd = at::empty({B, S, H, D}, q.options());
+    auto dq = dq_bshd.permute({0, 2, 1, 3});
+    {
+      long long stq[12];
+      pack_strides(stq, dq, 0);
+      pack_strides(stq, k, 3);
+      pack_strides(stq, v, 6);
+      pack_strides(stq, dout, 9);
+      // q-strides slot is used for BOTH q loads and dq stores in the kernel;
+      // they must match, so stage q into dq's layout if they differ
+      at::Tensor quse = q;
+      if (q.strides() != dq.strides()) {
+        auto q_bshd = at::empty({B, S, H, D}, q.options());
+        auto qv = q_bshd.permute({0, 2, 1, 3});
+        qv.copy_(q);
+        quse = qv;
+        pack_strides(stq, quse, 0);
+      }
+      fa_bwd_dq_launch(quse.data_ptr(), k.data_ptr(), v.data_ptr(),
+                       dout.data_ptr(), lse.data_ptr(), dvec.data_ptr(),
+                       dq_bshd.data_ptr(), B, H, HKV, S, (float)scale, stq,
+                       cur_stream());
+      auto dk32 = at::zeros({B, HKV, S, D}, k.options().dtype(at::kFloat));
+      auto dv32 = at::zeros({B, HKV, S, D}, v.options().dtype(at::kFloat));
+      fa_bwd_dkv_launch(quse.data_ptr(), k.data_ptr(), v.data_ptr(),
+                        dout.data_ptr(), lse.data_ptr(), dvec.data_ptr(),
+                        dk32.data_ptr(), dv32.data_ptr(), B, H, HKV, S,
+                        (float)scale, stq, cur_stream());
+      auto dk = at::empty({B, HKV, S, D}, k.options());
+      auto dv = at::empty({B, HKV, S, D}, v.options());
+      f32_to_bf16_launch(dk32.data_ptr(), dk.data_ptr(), dk32.numel(),
+                         cur_stream());
+      f32_to_bf16_launch(dv32.data_ptr(), dv.data_ptr(), dv32.numel(),
+                         cur_stream());
+      return {dq, dk, dv};
+    }
+  }
 }
 
 at::Tensor mfma16_probe(const at::Tensor& A, const at::Tensor& B) {
