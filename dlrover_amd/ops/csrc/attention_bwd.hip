@@ -40,6 +40,17 @@ __device__ __forceinline__ bf16x8 ld_a(const char* lds, int sub, int k0,
   return *reinterpret_cast<const bf16x8*>(lds + sub * row_bytes + swz2(sub, bc));
 }
 
+// B-fragment read straight from GLOBAL memory: B[k][n] = src[n][k] where src
+// rows are row_stride apart. The 64x128 tile is L1/L2-resident across the
+// 16 fragment reads per iteration, so skipping the LDS copy trades a little
+// cache traffic for 32 KB of LDS (occupancy: 1 -> 3 waves/SIMD in dkv).
+__device__ __forceinline__ bf16x8 ld_bT_global(const short* src,
+                                               long long row_stride, int n0,
+                                               int sub, int k0, int quarter) {
+  return *reinterpret_cast<const bf16x8*>(
+      src + (long long)(n0 + sub) * row_stride + k0 + quarter * 8);
+}
+
 // stage a [64][128] bf16 global tile into LDS row-major (swizzled), and
 // optionally also transposed into a [128][64] buffer.
 __device__ __forceinline__ void stage_tile(const short* __restrict__ src,
@@ -228,9 +239,9 @@ __global__ __launch_bounds__(256) void fa_bwd_dkv_kernel(
     long long ks_b, long long ks_h, long long ks_s,
     long long vs_b, long long vs_h, long long vs_s,
     long long ds_b, long long ds_h, long long ds_s) {
-  __shared__ char q_lds[FA_T * 256];     // Q row-major
+  // Q/dO row-major fragments read straight from global (L1/L2-resident);
+  // only the TRANSPOSED layouts need LDS. 48 KB/block -> 3 blocks/CU.
   __shared__ char qt_lds[FA_D * 128];    // Q^T
-  __shared__ char do_lds[FA_T * 256];    // dO row-major
   __shared__ char dot_lds[FA_D * 128];   // dO^T
   __shared__ char pt_lds[4 * 16 * 128];  // per-wave P^T [16kv][64q]
   __shared__ char dst_lds[4 * 16 * 128]; // per-wave dS^T [16kv][64q]
@@ -274,9 +285,10 @@ __global__ __launch_bounds__(256) void fa_bwd_dkv_kernel(
 
   const int n_qt = S / FA_T;
   for (int qt = kt; qt < n_qt; ++qt) {
-    stage_tile(q_head + (long long)(qt * FA_T) * qs_s, qs_s, q_lds, qt_lds, tid);
-    stage_tile(do_head + (long long)(qt * FA_T) * ds_s, ds_s, do_lds, dot_lds,
-               tid);
+    const short* q_tile = q_head + (long long)(qt * FA_T) * qs_s;
+    const short* do_tile = do_head + (long long)(qt * FA_T) * ds_s;
+    stage_tile(q_tile, qs_s, nullptr, qt_lds, tid);
+    stage_tile(do_tile, ds_s, nullptr, dot_lds, tid);
     __syncthreads();
 
     // S^T = K_band @ Q^T ; dP^T = V_band @ dO^T   (both [16kv, 64q])
@@ -287,11 +299,12 @@ __global__ __launch_bounds__(256) void fa_bwd_dkv_kernel(
       acc_dpt[n] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ks = 0; ks < 4; ++ks) {
-        // B[k=dim][n=q] = Q[n][k] (row-major Q) / dO[n][k]
-        bf16x8 bq = ld_bT(q_lds, n * 16, sub, ks * 32, quarter, 256);
+        // B[k=dim][n=q] = Q[n][k] / dO[n][k] — read from global (cached)
+        bf16x8 bq = ld_bT_global(q_tile, qs_s, n * 16, sub, ks * 32, quarter);
         acc_st[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak[ks], bq,
                                                             acc_st[n], 0, 0, 0);
-        bf16x8 bdo = ld_bT(do_lds, n * 16, sub, ks * 32, quarter, 256);
+        bf16x8 bdo =
+            ld_bT_global(do_tile, ds_s, n * 16, sub, ks * 32, quarter);
         acc_dpt[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             av[ks], bdo, acc_dpt[n], 0, 0, 0);
       }
