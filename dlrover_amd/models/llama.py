@@ -22,6 +22,7 @@ from dlrover_amd.ops import (
     causal_softmax,
     cross_entropy_loss,
     rmsnorm,
+    rmsnorm_add,
     rope_rotate,
     swiglu,
 )
@@ -173,6 +174,16 @@ class Block(nn.Module):
         x = x + self.mlp(self.mlp_norm(x))
         return x
 
+    def forward_fused(self, branch, resid, pos, cos, sin):
+        """Residual-stream form: (branch, resid) in -> (branch, resid) out,
+        with both residual adds fused into the following RMSNorm kernel."""
+        normed, resid = rmsnorm_add(branch, resid, self.attn_norm.weight,
+                                    self.attn_norm.eps)
+        a = self.attn(normed, pos, cos, sin)
+        normed, resid = rmsnorm_add(a, resid, self.mlp_norm.weight,
+                                    self.mlp_norm.eps)
+        return self.mlp(normed), resid
+
 
 class LlamaForCausalLM(nn.Module):
     def __init__(self, cfg: LlamaConfig):
@@ -213,9 +224,27 @@ class LlamaForCausalLM(nn.Module):
         B, S = input_ids.shape
         pos = torch.arange(S, device=input_ids.device, dtype=torch.int32)
         x = self.embed(input_ids)
-        for blk in self.blocks:
-            x = blk(x, pos, self.rope_cos, self.rope_sin)
-        x = self.final_norm(x)
+        if input_ids.is_cuda:
+            # fused residual-stream form (identical math, fewer HBM passes)
+            branch, resid = None, x
+            for blk in self.blocks:
+                if branch is None:
+                    normed = blk.attn_norm(resid)
+                    a = blk.attn(normed, pos, self.rope_cos, self.rope_sin)
+                    normed, resid = rmsnorm_add(
+                        a, resid, blk.mlp_norm.weight, blk.mlp_norm.eps
+                    )
+                    branch = blk.mlp(normed)
+                else:
+                    branch, resid = blk.forward_fused(
+                        branch, resid, pos, self.rope_cos, self.rope_sin
+                    )
+            x, _ = rmsnorm_add(branch, resid, self.final_norm.weight,
+                               self.final_norm.eps)
+        else:
+            for blk in self.blocks:
+                x = blk(x, pos, self.rope_cos, self.rope_sin)
+            x = self.final_norm(x)
         logits = self.lm_head(x)
         if labels is None:
             return logits

@@ -18,6 +18,7 @@ from dlrover_amd.ops.api import (  # noqa: F401
     hip_ops,
     hip_ops_available,
     rmsnorm,
+    rmsnorm_add,
     rope_rotate,
     swiglu,
 )

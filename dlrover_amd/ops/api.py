@@ -118,7 +118,7 @@ class _RMSNorm(torch.autograd.Function):
     def backward(ctx, dy):
         x, w, invrms = ctx.saved_tensors
         if ctx.use_hip:
-            dx, dw = hip_ops().rmsnorm_bwd(dy.contiguous(), x, w, invrms)
+            dx, dw = hip_ops().rmsnorm_bwd(dy.contiguous(), x, w, invrms, None)
             return dx, dw, None
         h = x.shape[-1]
         xf, dyf, wf = x.float(), dy.float(), w.float()
@@ -131,6 +131,53 @@ class _RMSNorm(torch.autograd.Function):
 
 def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
     return _RMSNorm.apply(x, w, eps)
+
+
+class _RMSNormAdd(torch.autograd.Function):
+    """Fused h = x + resid; y = rmsnorm(h) * w. Two outputs: (y, h) — h is
+    the continuing residual stream, so its incoming gradient folds into the
+    norm backward in the same kernel pass."""
+
+    @staticmethod
+    def forward(ctx, x, resid, w, eps):
+        if _use_hip(x):
+            h, y, invrms = hip_ops().rmsnorm_add_fwd(
+                x.contiguous(), resid.contiguous(), w.contiguous(), eps
+            )
+            ctx.save_for_backward(h, w, invrms)
+            ctx.use_hip = True
+            return y, h
+        hf = x.float() + resid.float()
+        invrms = torch.rsqrt(hf.pow(2).mean(-1) + eps).reshape(-1)
+        h = hf.to(x.dtype)
+        y = rmsnorm_ref(h, w, eps)
+        ctx.save_for_backward(h, w, invrms)
+        ctx.use_hip = False
+        return y, h
+
+    @staticmethod
+    def backward(ctx, dy, dh):
+        h, w, invrms = ctx.saved_tensors
+        if ctx.use_hip:
+            dh_c = dh.contiguous() if dh is not None else None
+            dx, dw = hip_ops().rmsnorm_bwd(dy.contiguous(), h, w, invrms, dh_c)
+            return dx, dx, dw, None
+        hid = h.shape[-1]
+        hf, dyf, wf = h.float(), dy.float(), w.float()
+        inv = invrms.view(*h.shape[:-1], 1)
+        dot = (dyf * wf * hf).sum(-1, keepdim=True)
+        dxf = dyf * wf * inv - hf * (dot * inv.pow(3) / hid)
+        if dh is not None:
+            dxf = dxf + dh.float()
+        dw = (dyf * hf * inv).reshape(-1, hid).sum(0)
+        dx = dxf.to(h.dtype)
+        return dx, dx, dw.to(w.dtype), None
+
+
+def rmsnorm_add(x: torch.Tensor, resid: torch.Tensor, w: torch.Tensor,
+                eps: float = 1e-5):
+    """(normed, new_residual) = fused residual-add + RMSNorm."""
+    return _RMSNormAdd.apply(x, resid, w, eps)
 
 
 class _SwiGLU(torch.autograd.Function):

@@ -68,9 +68,12 @@ __global__ __launch_bounds__(256) void flash_attn_fwd_kernel(
     long long os_b, long long os_h, long long os_s) {
   // LDS: K [64][128] bf16 (rows padded to 256B as-is), V^T [128][64] bf16,
   // P [4 waves][16][64] bf16 — all swizzled.
-  __shared__ char k_lds[FA_KT * FA_D * 2];        // 16 KB, row stride 256 B
-  __shared__ char vt_lds[FA_D * FA_KT * 2];       // 16 KB, row stride 128 B
-  __shared__ char p_lds[4 * 16 * FA_KT * 2];      // 8 KB, row stride 128 B
+  // double-buffered K/V so the NEXT tile's HBM loads overlap this tile's
+  // MFMA (guide §5.5 T3 minimum 2-phase: stage-next before compute, one
+  // barrier pair per tile). 72 KB total still fits 2 blocks/CU.
+  __shared__ char k_lds[2][FA_KT * FA_D * 2];     // 2 x 16 KB
+  __shared__ char vt_lds[2][FA_D * FA_KT * 2];    // 2 x 16 KB
+  __shared__ char p_lds[4 * 16 * FA_KT * 2];      // 8 KB
 
   const int tid = threadIdx.x;
   const int wave = tid >> 6;
@@ -111,34 +114,41 @@ __global__ __launch_bounds__(256) void flash_attn_fwd_kernel(
 
   char* p_wave = p_lds + wave * 16 * FA_KT * 2;
 
-  for (int kt = 0; kt <= qt; ++kt) {
-    // ---- stage K tile (row-major, swizzled) + V tile (transposed) ----
-    // 64*128 elems, 256 threads -> 32 elems (4x b128 chunks) per thread
-    {
-      const short* ksrc = k_head + (long long)(kt * FA_KT) * ks_s;
-      const short* vsrc = v_head + (long long)(kt * FA_KT) * vs_s;
+  // ---- stage one K/V tile into buffer `buf` ----
+  auto stage = [&](int kt, int buf) {
+    const short* ksrc = k_head + (long long)(kt * FA_KT) * ks_s;
+    const short* vsrc = v_head + (long long)(kt * FA_KT) * vs_s;
+    char* kb = k_lds[buf];
+    char* vb = vt_lds[buf];
 #pragma unroll
-      for (int c = 0; c < 4; ++c) {
-        const int linear = (tid * 4 + c) * 8;  // 8-elem chunk start
-        const int row = linear / FA_D;
-        const int col = linear % FA_D;
-        // K: strided-row global read, swizzled b128 LDS write
-        bf16x8 kv8 =
-            *reinterpret_cast<const bf16x8*>(ksrc + (long long)row * ks_s + col);
-        *reinterpret_cast<bf16x8*>(k_lds + row * 256 + swz(row, col * 2)) = kv8;
-        // V: same read pattern, transposed scalar writes (row<->col)
-        bf16x8 vv8 =
-            *reinterpret_cast<const bf16x8*>(vsrc + (long long)row * vs_s + col);
+    for (int c = 0; c < 4; ++c) {
+      const int linear = (tid * 4 + c) * 8;  // 8-elem chunk start
+      const int row = linear / FA_D;
+      const int col = linear % FA_D;
+      bf16x8 kv8 =
+          *reinterpret_cast<const bf16x8*>(ksrc + (long long)row * ks_s + col);
+      *reinterpret_cast<bf16x8*>(kb + row * 256 + swz(row, col * 2)) = kv8;
+      bf16x8 vv8 =
+          *reinterpret_cast<const bf16x8*>(vsrc + (long long)row * vs_s + col);
 #pragma unroll
-        for (int i = 0; i < 8; ++i) {
-          const int trow = col + i;  // head-dim index
-          const int tcol = row;      // kv position
-          *reinterpret_cast<__bf16*>(
-              vt_lds + trow * 128 + swz(trow, tcol * 2)) = vv8[i];
-        }
+      for (int i = 0; i < 8; ++i) {
+        const int trow = col + i;  // head-dim index
+        const int tcol = row;      // kv position
+        *reinterpret_cast<__bf16*>(vb + trow * 128 + swz(trow, tcol * 2)) =
+            vv8[i];
       }
     }
-    __syncthreads();
+  };
+
+  stage(0, 0);
+  __syncthreads();
+
+  for (int kt = 0; kt <= qt; ++kt) {
+    const int cur = kt & 1;
+    char* k_cur = k_lds[cur];
+    char* vt_cur = vt_lds[cur];
+    // prefetch the NEXT tile into the other buffer while computing this one
+    if (kt < qt) stage(kt + 1, cur ^ 1);
 
     // ---- S_band = Q_band @ K^T ----
     f32x4_t acc_s[4];
@@ -154,7 +164,7 @@ __global__ __launch_bounds__(256) void flash_attn_fwd_kernel(
           const int row = n * 16 + sub;
           const int byte_col = (ks * 32 + quarter * 8) * 2;
           bk = *reinterpret_cast<const bf16x8*>(
-              k_lds + row * 256 + swz(row, byte_col));
+              k_cur + row * 256 + swz(row, byte_col));
         }
         acc_s[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[ks], bk,
                                                            acc_s[n], 0, 0, 0);
@@ -209,7 +219,7 @@ __global__ __launch_bounds__(256) void flash_attn_fwd_kernel(
         acc_o[n][r] *= alpha[r];
       }
     }
-    __syncthreads();  // P visible wave-locally; K/V reuse next iter needs all
+    __syncthreads();  // P staged + next-tile K/V writes landed
 
     // ---- O_band += P_band @ V ----
 #pragma unroll
@@ -230,13 +240,13 @@ __global__ __launch_bounds__(256) void flash_attn_fwd_kernel(
           const int row = n * 16 + sub;
           const int byte_col = (ks * 32 + quarter * 8) * 2;
           bv = *reinterpret_cast<const bf16x8*>(
-              vt_lds + row * 128 + swz(row, byte_col));
+              vt_cur + row * 128 + swz(row, byte_col));
         }
         acc_o[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, bv, acc_o[n],
                                                            0, 0, 0);
       }
     }
-    __syncthreads();  // done with this tile's K/V/P
+    __syncthreads();  // all reads of buf[cur] done before kt+2 overwrites it
   }
 
   // ---- epilogue: normalize, store O and logsumexp ----

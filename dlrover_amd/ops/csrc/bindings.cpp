@@ -11,8 +11,11 @@
 extern "C" {
 void rmsnorm_fwd_launch(const void*, const void*, void*, void*, int, int,
                         float, void*);
+void rmsnorm_add_fwd_launch(const void*, const void*, const void*, void*,
+                            void*, void*, int, int, float, void*);
 void rmsnorm_bwd_launch(const void*, const void*, const void*, const void*,
-                        void*, void*, void*, int, int, int, void*);
+                        const void*, void*, void*, void*, int, int, int,
+                        void*);
 void rope_launch(void*, const void*, const void*, const void*, long long,
                  long long, int, int, int, void*);
 void swiglu_fwd_launch(const void*, void*, long long, int, void*);
@@ -66,10 +69,25 @@ std::tuple<at::Tensor, at::Tensor> rmsnorm_fwd(const at::Tensor& x,
   return {y, invrms};
 }
 
-std::tuple<at::Tensor, at::Tensor> rmsnorm_bwd(const at::Tensor& dy,
-                                               const at::Tensor& x,
-                                               const at::Tensor& w,
-                                               const at::Tensor& invrms) {
+std::tuple<at::Tensor, at::Tensor, at::Tensor> rmsnorm_add_fwd(
+    const at::Tensor& x, const at::Tensor& resid, const at::Tensor& w,
+    double eps) {
+  check_bf16(x, "x");
+  check_bf16(resid, "resid");
+  const int hidden = (int)x.size(-1);
+  const long long n_rows = x.numel() / hidden;
+  auto h = at::empty_like(x);
+  auto y = at::empty_like(x);
+  auto invrms = at::empty({n_rows}, x.options().dtype(at::kFloat));
+  rmsnorm_add_fwd_launch(x.data_ptr(), resid.data_ptr(), w.data_ptr(),
+                         h.data_ptr(), y.data_ptr(), invrms.data_ptr(),
+                         (int)n_rows, hidden, (float)eps, cur_stream());
+  return {h, y, invrms};
+}
+
+std::tuple<at::Tensor, at::Tensor> rmsnorm_bwd(
+    const at::Tensor& dy, const at::Tensor& x, const at::Tensor& w,
+    const at::Tensor& invrms, c10::optional<at::Tensor> dh_extra) {
   check_bf16(dy, "dy");
   check_bf16(x, "x");
   const int hidden = (int)x.size(-1);
@@ -81,10 +99,15 @@ std::tuple<at::Tensor, at::Tensor> rmsnorm_bwd(const at::Tensor& dy,
   auto dw = at::empty_like(w);
   auto partial =
       at::empty({n_partials, hidden}, x.options().dtype(at::kFloat));
+  const void* dh_ptr = nullptr;
+  if (dh_extra.has_value()) {
+    check_bf16(*dh_extra, "dh_extra");
+    dh_ptr = dh_extra->data_ptr();
+  }
   rmsnorm_bwd_launch(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
-                     invrms.data_ptr(), dx.data_ptr(), partial.data_ptr(),
-                     dw.data_ptr(), (int)n_rows, hidden, n_partials,
-                     cur_stream());
+                     invrms.data_ptr(), dh_ptr, dx.data_ptr(),
+                     partial.data_ptr(), dw.data_ptr(), (int)n_rows, hidden,
+                     n_partials, cur_stream());
   return {dx, dw};
 }
 
@@ -322,6 +345,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "dlrover_amd CDNA4 (gfx950) kernels";
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused RMSNorm forward (bf16)");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "fused RMSNorm backward (bf16)");
+  m.def("rmsnorm_add_fwd", &rmsnorm_add_fwd,
+        "fused residual-add + RMSNorm forward -> (h, y, invrms)");
   m.def("rope_apply", &rope_apply, "in-place RoPE rotate-half (bf16)");
   m.def("swiglu_fwd", &swiglu_fwd, "fused SwiGLU forward (bf16)");
   m.def("swiglu_bwd", &swiglu_bwd, "fused SwiGLU backward (bf16)");

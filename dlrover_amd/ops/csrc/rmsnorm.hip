@@ -57,6 +57,8 @@ __global__ void rmsnorm_fwd_kernel(
 __global__ void rmsnorm_bwd_kernel(
     const short* __restrict__ dy, const short* __restrict__ x,
     const short* __restrict__ w, const float* __restrict__ invrms,
+    const short* __restrict__ dh_extra,  // nullable: fused-add path's
+                                         // residual-stream gradient
     short* __restrict__ dx, float* __restrict__ dw_partial,
     int n_rows, int hidden, int n_partials) {
   __shared__ float scratch[16];
@@ -97,6 +99,12 @@ __global__ void rmsnorm_bwd_kernel(
         out[j] = dyv[j] * wv[j] * inv - xv[j] * k;
         dw_acc[vi][j] += dyv[j] * xv[j] * inv;
       }
+      if (dh_extra) {
+        float dhv[8];
+        load8(dh_extra + (long long)row * hidden + v * 8, dhv);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) out[j] += dhv[j];
+      }
       store8(dxr + v * 8, out);
     }
     __syncthreads();
@@ -122,6 +130,60 @@ __global__ void reduce_partials_kernel(
   }
 }
 
+// Fused residual-add + RMSNorm: h = x + resid; y = h * invrms(h) * w.
+// Saves a full HBM round-trip of the eager residual add (4.7% of the 8B
+// step as at::add — profiles/r01c). Writes BOTH h (needed as the next
+// residual stream) and y.
+__global__ void rmsnorm_add_fwd_kernel(
+    const short* __restrict__ x, const short* __restrict__ resid,
+    const short* __restrict__ w, short* __restrict__ h, short* __restrict__ y,
+    float* __restrict__ invrms, int n_rows, int hidden, float eps) {
+  __shared__ float scratch[16];
+  const int vecs = hidden >> 3;
+  for (int row = blockIdx.x; row < n_rows; row += gridDim.x) {
+    const short* xr = x + (long long)row * hidden;
+    const short* rr = resid + (long long)row * hidden;
+    short* hr = h + (long long)row * hidden;
+    short* yr = y + (long long)row * hidden;
+    float ssq = 0.f;
+    for (int v = threadIdx.x; v < vecs; v += blockDim.x) {
+      float xv[8], rv[8], hv[8];
+      load8(xr + v * 8, xv);
+      load8(rr + v * 8, rv);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        hv[j] = xv[j] + rv[j];
+        ssq += hv[j] * hv[j];
+      }
+      store8(hr + v * 8, hv);
+    }
+    ssq = block_reduce_sum(ssq, scratch);
+    const float inv = rsqrtf(ssq / hidden + eps);
+    if (threadIdx.x == 0 && invrms) invrms[row] = inv;
+    for (int v = threadIdx.x; v < vecs; v += blockDim.x) {
+      float hv[8], wv[8], out[8];
+      load8(hr + v * 8, hv);
+      load8(w + v * 8, wv);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) out[j] = hv[j] * inv * wv[j];
+      store8(yr + v * 8, out);
+    }
+    __syncthreads();
+  }
+}
+
+extern "C" void rmsnorm_add_fwd_launch(const void* x, const void* resid,
+                                       const void* w, void* h, void* y,
+                                       void* invrms, int n_rows, int hidden,
+                                       float eps, hipStream_t stream) {
+  int grid = n_rows < 2048 ? n_rows : 2048;
+  if (grid < 1) grid = 1;
+  hipLaunchKernelGGL(rmsnorm_add_fwd_kernel, dim3(grid), dim3(256), 0, stream,
+                     (const short*)x, (const short*)resid, (const short*)w,
+                     (short*)h, (short*)y, (float*)invrms, n_rows, hidden,
+                     eps);
+}
+
 void rmsnorm_fwd_launch(const void* x, const void* w, void* y, void* invrms,
                         int n_rows, int hidden, float eps, hipStream_t stream) {
   int grid = n_rows < 2048 ? n_rows : 2048;
@@ -132,17 +194,17 @@ void rmsnorm_fwd_launch(const void* x, const void* w, void* y, void* invrms,
 }
 
 void rmsnorm_bwd_launch(const void* dy, const void* x, const void* w,
-                        const void* invrms, void* dx, void* dw_partial,
-                        void* dw, int n_rows, int hidden, int n_partials,
-                        hipStream_t stream) {
+                        const void* invrms, const void* dh_extra, void* dx,
+                        void* dw_partial, void* dw, int n_rows, int hidden,
+                        int n_partials, hipStream_t stream) {
   int grid = n_rows < 2048 ? n_rows : 2048;
   if (grid < 1) grid = 1;
   hipMemsetAsync(dw_partial, 0, (size_t)n_partials * hidden * sizeof(float),
                  stream);
   hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(grid), dim3(256), 0, stream,
                      (const short*)dy, (const short*)x, (const short*)w,
-                     (const float*)invrms, (short*)dx, (float*)dw_partial,
-                     n_rows, hidden, n_partials);
+                     (const float*)invrms, (const short*)dh_extra, (short*)dx,
+                     (float*)dw_partial, n_rows, hidden, n_partials);
   hipLaunchKernelGGL(reduce_partials_kernel, dim3(grid_capped(hidden, 256)),
                      dim3(256), 0, stream, (const float*)dw_partial,
                      (short*)dw, n_partials, hidden);

@@ -109,3 +109,49 @@ def test_model_grads_flash_vs_composite():
         denom = g2.norm().clamp_min(1e-6)
         rel = (g1 - g2).norm() / denom
         assert rel < 0.05, (n1, rel.item())
+
+
+def test_fused_residual_stream_matches_plain_forward():
+    """GPU fused residual-stream model form == plain per-block form."""
+    from dlrover_amd.models import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig(
+        vocab_size=512, hidden_size=512, intermediate_size=1024, n_layers=3,
+        n_heads=4, n_kv_heads=2, max_seq_len=256, rope_base=10000.0,
+    )
+    model = LlamaForCausalLM(cfg).cuda().bfloat16()
+    ids = torch.randint(0, 512, (2, 128), device="cuda")
+    loss_fused = model(ids, ids.clone())
+    # plain path: run the CPU-structured loop manually on GPU
+    pos = torch.arange(128, device="cuda", dtype=torch.int32)
+    x = model.embed(ids)
+    for blk in model.blocks:
+        x = blk(x, pos, model.rope_cos, model.rope_sin)
+    x = model.final_norm(x)
+    logits = x @ model.lm_head.weight.t()
+    ref = torch.nn.functional.cross_entropy(
+        logits.float().view(-1, 512), ids.clone().view(-1)
+    )
+    torch.testing.assert_close(loss_fused.float(), ref, rtol=2e-2, atol=2e-2)
+
+
+def test_rmsnorm_add_gpu_matches_ref():
+    from dlrover_amd.ops import rmsnorm_add
+
+    torch.manual_seed(1)
+    x = torch.randn(64, 4096, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    r = torch.randn(64, 4096, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(4096, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    y, h = rmsnorm_add(x, r, w, 1e-5)
+    (y.float().pow(2).sum() + h.float().sum() * 0.5).backward()
+    x2 = x.detach().float().requires_grad_(True)
+    r2 = r.detach().float().requires_grad_(True)
+    w2 = w.detach().float().requires_grad_(True)
+    h2 = x2 + r2
+    inv = torch.rsqrt(h2.pow(2).mean(-1, keepdim=True) + 1e-5)
+    y2 = h2 * inv * w2
+    (y2.pow(2).sum() + h2.sum() * 0.5).backward()
+    torch.testing.assert_close(y.float(), y2, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(x.grad.float(), x2.grad, rtol=5e-2, atol=5e-1)
+    torch.testing.assert_close(w.grad.float(), w2.grad, rtol=5e-2, atol=5e-1)

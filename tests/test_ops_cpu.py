@@ -139,3 +139,28 @@ def test_fused_adamw_optimizer_class():
         opt.step()
     st = opt.state[model.weight]
     assert st["step"] == 3 and "master_param" in st
+
+
+def test_rmsnorm_add_matches_unfused():
+    from dlrover_amd.ops import rmsnorm_add
+
+    torch.manual_seed(3)
+    x = torch.randn(4, 6, 64, requires_grad=True)
+    r = torch.randn(4, 6, 64, requires_grad=True)
+    w = torch.randn(64, requires_grad=True)
+    y, h = rmsnorm_add(x, r, w, 1e-5)
+    # h is used downstream too: grads flow through BOTH outputs
+    (y.pow(2).sum() + h.sum() * 0.5).backward()
+
+    x2 = x.detach().clone().requires_grad_(True)
+    r2 = r.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    h2 = x2 + r2
+    inv = torch.rsqrt(h2.pow(2).mean(-1, keepdim=True) + 1e-5)
+    y2 = h2 * inv * w2
+    (y2.pow(2).sum() + h2.sum() * 0.5).backward()
+    torch.testing.assert_close(y, y2, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(h, h2, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(x.grad, x2.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(r.grad, r2.grad, rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(w.grad, w2.grad, rtol=1e-4, atol=1e-4)
