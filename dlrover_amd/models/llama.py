@@ -169,16 +169,21 @@ class Block(nn.Module):
         self.mlp_norm = RMSNorm(cfg.hidden_size, cfg.norm_eps)
         self.mlp = MLP(cfg)
 
-    def forward(self, x, pos, cos, sin):
-        x = x + self.attn(self.attn_norm(x), pos, cos, sin)
-        x = x + self.mlp(self.mlp_norm(x))
-        return x
-
-    def forward_fused(self, branch, resid, pos, cos, sin):
-        """Residual-stream form: (branch, resid) in -> (branch, resid) out,
-        with both residual adds fused into the following RMSNorm kernel."""
-        normed, resid = rmsnorm_add(branch, resid, self.attn_norm.weight,
-                                    self.attn_norm.eps)
+    def forward(self, x, pos, cos, sin, resid=None):
+        """Two forms. Plain (resid=None): the classic pre-norm block.
+        Residual-stream (resid given): x is the previous sublayer's BRANCH
+        output (None on the first block); both residual adds fuse into the
+        following RMSNorm kernel; returns (branch, resid). Must be entered
+        through __call__ so FSDP's unshard hooks fire."""
+        if resid is None:
+            x = x + self.attn(self.attn_norm(x), pos, cos, sin)
+            x = x + self.mlp(self.mlp_norm(x))
+            return x
+        if x is None:
+            normed = self.attn_norm(resid)
+        else:
+            normed, resid = rmsnorm_add(x, resid, self.attn_norm.weight,
+                                        self.attn_norm.eps)
         a = self.attn(normed, pos, cos, sin)
         normed, resid = rmsnorm_add(a, resid, self.mlp_norm.weight,
                                     self.mlp_norm.eps)
@@ -225,20 +230,13 @@ class LlamaForCausalLM(nn.Module):
         pos = torch.arange(S, device=input_ids.device, dtype=torch.int32)
         x = self.embed(input_ids)
         if input_ids.is_cuda:
-            # fused residual-stream form (identical math, fewer HBM passes)
+            # fused residual-stream form (identical math, fewer HBM passes);
+            # blocks entered via __call__ so FSDP unshard hooks fire
             branch, resid = None, x
             for blk in self.blocks:
-                if branch is None:
-                    normed = blk.attn_norm(resid)
-                    a = blk.attn(normed, pos, self.rope_cos, self.rope_sin)
-                    normed, resid = rmsnorm_add(
-                        a, resid, blk.mlp_norm.weight, blk.mlp_norm.eps
-                    )
-                    branch = blk.mlp(normed)
-                else:
-                    branch, resid = blk.forward_fused(
-                        branch, resid, pos, self.rope_cos, self.rope_sin
-                    )
+                branch, resid = blk(
+                    branch, pos, self.rope_cos, self.rope_sin, resid=resid
+                )
             x, _ = rmsnorm_add(branch, resid, self.final_norm.weight,
                                self.final_norm.eps)
         else:
