@@ -223,6 +223,97 @@ class HttpRpcClient:
 
 
 # ---------------------------------------------------------------------------
+# gRPC (generic bytes handlers — no protoc-generated stubs needed)
+# ---------------------------------------------------------------------------
+
+
+class GrpcRpcServer:
+    """Two-verb service over gRPC, mirroring the reference's default
+    transport (ref: servicer.py:910 GrpcMasterServicer). Methods are
+    registered generically with bytes (de)serializers, so the wire payload is
+    the same restricted-pickle frame the tcp/http transports use."""
+
+    SERVICE = "dlrover.Master"
+
+    def __init__(self, port: int, handler: Handler, host: str = "0.0.0.0"):
+        import grpc
+        from concurrent import futures
+
+        self._handler = handler
+
+        def _unary(verb):
+            def call(request_bytes, context):
+                try:
+                    if verb == "ping":
+                        resp = BaseResponse(success=True)
+                    else:
+                        resp = handler(verb, serialize.loads(request_bytes))
+                except Exception as e:  # noqa: BLE001
+                    logger.exception("gRPC %s failed", verb)
+                    resp = BaseResponse(success=False, reason=repr(e))
+                return serialize.dumps(resp)
+
+            return grpc.unary_unary_rpc_method_handler(
+                call,
+                request_deserializer=lambda b: b,
+                response_serializer=lambda b: b,
+            )
+
+        service = grpc.method_handlers_generic_handler(
+            self.SERVICE,
+            {"get": _unary("get"), "report": _unary("report"),
+             "ping": _unary("ping")},
+        )
+        self._server = grpc.server(
+            futures.ThreadPoolExecutor(max_workers=32),
+            options=[
+                ("grpc.max_send_message_length", 256 << 20),
+                ("grpc.max_receive_message_length", 256 << 20),
+            ],
+        )
+        self._server.add_generic_rpc_handlers((service,))
+        self.port = self._server.add_insecure_port(f"{host}:{port}")
+
+    def start(self):
+        self._server.start()
+        logger.info("gRPC RPC server on port %s", self.port)
+        return self
+
+    def stop(self):
+        self._server.stop(grace=1)
+
+
+class GrpcRpcClient:
+    def __init__(self, addr: str, timeout: float = 30.0):
+        import grpc
+
+        self._timeout = timeout
+        self._channel = grpc.insecure_channel(
+            addr,
+            options=[
+                ("grpc.max_send_message_length", 256 << 20),
+                ("grpc.max_receive_message_length", 256 << 20),
+            ],
+        )
+        self._methods = {
+            verb: self._channel.unary_unary(
+                f"/{GrpcRpcServer.SERVICE}/{verb}",
+                request_serializer=lambda b: b,
+                response_deserializer=lambda b: b,
+            )
+            for verb in ("get", "report", "ping")
+        }
+
+    def call(self, verb: str, req: BaseRequest) -> BaseResponse:
+        payload = serialize.dumps(req) if verb != "ping" else b""
+        out = self._methods[verb](payload, timeout=self._timeout)
+        return serialize.loads(out)
+
+    def close(self):
+        self._channel.close()
+
+
+# ---------------------------------------------------------------------------
 # factories
 # ---------------------------------------------------------------------------
 
@@ -230,12 +321,16 @@ class HttpRpcClient:
 def create_rpc_server(service_type: str, port: int, handler: Handler):
     if service_type == CommServiceType.HTTP:
         return HttpRpcServer(port, handler)
+    if service_type == CommServiceType.GRPC:
+        return GrpcRpcServer(port, handler)
     return TcpRpcServer(port, handler)
 
 
 def create_rpc_client(service_type: str, addr: str, timeout: float = 30.0):
     if service_type == CommServiceType.HTTP:
         return HttpRpcClient(addr, timeout)
+    if service_type == CommServiceType.GRPC:
+        return GrpcRpcClient(addr, timeout)
     return TcpRpcClient(addr, timeout)
 
 
@@ -247,6 +342,6 @@ def wait_for_server(addr: str, timeout: float = 60.0, service_type: str = "tcp")
             client.call("ping", BaseRequest())
             client.close()
             return True
-        except (OSError, urlerror.URLError, ConnectionError):
+        except Exception:  # noqa: BLE001 — includes grpc.RpcError
             time.sleep(0.3)
     return False

@@ -106,3 +106,20 @@ def test_resource_monitor_snapshot():
 
     snap = ResourceMonitor().snapshot()
     assert snap.memory_mb > 0
+
+
+def test_master_over_grpc():
+    from dlrover_amd.utils.transport import GrpcRpcClient, GrpcRpcServer
+
+    JobContext._reset_for_tests()
+    m = LocalJobMaster(port=0, service_type=CommServiceType.GRPC).prepare()
+    try:
+        c = MasterClient(f"127.0.0.1:{m.port}", node_id=0,
+                         service_type=CommServiceType.GRPC)
+        c.kv_store_set("g", b"42")
+        assert c.kv_store_get("g") == b"42"
+        assert c.join_rendezvous(0, 8) >= 0
+        c.close()
+    finally:
+        m.stop()
+        JobContext._reset_for_tests()
