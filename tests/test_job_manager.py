@@ -1,0 +1,117 @@
+"""Job-manager failure policy unit tests: relaunch ladder, heartbeat death,
+elimination vs abort, status flow."""
+
+import pytest
+
+from dlrover_amd.common import comm
+from dlrover_amd.common.constants import (
+    JobStage,
+    NodeEventType,
+    NodeExitReason,
+    NodeStatus,
+    NodeType,
+)
+from dlrover_amd.common.node import Node
+from dlrover_amd.diagnosis.actions import DiagnosisActionType
+from dlrover_amd.master.elastic.rdzv_manager import ElasticTrainingRendezvousManager
+from dlrover_amd.master.node.job_context import JobContext
+from dlrover_amd.master.node.job_manager import LocalJobManager
+from dlrover_amd.master.node.status_flow import allowed_transition
+
+
+@pytest.fixture()
+def mgr():
+    JobContext._reset_for_tests()
+    ctx = JobContext.singleton_instance()
+    rdzv = {"elastic-training": ElasticTrainingRendezvousManager()}
+    m = LocalJobManager(job_context=ctx, rdzv_managers=rdzv)
+    yield m
+    JobContext._reset_for_tests()
+
+
+def _meta(nid, status=""):
+    return comm.NodeMeta(type=NodeType.WORKER, id=nid, rank=nid, status=status)
+
+
+def test_status_flow_table():
+    assert allowed_transition(NodeStatus.PENDING, NodeStatus.RUNNING)
+    assert allowed_transition(NodeStatus.RUNNING, NodeStatus.FAILED)
+    assert not allowed_transition(NodeStatus.SUCCEEDED, NodeStatus.RUNNING)
+    assert not allowed_transition(NodeStatus.RUNNING, NodeStatus.RUNNING)
+
+
+def test_software_failure_gets_restart_action(mgr):
+    mgr.on_node_joined(0)
+    mgr.on_node_event(
+        comm.NodeEvent(event_type=NodeEventType.FAILED_EXITED, node=_meta(0))
+    )
+    action = mgr.ctx.next_action(0)
+    assert action is not None
+    assert action.action_type == DiagnosisActionType.RESTART_WORKER
+    # node is kept alive for the restart
+    node = mgr.ctx.get_node(NodeType.WORKER, 0)
+    assert node.status == NodeStatus.RUNNING and node.relaunch_count == 1
+
+
+def test_heartbeat_death_shrinks_not_aborts(mgr):
+    mgr.on_node_joined(0)
+    mgr.on_node_joined(1)
+    node1 = mgr.ctx.get_node(NodeType.WORKER, 1)
+    node1.exit_reason = NodeExitReason.NO_HEARTBEAT
+    mgr._transition(node1, NodeStatus.FAILED)
+    mgr._handle_node_failure(node1, "heartbeat timeout")
+    assert not mgr.ctx.is_stopping()
+    assert not node1.relaunchable
+    # removed from rendezvous liveness
+    assert 1 not in mgr.rdzv_managers["elastic-training"]._alive_nodes
+
+
+def test_last_node_heartbeat_death_stops_job(mgr):
+    mgr.on_node_joined(0)
+    node = mgr.ctx.get_node(NodeType.WORKER, 0)
+    node.exit_reason = NodeExitReason.NO_HEARTBEAT
+    mgr._transition(node, NodeStatus.FAILED)
+    mgr._handle_node_failure(node, "heartbeat timeout")
+    assert mgr.ctx.is_stopping()
+
+
+def test_unrecoverable_with_peers_eliminates_node(mgr):
+    mgr.on_node_joined(0)
+    mgr.on_node_joined(1)
+    node = mgr.ctx.get_node(NodeType.WORKER, 0)
+    node.relaunch_count = node.max_relaunch_count  # budget exhausted
+    mgr._transition(node, NodeStatus.FAILED)
+    mgr._handle_node_failure(node, "still broken")
+    assert not mgr.ctx.is_stopping()
+    assert node.eliminated
+
+
+def test_unrecoverable_alone_aborts(mgr):
+    mgr.on_node_joined(0)
+    node = mgr.ctx.get_node(NodeType.WORKER, 0)
+    node.relaunch_count = node.max_relaunch_count
+    mgr._transition(node, NodeStatus.FAILED)
+    mgr._handle_node_failure(node, "still broken")
+    assert mgr.ctx.is_stopping()
+    assert mgr.ctx.exit_reason == "WorkerError"
+
+
+def test_all_succeed_finishes_job(mgr):
+    for nid in (0, 1):
+        mgr.on_node_joined(nid)
+    for nid in (0, 1):
+        mgr.on_node_event(
+            comm.NodeEvent(event_type=NodeEventType.SUCCEEDED_EXITED, node=_meta(nid))
+        )
+    assert mgr.ctx.is_stopping()
+    assert mgr.ctx.exit_code == 0
+
+
+def test_broadcast_action_fans_out(mgr):
+    from dlrover_amd.diagnosis.actions import NodeAction
+
+    for nid in (0, 1, 2):
+        mgr.on_node_joined(nid)
+    mgr.ctx.enqueue_action(NodeAction(node_id=-1, reason="hang"))
+    got = [mgr.ctx.next_action(nid) is not None for nid in (0, 1, 2)]
+    assert all(got)
