@@ -80,14 +80,23 @@ __device__ __forceinline__ void stage_tile(const short* __restrict__ src,
 
 __global__ void fa_bwd_pre_kernel(const short* __restrict__ dout,
                                   const short* __restrict__ out,
-                                  float* __restrict__ dvec, long long n_rows) {
+                                  float* __restrict__ dvec, int B, int H,
+                                  int S, long long os_b, long long os_h,
+                                  long long os_s, long long ds_b,
+                                  long long ds_h, long long ds_s) {
   const int lane32 = threadIdx.x & 31;
   const long long group0 =
       ((long long)blockIdx.x * blockDim.x + threadIdx.x) >> 5;
   const long long stride = ((long long)gridDim.x * blockDim.x) >> 5;
-  for (long long row = group0; row < n_rows; row += stride) {
-    const short* dp = dout + row * FA_D + lane32 * 4;
-    const short* op = out + row * FA_D + lane32 * 4;
+  const long long rows = (long long)B * H * S;
+  for (long long row = group0; row < rows; row += stride) {
+    const int b = (int)(row / ((long long)H * S));
+    const int h = (int)((row / S) % H);
+    const int sI = (int)(row % S);
+    const short* dp =
+        dout + b * ds_b + h * ds_h + (long long)sI * ds_s + lane32 * 4;
+    const short* op =
+        out + b * os_b + h * os_h + (long long)sI * os_s + lane32 * 4;
     float acc = 0.f;
     short4_t d4 = *reinterpret_cast<const short4_t*>(dp);
     short4_t o4 = *reinterpret_cast<const short4_t*>(op);
@@ -239,9 +248,11 @@ __global__ __launch_bounds__(256) void fa_bwd_dkv_kernel(
     long long ks_b, long long ks_h, long long ks_s,
     long long vs_b, long long vs_h, long long vs_s,
     long long ds_b, long long ds_h, long long ds_s) {
-  // Q/dO row-major fragments read straight from global (L1/L2-resident);
-  // only the TRANSPOSED layouts need LDS. 48 KB/block -> 3 blocks/CU.
+  // measured: LDS-staged row-major Q/dO beats global-cached fragment reads
+  // (1992 vs 2260 us/call) even at occupancy 1 — keep them in LDS
+  __shared__ char q_lds[FA_T * 256];     // Q row-major
   __shared__ char qt_lds[FA_D * 128];    // Q^T
+  __shared__ char do_lds[FA_T * 256];    // dO row-major
   __shared__ char dot_lds[FA_D * 128];   // dO^T
   __shared__ char pt_lds[4 * 16 * 128];  // per-wave P^T [16kv][64q]
   __shared__ char dst_lds[4 * 16 * 128]; // per-wave dS^T [16kv][64q]
@@ -287,8 +298,8 @@ __global__ __launch_bounds__(256) void fa_bwd_dkv_kernel(
   for (int qt = kt; qt < n_qt; ++qt) {
     const short* q_tile = q_head + (long long)(qt * FA_T) * qs_s;
     const short* do_tile = do_head + (long long)(qt * FA_T) * ds_s;
-    stage_tile(q_tile, qs_s, nullptr, qt_lds, tid);
-    stage_tile(do_tile, ds_s, nullptr, dot_lds, tid);
+    stage_tile(q_tile, qs_s, q_lds, qt_lds, tid);
+    stage_tile(do_tile, ds_s, do_lds, dot_lds, tid);
     __syncthreads();
 
     // S^T = K_band @ Q^T ; dP^T = V_band @ dO^T   (both [16kv, 64q])
@@ -299,12 +310,11 @@ __global__ __launch_bounds__(256) void fa_bwd_dkv_kernel(
       acc_dpt[n] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ks = 0; ks < 4; ++ks) {
-        // B[k=dim][n=q] = Q[n][k] / dO[n][k] — read from global (cached)
-        bf16x8 bq = ld_bT_global(q_tile, qs_s, n * 16, sub, ks * 32, quarter);
+        // B[k=dim][n=q] = Q[n][k] (row-major Q) / dO[n][k]
+        bf16x8 bq = ld_bT(q_lds, n * 16, sub, ks * 32, quarter, 256);
         acc_st[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak[ks], bq,
                                                             acc_st[n], 0, 0, 0);
-        bf16x8 bdo =
-            ld_bT_global(do_tile, ds_s, n * 16, sub, ks * 32, quarter);
+        bf16x8 bdo = ld_bT(do_lds, n * 16, sub, ks * 32, quarter, 256);
         acc_dpt[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             av[ks], bdo, acc_dpt[n], 0, 0, 0);
       }
@@ -373,12 +383,14 @@ __global__ void f32_to_bf16_kernel(const float* __restrict__ src,
 
 extern "C" {
 
-void fa_bwd_pre_launch(const void* dout, const void* out, void* dvec,
-                       long long n_rows, hipStream_t stream) {
-  const long long threads = n_rows * 32;
+void fa_bwd_pre_launch(const void* dout, const void* out, void* dvec, int B,
+                       int H, int S, const long long* ost,
+                       const long long* dst, hipStream_t stream) {
+  const long long threads = (long long)B * H * S * 32;
   hipLaunchKernelGGL(fa_bwd_pre_kernel, dim3(grid_capped(threads, 256)),
                      dim3(256), 0, stream, (const short*)dout,
-                     (const short*)out, (float*)dvec, n_rows);
+                     (const short*)out, (float*)dvec, B, H, S, ost[0], ost[1],
+                     ost[2], dst[0], dst[1], dst[2]);
 }
 
 void fa_bwd_dq_launch(const void* q, const void* k, const void* v,

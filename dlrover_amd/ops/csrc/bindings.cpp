@@ -31,7 +31,8 @@ void mfma16_probe_launch(const void*, const void*, void*, void*);
 void flash_attn_fwd_launch(const void*, const void*, const void*, void*,
                            void*, int, int, int, int, float,
                            const long long*, void*);
-void fa_bwd_pre_launch(const void*, const void*, void*, long long, void*);
+void fa_bwd_pre_launch(const void*, const void*, void*, int, int, int,
+                       const long long*, const long long*, void*);
 void fa_bwd_dq_launch(const void*, const void*, const void*, const void*,
                       const void*, const void*, void*, int, int, int, int,
                       float, const long long*, void*);
@@ -259,14 +260,13 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> flash_attn_bwd(
     dout.copy_(dout_in);
   }
   TORCH_CHECK(dout.stride(3) == 1, "dout needs contiguous D");
-  const long long n_rows = (long long)B * H * S;
   auto dvec = at::empty({B, H, S}, q.options().dtype(at::kFloat));
   {
-    // the pre kernel wants row-contiguous [N, D]: build contiguous views
-    auto dof = dout.is_contiguous() ? dout : dout.contiguous();
-    auto of = out.is_contiguous() ? out : out.contiguous();
-    fa_bwd_pre_launch(dof.data_ptr(), of.data_ptr(), dvec.data_ptr(), n_rows,
-                      cur_stream());
+    long long ost[3], dstr[3];
+    pack_strides(ost, out, 0);
+    pack_strides(dstr, dout, 0);
+    fa_bwd_pre_launch(dout.data_ptr(), out.data_ptr(), dvec.data_ptr(), B, H,
+                      S, ost, dstr, cur_stream());
     long long st[12];
     pack_strides(st, q, 0);
     pack_strides(st, k, 3);
