@@ -130,12 +130,8 @@ __global__ __launch_bounds__(256) void flash_attn_fwd_kernel(
       *reinterpret_cast<bf16x8*>(kb + row * 256 + swz(row, col * 2)) = kv8;
       bf16x8 vv8 =
           *reinterpret_cast<const bf16x8*>(vsrc + (long long)row * vs_s + col);
-      // rotate each lane's write order: without this every lane writes the
-      // same (trow&7) class per instruction and 8 lanes collide on one bank
-      // word (PMC: 35% of fwd cycles were SQ_LDS_BANK_CONFLICT stalls)
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int i = (j + lane) & 7;
+      for (int i = 0; i < 8; ++i) {
         const int trow = col + i;  // head-dim index
         const int tcol = row;      // kv position
         *reinterpret_cast<__bf16*>(vb + trow * 128 + swz(trow, tcol * 2)) =

@@ -66,14 +66,10 @@ __device__ __forceinline__ void stage_tile(const short* __restrict__ src,
     if (row_lds)
       *reinterpret_cast<bf16x8*>(row_lds + row * 256 + swz2(row, col * 2)) = v8;
     if (tr_lds) {
-      // per-lane rotated order: spreads the (row&7) swizzle class across
-      // lanes per instruction (bank-conflict fix, see attention.hip)
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int i = (j + tid) & 7;
+      for (int i = 0; i < 8; ++i)
         *reinterpret_cast<__bf16*>(
             tr_lds + (col + i) * 128 + swz2(col + i, row * 2)) = v8[i];
-      }
     }
   }
 }
