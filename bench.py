@@ -157,9 +157,10 @@ def main():
     if cp is not None:
         # size the shm segment + staging outside the timed window (the
         # reference also excludes first-export spin-up, ~20 s: BASELINE.md)
+        # step must be > 0: the shm commit word treats 0 as "empty"
         sd = ckpt_state()
-        sd["step"] = 0
-        cp.engine.save_to_memory(0, sd)
+        sd["step"] = 1
+        cp.engine.save_to_memory(1, sd)
         cp.engine.wait_saving()
     sync()
     dist.barrier()
