@@ -39,6 +39,8 @@ class _Handler(BaseHTTPRequestHandler):
                 self._send(200, json.dumps(self._nodes(master)).encode())
             elif self.path == "/api/metrics":
                 self._send(200, json.dumps(self._metrics(master)).encode())
+            elif self.path.startswith("/api/logs"):
+                self._send(200, json.dumps(self._logs()).encode())
             else:
                 self._send(404, b'{"error": "not found"}')
         except Exception as e:  # noqa: BLE001
@@ -79,6 +81,23 @@ class _Handler(BaseHTTPRequestHandler):
                     "exit_reason": node.exit_reason,
                 }
             )
+        return out
+
+    @staticmethod
+    def _logs(n: int = 100) -> dict:
+        """Tail the master/agent log files (ref: dashboard LogsHandler :229)."""
+        import glob
+        import os
+
+        out = {}
+        log_dir = os.getenv("DLROVER_LOG_DIR", "")
+        if log_dir:
+            for path in sorted(glob.glob(os.path.join(log_dir, "dlrover_*.log")))[-8:]:
+                try:
+                    with open(path, errors="replace") as f:
+                        out[os.path.basename(path)] = f.readlines()[-n:]
+                except OSError:
+                    continue
         return out
 
     @staticmethod

@@ -20,6 +20,8 @@ def parse_args(argv=None):
     p.add_argument("--namespace", default="default")
     p.add_argument("--port_file", default="",
                    help="write the bound port here (standalone launcher handshake)")
+    p.add_argument("--enable_dashboard", action="store_true")
+    p.add_argument("--dashboard_port", type=int, default=0)
     return p.parse_args(argv)
 
 
@@ -40,6 +42,11 @@ def run(args) -> int:
 
         master = LocalJobMaster(port=args.port, service_type=args.service_type)
     master.prepare()
+    dashboard = None
+    if args.enable_dashboard:
+        from dlrover_amd.master.dashboard import Dashboard
+
+        dashboard = Dashboard(master, port=args.dashboard_port).start()
     if args.port_file:
         with open(args.port_file, "w") as f:
             f.write(str(master.port))
@@ -47,6 +54,8 @@ def run(args) -> int:
     try:
         return master.run()
     finally:
+        if dashboard is not None:
+            dashboard.stop()
         master.stop()
 
 
