@@ -15,7 +15,8 @@ def parse_args(argv=None):
                    choices=[PlatformType.LOCAL, PlatformType.KUBERNETES])
     p.add_argument("--port", type=int, default=0)
     p.add_argument("--service_type", default=CommServiceType.TCP,
-                   choices=[CommServiceType.TCP, CommServiceType.HTTP])
+                   choices=[CommServiceType.TCP, CommServiceType.HTTP,
+                            CommServiceType.GRPC])
     p.add_argument("--job_name", default=os.getenv("ELASTIC_JOB_NAME", "dlrover-job"))
     p.add_argument("--namespace", default="default")
     p.add_argument("--port_file", default="",
