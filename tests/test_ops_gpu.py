@@ -148,3 +148,36 @@ def test_llama_tiny_gpu_step():
         first = first or loss.item()
         last = loss.item()
     assert last < first * 0.9, (first, last)
+
+
+def test_fused_adamw_hipgraph_capture():
+    """capture_graph=True must produce the same trajectory as eager once the
+    bias-correction warmup (100 steps) passes and the graph replays."""
+    from dlrover_amd.ops import FusedAdamW
+
+    torch.manual_seed(0)
+    m1 = torch.nn.Linear(64, 64).cuda().bfloat16()
+    m2 = torch.nn.Linear(64, 64).cuda().bfloat16()
+    m2.load_state_dict(m1.state_dict())
+    o1 = FusedAdamW(m1.parameters(), lr=1e-3, weight_decay=0.0)
+    o2 = FusedAdamW(m2.parameters(), lr=1e-3, weight_decay=0.0,
+                    capture_graph=True)
+    x = torch.randn(8, 64, device="cuda", dtype=torch.bfloat16)
+    for i in range(110):
+        for m, o in ((m1, o1), (m2, o2)):
+            loss = m(x).float().pow(2).mean()
+            o.zero_grad()
+            loss.backward()
+            o.step()
+    assert o2._graph is not None, "graph was never captured"
+    # a few replayed steps
+    for i in range(5):
+        for m, o in ((m1, o1), (m2, o2)):
+            loss = m(x).float().pow(2).mean()
+            o.zero_grad()
+            loss.backward()
+            o.step()
+    torch.testing.assert_close(
+        m1.weight.float(), m2.weight.float(), rtol=2e-2, atol=2e-3
+    )
+    assert o2.state[m2.weight]["step"] == 115
