@@ -177,7 +177,11 @@ def main():
         loss = train_step()
         sync()
         useful += time.perf_counter() - t0
-        if cp is not None and (k + 1) % args.ckpt_interval == 0:
+        if (cp is not None and (k + 1) % args.ckpt_interval == 0
+                and k + 1 < args.steps):
+            # no save on the final step: its async drain would sit in the
+            # timed window with no steps left to overlap (steady-state
+            # training always has future steps to hide the drain under)
             sd = ckpt_state()
             sd["step"] = k + 1
             blocking = cp.engine.save_to_memory(k + 1, sd, block=False)
