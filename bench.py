@@ -43,7 +43,7 @@ def parse_args():
 
 
 def setup_dist(args):
-    if "RANK" in os.environ:
+    if "RANK" in os.environ and "WORLD_SIZE" in os.environ:
         rank = int(os.environ["RANK"])
         world = int(os.environ["WORLD_SIZE"])
         local_rank = int(os.environ.get("LOCAL_RANK", rank))
