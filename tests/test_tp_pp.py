@@ -109,6 +109,22 @@ def _worker(rank, port, tmpdir, results):
         grads = [p.grad for p in stage.parameters()]
         assert all(g is not None for g in grads), "missing grads on a stage"
 
+        # ---- 1F1B and GPipe must produce the same gradients (4 microbatches;
+        # the schedules only reorder the backward accumulation) ----
+        micros = [torch.randint(0, cfg.vocab_size, (B, S)) for _ in range(4)]
+        mlabels = [m.clone() for m in micros]
+        for p in stage.parameters():
+            p.grad = None
+        loss_g = runner.train_step(micros, mlabels, schedule="gpipe")
+        grads_g = [p.grad.detach().clone() for p in stage.parameters()]
+        for p in stage.parameters():
+            p.grad = None
+        loss_1 = runner.train_step(micros, mlabels, schedule="1f1b")
+        if groups.is_last_stage:
+            assert torch.allclose(loss_1, loss_g, rtol=1e-5, atol=1e-6)
+        for g1, gg in zip([p.grad for p in stage.parameters()], grads_g):
+            assert torch.allclose(g1, gg, rtol=1e-4, atol=1e-6)
+
         # ---- Megatron-style checkpoint roundtrip ----
         opt = torch.optim.AdamW(stage.parameters(), lr=1e-4)
         opt.step()
