@@ -21,3 +21,13 @@ Top-level subpackages:
 """
 
 __version__ = "0.1.0"
+
+# hang diagnosis: when the elastic agent exports DLROVER_PY_TRACER_DIR,
+# every worker that imports this package gains a SIGUSR2 handler dumping
+# all-thread Python stacks (diagnosis/py_tracer.py)
+import os as _os
+
+if _os.getenv("DLROVER_PY_TRACER_DIR"):
+    from dlrover_amd.diagnosis.py_tracer import maybe_install_from_env
+
+    maybe_install_from_env()

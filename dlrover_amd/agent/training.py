@@ -311,6 +311,7 @@ class ElasticTrainingAgent(LocalElasticAgent):
             if state == WorkerState.HEALTHY:
                 if self._restart_requested.is_set():
                     self._restart_requested.clear()
+                    self._dump_worker_py_stacks()
                     self._save_ckpt_to_storage()
                     # a diagnosis-driven restart is a real failure recovery:
                     # it consumes the restart budget (and bumps
@@ -342,6 +343,27 @@ class ElasticTrainingAgent(LocalElasticAgent):
                 continue
 
             raise RuntimeError(f"unknown worker state {state}")
+
+    def _dump_worker_py_stacks(self):
+        """Before a hang-driven restart kills the workers, capture where each
+        rank's Python main thread is stuck and log the cross-rank aggregate
+        (ref behavior: xpu_timer dump_driver + stack_viewer hang dossier)."""
+        dump_dir = os.getenv("DLROVER_PY_TRACER_DIR", "")
+        if not dump_dir:
+            return
+        try:
+            pids = dict(self._pcontext.pids()) if self._pcontext else {}
+            if not pids:
+                return
+            from dlrover_amd.diagnosis import py_tracer
+
+            stacks = py_tracer.dump_worker_stacks(pids, dump_dir)
+            logger.info(
+                "worker python stacks at hang:\n%s",
+                py_tracer.aggregate_stacks(stacks),
+            )
+        except Exception as e:  # noqa: BLE001 — diagnosis must not block recovery
+            logger.warning("python stack dump failed: %s", e)
 
     def _report_failures(self, run_result: RunResult) -> str:
         errs = {}
@@ -416,6 +438,13 @@ def launch_agent(
         config.max_nodes,
         config.waiting_timeout,
         config.node_unit,
+    )
+
+    # workers auto-register a SIGUSR2 python-stack dumper on import
+    # (diagnosis/py_tracer.py); the agent signals them on hang-restart
+    os.environ.setdefault(
+        "DLROVER_PY_TRACER_DIR",
+        f"/tmp/dlrover_py_tracer_{os.getenv('ELASTIC_JOB_NAME', 'job')}",
     )
 
     ipc_server = IPCServer().start()
