@@ -121,5 +121,15 @@ class HiptimerCollector:
                     self._client.report_diagnosis_data("hang", json.dumps(state))
                 if state.get("hang"):
                     logger.warning("hiptimer: node reports GPU hang: %s", state)
+                    # name the communicators involved (rccl_env introspection)
+                    from dlrover_amd.xpu_timer.rccl_env import format_comm_report
+
+                    for rank, metrics in sorted(self.snapshot().items()):
+                        report = format_comm_report(metrics)
+                        if report:
+                            logger.warning(
+                                "hiptimer: rank %s communicators:\n%s",
+                                rank, report,
+                            )
             except Exception:  # noqa: BLE001
                 logger.exception("hiptimer collector iteration failed")

@@ -164,6 +164,38 @@ class Manager {
     pending_.push_back({start, stop, cat, bytes, now(), name});
   }
 
+  // ---- RCCL communicator introspection (ref behavior: the reference's
+  // nccl_parser extracts comm world size/rank + per-comm traffic so the
+  // diagnostician can name WHICH process group a hang or imbalance is in) --
+  struct CommInfo {
+    int nranks = 0;
+    int rank = 0;
+    long calls = 0;
+    double bytes = 0;
+    bool alive = true;
+  };
+
+  void register_comm(void* c, int nranks, int rank) {
+    std::lock_guard<std::mutex> g(comm_mu_);
+    CommInfo& ci = comms_[c];
+    ci.nranks = nranks;
+    ci.rank = rank;
+    ci.alive = true;
+  }
+
+  void unregister_comm(void* c) {
+    std::lock_guard<std::mutex> g(comm_mu_);
+    auto it = comms_.find(c);
+    if (it != comms_.end()) it->second.alive = false;  // keep stats visible
+  }
+
+  void count_comm_call(void* c, double bytes) {
+    std::lock_guard<std::mutex> g(comm_mu_);
+    CommInfo& ci = comms_[c];
+    ci.calls += 1;
+    ci.bytes += bytes;
+  }
+
   void count_alloc(long bytes) { alloc_bytes_ += bytes; }
   void count_free() { free_count_ += 1; }
   void count_host_alloc(long bytes) { host_alloc_bytes_ += bytes; }
@@ -332,6 +364,21 @@ class Manager {
                 nm.c_str(), kv.second.second);
       }
     }
+    {
+      std::lock_guard<std::mutex> g(comm_mu_);
+      for (auto& kv : comms_) {
+        fprintf(f,
+                "hiptimer_comm_calls{comm=\"%p\",nranks=\"%d\",rank=\"%d\","
+                "alive=\"%d\"} %ld\n",
+                kv.first, kv.second.nranks, kv.second.rank,
+                kv.second.alive ? 1 : 0, kv.second.calls);
+        fprintf(f,
+                "hiptimer_comm_elems{comm=\"%p\",nranks=\"%d\",rank=\"%d\","
+                "alive=\"%d\"} %.0f\n",
+                kv.first, kv.second.nranks, kv.second.rank,
+                kv.second.alive ? 1 : 0, kv.second.bytes);
+      }
+    }
     fprintf(f, "hiptimer_device_alloc_bytes %.0f\n", (double)alloc_bytes_.load());
     fprintf(f, "hiptimer_device_free_total %ld\n", free_count_.load());
     fprintf(f, "hiptimer_host_alloc_bytes %.0f\n",
@@ -353,6 +400,8 @@ class Manager {
   std::vector<hipEvent_t> pool_;
   std::mutex q_mu_;
   std::deque<PendingOp> pending_;
+  std::mutex comm_mu_;
+  std::map<void*, CommInfo> comms_;
   CatStats stats_[CAT_COUNT];
   std::map<std::string, std::pair<long, double>> kernel_stats_;
   std::atomic<long> launched_{0};
@@ -472,11 +521,50 @@ int hipblasLtMatmul(void* handle, void* matmulDesc, const void* alpha,
     using fn_t = ncclResult_t (*)(const void*, void*, size_t, ncclDataType_t,  \
                                   ncclRedOp_t, ncclComm_t, hipStream_t);       \
     static fn_t fn = (fn_t)real(#NAME);                                        \
+    Manager::inst().count_comm_call(comm, (double)(COUNT_EXPR));               \
     hiptimer::Scoped sc(stream, CAT_COMM, (double)(COUNT_EXPR));               \
     ncclResult_t rc = fn(sendbuff, recvbuff, count, dt, op, comm, stream);     \
     sc.finish();                                                               \
     return rc;                                                                 \
   }
+
+// comm lifecycle: record {comm -> (nranks, rank)} so metrics can attribute
+// traffic to a specific process group (DP vs TP vs PP)
+ncclResult_t ncclCommInitRank(ncclComm_t* comm, int nranks,
+                              ncclUniqueId_dummy commId, int rank) {
+  using fn_t = ncclResult_t (*)(ncclComm_t*, int, ncclUniqueId_dummy, int);
+  static fn_t fn = (fn_t)real("ncclCommInitRank");
+  ncclResult_t rc = fn(comm, nranks, commId, rank);
+  if (rc == 0 && comm != nullptr)
+    Manager::inst().register_comm(*comm, nranks, rank);
+  return rc;
+}
+
+ncclResult_t ncclCommInitRankConfig(ncclComm_t* comm, int nranks,
+                                    ncclUniqueId_dummy commId, int rank,
+                                    void* config) {
+  using fn_t =
+      ncclResult_t (*)(ncclComm_t*, int, ncclUniqueId_dummy, int, void*);
+  static fn_t fn = (fn_t)real("ncclCommInitRankConfig");
+  ncclResult_t rc = fn(comm, nranks, commId, rank, config);
+  if (rc == 0 && comm != nullptr)
+    Manager::inst().register_comm(*comm, nranks, rank);
+  return rc;
+}
+
+ncclResult_t ncclCommDestroy(ncclComm_t comm) {
+  using fn_t = ncclResult_t (*)(ncclComm_t);
+  static fn_t fn = (fn_t)real("ncclCommDestroy");
+  Manager::inst().unregister_comm(comm);
+  return fn(comm);
+}
+
+ncclResult_t ncclCommAbort(ncclComm_t comm) {
+  using fn_t = ncclResult_t (*)(ncclComm_t);
+  static fn_t fn = (fn_t)real("ncclCommAbort");
+  Manager::inst().unregister_comm(comm);
+  return fn(comm);
+}
 
 HIPTIMER_NCCL_COLL(ncclAllReduce, count)
 HIPTIMER_NCCL_COLL(ncclReduce, count)
@@ -487,6 +575,7 @@ ncclResult_t ncclAllGather(const void* sendbuff, void* recvbuff,
   using fn_t = ncclResult_t (*)(const void*, void*, size_t, ncclDataType_t,
                                 ncclComm_t, hipStream_t);
   static fn_t fn = (fn_t)real("ncclAllGather");
+  Manager::inst().count_comm_call(comm, (double)sendcount);
   hiptimer::Scoped sc(stream, CAT_COMM, (double)sendcount);
   ncclResult_t rc = fn(sendbuff, recvbuff, sendcount, dt, comm, stream);
   sc.finish();
@@ -500,6 +589,7 @@ ncclResult_t ncclReduceScatter(const void* sendbuff, void* recvbuff,
   using fn_t = ncclResult_t (*)(const void*, void*, size_t, ncclDataType_t,
                                 ncclRedOp_t, ncclComm_t, hipStream_t);
   static fn_t fn = (fn_t)real("ncclReduceScatter");
+  Manager::inst().count_comm_call(comm, (double)recvcount);
   hiptimer::Scoped sc(stream, CAT_COMM, (double)recvcount);
   ncclResult_t rc = fn(sendbuff, recvbuff, recvcount, dt, op, comm, stream);
   sc.finish();
@@ -512,6 +602,7 @@ ncclResult_t ncclBroadcast(const void* sendbuff, void* recvbuff, size_t count,
   using fn_t = ncclResult_t (*)(const void*, void*, size_t, ncclDataType_t,
                                 int, ncclComm_t, hipStream_t);
   static fn_t fn = (fn_t)real("ncclBroadcast");
+  Manager::inst().count_comm_call(comm, (double)count);
   hiptimer::Scoped sc(stream, CAT_COMM, (double)count);
   ncclResult_t rc = fn(sendbuff, recvbuff, count, dt, root, comm, stream);
   sc.finish();
@@ -524,6 +615,7 @@ ncclResult_t ncclSend(const void* sendbuff, size_t count, ncclDataType_t dt,
       ncclResult_t (*)(const void*, size_t, ncclDataType_t, int, ncclComm_t,
                        hipStream_t);
   static fn_t fn = (fn_t)real("ncclSend");
+  Manager::inst().count_comm_call(comm, (double)count);
   hiptimer::Scoped sc(stream, CAT_COMM, (double)count);
   ncclResult_t rc = fn(sendbuff, count, dt, peer, comm, stream);
   sc.finish();
@@ -535,6 +627,7 @@ ncclResult_t ncclRecv(void* recvbuff, size_t count, ncclDataType_t dt, int peer,
   using fn_t = ncclResult_t (*)(void*, size_t, ncclDataType_t, int, ncclComm_t,
                                 hipStream_t);
   static fn_t fn = (fn_t)real("ncclRecv");
+  Manager::inst().count_comm_call(comm, (double)count);
   hiptimer::Scoped sc(stream, CAT_COMM, (double)count);
   ncclResult_t rc = fn(recvbuff, count, dt, peer, comm, stream);
   sc.finish();

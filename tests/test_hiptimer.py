@@ -72,3 +72,33 @@ def test_hiptimer_intercepts_real_kernels(tmp_path):
     assert m.get("hiptimer_launched_total", 0) > 0
     assert m.get('hiptimer_op_count{cat="kernel"}', 0) > 0
     assert m.get("XPU_TIMER_COMMON_HANG") == 0
+
+
+def test_rccl_env_comm_summary():
+    from dlrover_amd.xpu_timer import rccl_env
+
+    metrics = {
+        'hiptimer_comm_calls{comm="0xa",nranks="8",rank="3",alive="1"}': 120.0,
+        'hiptimer_comm_elems{comm="0xa",nranks="8",rank="3",alive="1"}': 4.0e9,
+        'hiptimer_comm_calls{comm="0xb",nranks="2",rank="1",alive="0"}': 7.0,
+        'hiptimer_comm_elems{comm="0xb",nranks="2",rank="1",alive="0"}': 1.0e6,
+        "hiptimer_launched_total": 99.0,
+    }
+    comms = rccl_env.summarize_comms(metrics)
+    assert [c["comm"] for c in comms] == ["0xa", "0xb"]  # traffic-sorted
+    assert comms[0]["nranks"] == 8 and comms[0]["rank"] == 3 and comms[0]["alive"]
+    assert comms[1]["calls"] == 7.0 and not comms[1]["alive"]
+    report = rccl_env.format_comm_report(metrics)
+    assert "nranks=8" in report and "destroyed" in report
+    assert rccl_env.format_comm_report({"x": 1.0}) is None
+
+
+def test_rccl_env_bandwidth_expectations(monkeypatch):
+    from dlrover_amd.xpu_timer import rccl_env
+
+    assert rccl_env.expected_ring_busbw_gbps() == 153.0
+    # all-reduce algbw bound: busbw * n / (2(n-1)) -> 8/14 of link bw at n=8
+    assert abs(rccl_env.expected_allreduce_algbw_gbps(8) - 153.0 * 8 / 14) < 1e-9
+    monkeypatch.setenv("NCCL_MIN_NCHANNELS", "4")
+    env = rccl_env.effective_env()
+    assert env["NCCL_MIN_NCHANNELS"] == "4"
