@@ -30,7 +30,7 @@ import torch.distributed as dist
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=8, help="timed steps (K)")
+    p.add_argument("--steps", type=int, default=12, help="timed steps (K)")
     p.add_argument("--warmup", type=int, default=3, help="untimed steps (W)")
     p.add_argument("--model", default="llama3_8b", choices=["llama3_8b", "small_1b", "tiny"])
     p.add_argument("--batch", type=int, default=1, help="per-GPU micro batch")
