@@ -18,40 +18,59 @@ __global__ void adamw_kernel(
     short* __restrict__ param_bf16, long long numel, float lr, float beta1,
     float beta2, float eps, float weight_decay, float bc1, float bc2,
     float grad_scale) {
-  const long long vecs = numel >> 2;
+  // 8 elements (2 independent b128 chains per array) per thread-iteration:
+  // one chain alone leaves the memory pipes underfed on HBM3E — this loop is
+  // pure bandwidth (28 B/param) and needs the outstanding-load depth
+  const long long vecs = numel >> 3;
   for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
        i < vecs; i += (long long)gridDim.x * blockDim.x) {
-    f32x4 p = *reinterpret_cast<f32x4*>(param + i * 4);
-    f32x4 m = *reinterpret_cast<f32x4*>(exp_avg + i * 4);
-    f32x4 v = *reinterpret_cast<f32x4*>(exp_avg_sq + i * 4);
-    float g[4];
+    f32x4 p[2], m[2], v[2];
+    float g[8];
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      p[h] = *reinterpret_cast<f32x4*>(param + i * 8 + h * 4);
+      m[h] = *reinterpret_cast<f32x4*>(exp_avg + i * 8 + h * 4);
+      v[h] = *reinterpret_cast<f32x4*>(exp_avg_sq + i * 8 + h * 4);
+    }
     if (GRAD_BF16) {
-      short4_t gv = *reinterpret_cast<const short4_t*>(
-          (const short*)grad_in + i * 4);
+      short8_t gv = *reinterpret_cast<const short8_t*>(
+          (const short*)grad_in + i * 8);
 #pragma unroll
-      for (int j = 0; j < 4; ++j) g[j] = bf2f(gv[j]) * grad_scale;
+      for (int j = 0; j < 8; ++j) g[j] = bf2f(gv[j]) * grad_scale;
     } else {
-      f32x4 gv = *reinterpret_cast<const f32x4*>((const float*)grad_in + i * 4);
 #pragma unroll
-      for (int j = 0; j < 4; ++j) g[j] = gv[j] * grad_scale;
-    }
-    short4_t pb;
+      for (int h = 0; h < 2; ++h) {
+        f32x4 gv =
+            *reinterpret_cast<const f32x4*>((const float*)grad_in + i * 8 + h * 4);
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      m[j] = beta1 * m[j] + (1.f - beta1) * g[j];
-      v[j] = beta2 * v[j] + (1.f - beta2) * g[j] * g[j];
-      const float mhat = m[j] * bc1;
-      const float vhat = v[j] * bc2;
-      p[j] -= lr * (mhat / (sqrtf(vhat) + eps) + weight_decay * p[j]);
-      pb[j] = f2bf(p[j]);
+        for (int j = 0; j < 4; ++j) g[h * 4 + j] = gv[j] * grad_scale;
+      }
     }
-    *reinterpret_cast<f32x4*>(param + i * 4) = p;
-    *reinterpret_cast<f32x4*>(exp_avg + i * 4) = m;
-    *reinterpret_cast<f32x4*>(exp_avg_sq + i * 4) = v;
-    if (param_bf16) *reinterpret_cast<short4_t*>(param_bf16 + i * 4) = pb;
+    short8_t pb;
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const float gj = g[h * 4 + j];
+        m[h][j] = beta1 * m[h][j] + (1.f - beta1) * gj;
+        v[h][j] = beta2 * v[h][j] + (1.f - beta2) * gj * gj;
+        const float mhat = m[h][j] * bc1;
+        const float vhat = v[h][j] * bc2;
+        p[h][j] -= lr * (mhat / (sqrtf(vhat) + eps) + weight_decay * p[h][j]);
+        pb[h * 4 + j] = f2bf(p[h][j]);
+      }
+    }
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      *reinterpret_cast<f32x4*>(param + i * 8 + h * 4) = p[h];
+      *reinterpret_cast<f32x4*>(exp_avg + i * 8 + h * 4) = m[h];
+      *reinterpret_cast<f32x4*>(exp_avg_sq + i * 8 + h * 4) = v[h];
+    }
+    if (param_bf16)
+      *reinterpret_cast<short8_t*>(param_bf16 + i * 8) = pb;
   }
-  // scalar tail (numel % 4)
-  const long long tail0 = vecs * 4;
+  // scalar tail (numel % 8)
+  const long long tail0 = vecs * 8;
   for (long long i = tail0 + (long long)blockIdx.x * blockDim.x + threadIdx.x;
        i < numel; i += (long long)gridDim.x * blockDim.x) {
     float g = GRAD_BF16 ? bf2f(((const short*)grad_in)[i])
@@ -75,7 +94,7 @@ extern "C" void adamw_launch(void* param, const void* grad, void* exp_avg,
                   float grad_scale, hipStream_t stream) {
   const float bc1 = 1.f / (1.f - powf(beta1, (float)step));
   const float bc2 = 1.f / (1.f - powf(beta2, (float)step));
-  const int grid = grid_capped(numel >> 2, 256);
+  const int grid = grid_capped(numel >> 3, 256);
   if (grad_is_bf16) {
     hipLaunchKernelGGL((adamw_kernel<true>), dim3(grid), dim3(256), 0, stream,
                        (float*)param, grad, (float*)exp_avg,
