@@ -14,8 +14,8 @@ void rmsnorm_fwd_launch(const void*, const void*, void*, void*, int, int,
 void rmsnorm_add_fwd_launch(const void*, const void*, const void*, void*,
                             void*, void*, int, int, float, void*);
 void rmsnorm_bwd_launch(const void*, const void*, const void*, const void*,
-                        const void*, void*, void*, void*, int, int, int,
-                        void*);
+                        const void*, void*, void*, void*, void*, int, int,
+                        int, void*);
 void rope_launch(void*, const void*, const void*, const void*, long long,
                  long long, int, int, int, void*);
 void swiglu_fwd_launch(const void*, void*, long long, int, void*);
@@ -95,11 +95,14 @@ std::tuple<at::Tensor, at::Tensor> rmsnorm_bwd(
   TORCH_CHECK(hidden <= 16384, "rmsnorm_bwd: hidden > 16384 unsupported "
               "(register dw accumulator: RMSN_MAX_VPT)");
   const long long n_rows = x.numel() / hidden;
-  const int n_partials = 64;
+  // one private slice per block (plain stores, no atomics): n_partials must
+  // match the kernel grid = min(n_rows, 2048)
+  const int n_partials = (int)(n_rows < 2048 ? (n_rows > 0 ? n_rows : 1) : 2048);
   auto dx = at::empty_like(x);
   auto dw = at::empty_like(w);
   auto partial =
       at::empty({n_partials, hidden}, x.options().dtype(at::kFloat));
+  auto dw_f32 = at::zeros({hidden}, x.options().dtype(at::kFloat));
   const void* dh_ptr = nullptr;
   if (dh_extra.has_value()) {
     check_bf16(*dh_extra, "dh_extra");
@@ -107,8 +110,8 @@ std::tuple<at::Tensor, at::Tensor> rmsnorm_bwd(
   }
   rmsnorm_bwd_launch(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
                      invrms.data_ptr(), dh_ptr, dx.data_ptr(),
-                     partial.data_ptr(), dw.data_ptr(), (int)n_rows, hidden,
-                     n_partials, cur_stream());
+                     partial.data_ptr(), dw_f32.data_ptr(), dw.data_ptr(),
+                     (int)n_rows, hidden, n_partials, cur_stream());
   return {dx, dw};
 }
 

@@ -53,22 +53,40 @@ __global__ void rmsnorm_fwd_kernel(
 // ran at 0.65 TB/s (profiles/r01_small1b_kernel_stats.txt); this removes
 // n_rows*H atomics down to gridDim*H.
 #define RMSN_MAX_VPT 8  // vecs/thread: supports hidden up to 8*8*256 = 16384
+#define RMSN_CACHE_VPT 2  // dy/x register-cached when vecs <= 2*blockDim
 
-__global__ void rmsnorm_bwd_kernel(
+}  // extern "C" — the templated bwd kernel needs C++ linkage
+
+// CACHE=true (hidden <= 8*2*256 = 4096, the Llama case): pass 1's dy/x loads
+// are kept in registers so pass 2 issues NO global reads except dh_extra —
+// the old re-read version measured 257 us vs a ~40 us traffic bound
+// (profiles/r01g). Every block owns its own dw_partial slice (n_partials ==
+// gridDim, set host-side) and fully overwrites it with plain stores: the old
+// 64-slice atomicAdd flush was ~8M contended RMW ops per call.
+template <bool CACHE>
+__global__ __launch_bounds__(256) void rmsnorm_bwd_kernel(
     const short* __restrict__ dy, const short* __restrict__ x,
     const short* __restrict__ w, const float* __restrict__ invrms,
     const short* __restrict__ dh_extra,  // nullable: fused-add path's
                                          // residual-stream gradient
     short* __restrict__ dx, float* __restrict__ dw_partial,
     int n_rows, int hidden, int n_partials) {
+  // ALL per-thread arrays are indexed only by compile-time unrolled vi —
+  // runtime indexing would demote them to scratch (guide rule #20), which
+  // is what held the previous version to ~1.4 TB/s
+  constexpr int BD = 256;
+  constexpr int VPT = CACHE ? RMSN_CACHE_VPT : RMSN_MAX_VPT;
   __shared__ float scratch[16];
   const int vecs = hidden >> 3;
-  float* dwp = dw_partial + (long long)(blockIdx.x % n_partials) * hidden;
-  float dw_acc[RMSN_MAX_VPT][8];
+  float* dwp = dw_partial + (long long)blockIdx.x * hidden;
+  float dw_acc[VPT][8];
 #pragma unroll
-  for (int i = 0; i < RMSN_MAX_VPT; ++i)
+  for (int i = 0; i < VPT; ++i)
 #pragma unroll
     for (int j = 0; j < 8; ++j) dw_acc[i][j] = 0.f;
+
+  float dy_c[CACHE ? RMSN_CACHE_VPT : 1][8];
+  float x_c[CACHE ? RMSN_CACHE_VPT : 1][8];
 
   for (int row = blockIdx.x; row < n_rows; row += gridDim.x) {
     const short* dyr = dy + (long long)row * hidden;
@@ -77,57 +95,95 @@ __global__ void rmsnorm_bwd_kernel(
     const float inv = invrms[row];
     // pass 1: dot = sum(dy * w * x)
     float dot = 0.f;
-    for (int v = threadIdx.x; v < vecs; v += blockDim.x) {
-      float dyv[8], wv[8], xv[8];
-      load8(dyr + v * 8, dyv);
-      load8(w + v * 8, wv);
-      load8(xr + v * 8, xv);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) dot += dyv[j] * wv[j] * xv[j];
+    for (int vi = 0; vi < VPT; ++vi) {
+      const int v = threadIdx.x + vi * BD;
+      if (v < vecs) {
+        float dyv[8], wv[8], xv[8];
+        load8(dyr + v * 8, dyv);
+        load8(w + v * 8, wv);
+        load8(xr + v * 8, xv);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          dot += dyv[j] * wv[j] * xv[j];
+          if (CACHE) {
+            dy_c[vi][j] = dyv[j];
+            x_c[vi][j] = xv[j];
+          }
+        }
+      }
     }
     dot = block_reduce_sum(dot, scratch);
     const float k = dot * inv * inv * inv / hidden;
-    // pass 2: dx + register dw accumulation
-    int vi = 0;
-    for (int v = threadIdx.x; v < vecs; v += blockDim.x, ++vi) {
-      float dyv[8], wv[8], xv[8], out[8];
-      load8(dyr + v * 8, dyv);
-      load8(w + v * 8, wv);
-      load8(xr + v * 8, xv);
+    // pass 2: dx + register dw accumulation (no dy/x re-read when CACHE)
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        out[j] = dyv[j] * wv[j] * inv - xv[j] * k;
-        dw_acc[vi][j] += dyv[j] * xv[j] * inv;
-      }
-      if (dh_extra) {
-        float dhv[8];
-        load8(dh_extra + (long long)row * hidden + v * 8, dhv);
+    for (int vi = 0; vi < VPT; ++vi) {
+      const int v = threadIdx.x + vi * BD;
+      if (v < vecs) {
+        float dyv[8], wv[8], xv[8], out[8];
+        if (CACHE) {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) out[j] += dhv[j];
+          for (int j = 0; j < 8; ++j) {
+            dyv[j] = dy_c[vi][j];
+            xv[j] = x_c[vi][j];
+          }
+        } else {
+          load8(dyr + v * 8, dyv);
+          load8(xr + v * 8, xv);
+        }
+        load8(w + v * 8, wv);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          out[j] = dyv[j] * wv[j] * inv - xv[j] * k;
+          dw_acc[vi][j] += dyv[j] * xv[j] * inv;
+        }
+        if (dh_extra) {
+          float dhv[8];
+          load8(dh_extra + (long long)row * hidden + v * 8, dhv);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) out[j] += dhv[j];
+        }
+        store8(dxr + v * 8, out);
       }
-      store8(dxr + v * 8, out);
     }
     __syncthreads();
   }
-  // flush: one atomicAdd per column per block into this block's partial slice
-  int vi = 0;
-  for (int v = threadIdx.x; v < vecs; v += blockDim.x, ++vi) {
+  // flush: this block's private slice, plain b128 stores
 #pragma unroll
-    for (int j = 0; j < 8; ++j) atomicAdd(&dwp[v * 8 + j], dw_acc[vi][j]);
+  for (int vi = 0; vi < VPT; ++vi) {
+    const int v = threadIdx.x + vi * BD;
+    if (v < vecs) {
+      float4_t* dst = reinterpret_cast<float4_t*>(dwp + v * 8);
+      dst[0] = {dw_acc[vi][0], dw_acc[vi][1], dw_acc[vi][2], dw_acc[vi][3]};
+      dst[1] = {dw_acc[vi][4], dw_acc[vi][5], dw_acc[vi][6], dw_acc[vi][7]};
+    }
   }
 }
 
-// reduce [n_partials, H] fp32 -> dw [H] bf16
+// stage 1: [n_partials, H] fp32 -> dw_f32 [H] via <=16 atomics per column
+// (blockIdx.y picks a partial range; dw_f32 is zeroed host-side)
 __global__ void reduce_partials_kernel(
-    const float* __restrict__ partials, short* __restrict__ out,
+    const float* __restrict__ partials, float* __restrict__ dw_f32,
     int n_partials, int hidden) {
-  for (int col = blockIdx.x * blockDim.x + threadIdx.x; col < hidden;
-       col += gridDim.x * blockDim.x) {
-    float acc = 0.f;
-    for (int p = 0; p < n_partials; ++p)
-      acc += partials[(long long)p * hidden + col];
-    out[col] = f2bf(acc);
-  }
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= hidden) return;
+  const int per = (n_partials + gridDim.y - 1) / gridDim.y;
+  const int p0 = blockIdx.y * per;
+  const int p1 = min(p0 + per, n_partials);
+  float acc = 0.f;
+  for (int p = p0; p < p1; ++p)
+    acc += partials[(long long)p * hidden + col];
+  if (gridDim.y == 1)
+    dw_f32[col] = acc;
+  else
+    atomicAdd(&dw_f32[col], acc);
+}
+
+// stage 2: fp32 -> bf16
+__global__ void dw_cast_kernel(const float* __restrict__ dw_f32,
+                               short* __restrict__ out, int hidden) {
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col < hidden) out[col] = f2bf(dw_f32[col]);
 }
 
 // Fused residual-add + RMSNorm: h = x + resid; y = h * invrms(h) * w.
@@ -184,8 +240,9 @@ extern "C" void rmsnorm_add_fwd_launch(const void* x, const void* resid,
                      eps);
 }
 
-void rmsnorm_fwd_launch(const void* x, const void* w, void* y, void* invrms,
-                        int n_rows, int hidden, float eps, hipStream_t stream) {
+extern "C" void rmsnorm_fwd_launch(const void* x, const void* w, void* y,
+                                   void* invrms, int n_rows, int hidden,
+                                   float eps, hipStream_t stream) {
   int grid = n_rows < 2048 ? n_rows : 2048;
   if (grid < 1) grid = 1;
   hipLaunchKernelGGL(rmsnorm_fwd_kernel, dim3(grid), dim3(256), 0, stream,
@@ -193,21 +250,39 @@ void rmsnorm_fwd_launch(const void* x, const void* w, void* y, void* invrms,
                      (float*)invrms, n_rows, hidden, eps);
 }
 
+extern "C" {
+
+// n_partials MUST equal the launch grid (each block owns one slice);
+// dw_f32 is a zero-init [hidden] fp32 scratch for the two-stage reduce.
 void rmsnorm_bwd_launch(const void* dy, const void* x, const void* w,
                         const void* invrms, const void* dh_extra, void* dx,
-                        void* dw_partial, void* dw, int n_rows, int hidden,
-                        int n_partials, hipStream_t stream) {
+                        void* dw_partial, void* dw_f32, void* dw, int n_rows,
+                        int hidden, int n_partials, hipStream_t stream) {
   int grid = n_rows < 2048 ? n_rows : 2048;
   if (grid < 1) grid = 1;
-  hipMemsetAsync(dw_partial, 0, (size_t)n_partials * hidden * sizeof(float),
-                 stream);
-  hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(grid), dim3(256), 0, stream,
-                     (const short*)dy, (const short*)x, (const short*)w,
-                     (const float*)invrms, (const short*)dh_extra, (short*)dx,
-                     (float*)dw_partial, n_rows, hidden, n_partials);
-  hipLaunchKernelGGL(reduce_partials_kernel, dim3(grid_capped(hidden, 256)),
-                     dim3(256), 0, stream, (const float*)dw_partial,
-                     (short*)dw, n_partials, hidden);
+  const int vecs = hidden >> 3;
+  if (vecs <= RMSN_CACHE_VPT * 256) {
+    hipLaunchKernelGGL((rmsnorm_bwd_kernel<true>), dim3(grid), dim3(256), 0,
+                       stream, (const short*)dy, (const short*)x,
+                       (const short*)w, (const float*)invrms,
+                       (const short*)dh_extra, (short*)dx, (float*)dw_partial,
+                       n_rows, hidden, n_partials);
+  } else {
+    hipLaunchKernelGGL((rmsnorm_bwd_kernel<false>), dim3(grid), dim3(256), 0,
+                       stream, (const short*)dy, (const short*)x,
+                       (const short*)w, (const float*)invrms,
+                       (const short*)dh_extra, (short*)dx, (float*)dw_partial,
+                       n_rows, hidden, n_partials);
+  }
+  int ny = (n_partials + 255) / 256;
+  if (ny > 16) ny = 16;
+  if (ny < 1) ny = 1;
+  hipLaunchKernelGGL(reduce_partials_kernel,
+                     dim3((hidden + 255) / 256, ny), dim3(256), 0, stream,
+                     (const float*)dw_partial, (float*)dw_f32, n_partials,
+                     hidden);
+  hipLaunchKernelGGL(dw_cast_kernel, dim3((hidden + 255) / 256), dim3(256), 0,
+                     stream, (const float*)dw_f32, (short*)dw, hidden);
 }
 
 }  // extern "C"
