@@ -28,9 +28,13 @@ void causal_softmax_bwd_launch(void*, const void*, long long, int, float,
 void cross_entropy_launch(void*, const void*, void*, long long, int, int,
                           float, int, void*);
 void mfma16_probe_launch(const void*, const void*, void*, void*);
+void tr_b16_probe_launch(void*, void*);
 void flash_attn_fwd_launch(const void*, const void*, const void*, void*,
                            void*, int, int, int, int, float,
                            const long long*, void*);
+void flash_attn_fwd_v2_launch(const void*, const void*, const void*, void*,
+                              void*, int, int, int, int, float,
+                              const long long*, void*);
 void fa_bwd_pre_launch(const void*, const void*, void*, int, int, int,
                        const long long*, const long long*, void*);
 void fa_bwd_dq_launch(const void*, const void*, const void*, const void*,
@@ -238,9 +242,21 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
   pack_strides(st, k, 3);
   pack_strides(st, v, 6);
   pack_strides(st, out, 9);
-  flash_attn_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(),
-                        out_bshd.data_ptr(), lse.data_ptr(), B, H, HKV, S,
-                        (float)scale, st, cur_stream());
+  // v2 = ds_read_b64_tr_b16 V path (see attention.hip); DLROVER_FA_V1=1
+  // falls back to the swizzled-V^T kernel for A/B comparison
+  static const bool use_v1 = []() {
+    const char* e = getenv("DLROVER_FA_V1");
+    return e != nullptr && e[0] == '1';
+  }();
+  if (use_v1) {
+    flash_attn_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                          out_bshd.data_ptr(), lse.data_ptr(), B, H, HKV, S,
+                          (float)scale, st, cur_stream());
+  } else {
+    flash_attn_fwd_v2_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                             out_bshd.data_ptr(), lse.data_ptr(), B, H, HKV,
+                             S, (float)scale, st, cur_stream());
+  }
   return {out, lse};
 }
 
@@ -360,6 +376,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "in-place softmax backward (bf16)");
   m.def("cross_entropy_fwd_bwd", &cross_entropy_fwd_bwd,
         "fused CE loss + in-place dlogits (bf16)");
+  m.def("tr_b16_probe", []() {
+    auto out = at::empty({64, 8}, at::TensorOptions()
+                                      .dtype(at::kBFloat16)
+                                      .device(at::kCUDA));
+    tr_b16_probe_launch(out.data_ptr(), cur_stream());
+    return out;
+  }, "ds_read_b64_tr_b16 semantics probe");
   m.def("mfma16_probe", &mfma16_probe, "MFMA 16x16x32 bf16 layout self-test");
   m.def("flash_attn_fwd", &flash_attn_fwd,
         "flash attention forward (bf16, causal, GQA, D=128) -> (out, lse)");

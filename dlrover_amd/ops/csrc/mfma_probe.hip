@@ -40,3 +40,42 @@ extern "C" void mfma16_probe_launch(const void* A, const void* B, void* C,
   hipLaunchKernelGGL(mfma16_probe_kernel, dim3(1), dim3(64), 0, stream,
                      (const short*)A, (const short*)B, (float*)C);
 }
+
+// ---- ds_read_b64_tr_b16 semantics probe -------------------------------------
+//
+// Guide-documented mapping (cdna_hip_programming.md m156/m162): each lane
+// passes its own LDS byte address A and receives 4 bf16 elements from
+// A + {0, 32, 64, 96} bytes (a strided gather = a free 4x4 transpose when
+// the source is a row-major [4][16] bf16 subtile); `offset:N` is additive.
+// This probe fills LDS with value(i) = i (as bf16), has lane l read at
+// A = 2*l with offset 0 and offset 128, and writes what each lane saw so the
+// host can verify elem j == l + 16*j (and +64 for the offset immediate).
+__global__ void tr_b16_probe_kernel(short* __restrict__ out) {
+  __shared__ short lds[512];
+  const int lane = threadIdx.x & 63;
+  for (int i = threadIdx.x; i < 512; i += blockDim.x) {
+    float v = (float)i;
+    lds[i] = f2bf(v);
+  }
+  __syncthreads();
+  // per-lane address: byte 2*l
+  typedef __attribute__((ext_vector_type(2))) unsigned int u32x2;
+  u32x2 r0, r1;
+  // ds ops take a 32-bit LDS byte address (addrspace(3) starts at 0)
+  const unsigned addr = (unsigned)(uintptr_t)&lds[lane];
+  asm volatile("ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
+               "ds_read_b64_tr_b16 %1, %2 offset:128\n\t"
+               "s_waitcnt lgkmcnt(0)"
+               : "=v"(r0), "=v"(r1)
+               : "v"(addr));
+  short vals[8];
+  *reinterpret_cast<u32x2*>(&vals[0]) = r0;
+  *reinterpret_cast<u32x2*>(&vals[4]) = r1;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) out[lane * 8 + j] = vals[j];
+}
+
+extern "C" void tr_b16_probe_launch(void* out, hipStream_t stream) {
+  hipLaunchKernelGGL(tr_b16_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     (short*)out);
+}
