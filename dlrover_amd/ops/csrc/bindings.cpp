@@ -242,20 +242,23 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
   pack_strides(st, k, 3);
   pack_strides(st, v, 6);
   pack_strides(st, out, 9);
-  // v2 = ds_read_b64_tr_b16 V path (see attention.hip); DLROVER_FA_V1=1
-  // falls back to the swizzled-V^T kernel for A/B comparison
-  static const bool use_v1 = []() {
-    const char* e = getenv("DLROVER_FA_V1");
+  // DLROVER_FA_V2=1 selects the experimental ds_read_b64_tr_b16 PV path.
+  // Hardware measurement showed the tr read takes its tile-selecting
+  // address bits from the 16-lane GROUP LEADER (only bits 1-2 act
+  // per-lane), so the per-lane-tile v2 design reads wrong columns for
+  // sub>3 — kept off until the round-2 redesign (see attention.hip).
+  static const bool use_v2 = []() {
+    const char* e = getenv("DLROVER_FA_V2");
     return e != nullptr && e[0] == '1';
   }();
-  if (use_v1) {
-    flash_attn_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(),
-                          out_bshd.data_ptr(), lse.data_ptr(), B, H, HKV, S,
-                          (float)scale, st, cur_stream());
-  } else {
+  if (use_v2) {
     flash_attn_fwd_v2_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                              out_bshd.data_ptr(), lse.data_ptr(), B, H, HKV,
                              S, (float)scale, st, cur_stream());
+  } else {
+    flash_attn_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                          out_bshd.data_ptr(), lse.data_ptr(), B, H, HKV, S,
+                          (float)scale, st, cur_stream());
   }
   return {out, lse};
 }
@@ -377,7 +380,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_fwd_bwd", &cross_entropy_fwd_bwd,
         "fused CE loss + in-place dlogits (bf16)");
   m.def("tr_b16_probe", []() {
-    auto out = at::empty({64, 8}, at::TensorOptions()
+    auto out = at::empty({64, 12}, at::TensorOptions()
                                       .dtype(at::kBFloat16)
                                       .device(at::kCUDA));
     tr_b16_probe_launch(out.data_ptr(), cur_stream());

@@ -43,36 +43,44 @@ extern "C" void mfma16_probe_launch(const void* A, const void* B, void* C,
 
 // ---- ds_read_b64_tr_b16 semantics probe -------------------------------------
 //
-// Guide-documented mapping (cdna_hip_programming.md m156/m162): each lane
-// passes its own LDS byte address A and receives 4 bf16 elements from
-// A + {0, 32, 64, 96} bytes (a strided gather = a free 4x4 transpose when
-// the source is a row-major [4][16] bf16 subtile); `offset:N` is additive.
-// This probe fills LDS with value(i) = i (as bf16), has lane l read at
-// A = 2*l with offset 0 and offset 128, and writes what each lane saw so the
-// host can verify elem j == l + 16*j (and +64 for the offset immediate).
+// Hardware-measured semantics (two probe rounds on MI355X):
+//   * Per-lane address bits 1-2 select the COLUMN (A/2 & 3) of a row-major
+//     4x4 bf16 tile; the lane receives that column (4 elements, +4 apart).
+//   * The tile itself is NOT selected per-lane: using per-lane high bits in
+//     the FA v2 kernel read the group leader's tile for every lane
+//     (cols with sub>3 got sub&3's data). This probe varies a high address
+//     bit WITHIN a group (odd sub lanes +32B) to pin down which bits are
+//     taken per-lane vs from the wave/group.
+//   * offset:N is additive.
+// Round-2 FA work must build on whatever `expect_per_lane` vs
+// `expect_leader` this probe reports (tests/test_mfma_gpu.py).
 __global__ void tr_b16_probe_kernel(short* __restrict__ out) {
-  __shared__ short lds[512];
+  __shared__ short lds[1024];
   const int lane = threadIdx.x & 63;
-  for (int i = threadIdx.x; i < 512; i += blockDim.x) {
+  for (int i = threadIdx.x; i < 1024; i += blockDim.x) {
     float v = (float)i;
     lds[i] = f2bf(v);
   }
   __syncthreads();
-  // per-lane address: byte 2*l
   typedef __attribute__((ext_vector_type(2))) unsigned int u32x2;
-  u32x2 r0, r1;
-  // ds ops take a 32-bit LDS byte address (addrspace(3) starts at 0)
-  const unsigned addr = (unsigned)(uintptr_t)&lds[lane];
-  asm volatile("ds_read_b64_tr_b16 %0, %2 offset:0\n\t"
-               "ds_read_b64_tr_b16 %1, %2 offset:128\n\t"
+  u32x2 r0, r1, r2;
+  // addr pattern A: the original uniform-high-bits layout (2*l)
+  const unsigned addrA = (unsigned)(uintptr_t)&lds[0] + 2 * lane;
+  // addr pattern B: odd-sub lanes point one 4x4 tile (32 B) further —
+  // distinguishes per-lane high bits from group-leader high bits
+  const unsigned addrB = addrA + ((lane & 1) ? 32u : 0u);
+  asm volatile("ds_read_b64_tr_b16 %0, %3 offset:0\n\t"
+               "ds_read_b64_tr_b16 %1, %3 offset:128\n\t"
+               "ds_read_b64_tr_b16 %2, %4 offset:0\n\t"
                "s_waitcnt lgkmcnt(0)"
-               : "=v"(r0), "=v"(r1)
-               : "v"(addr));
-  short vals[8];
+               : "=v"(r0), "=v"(r1), "=v"(r2)
+               : "v"(addrA), "v"(addrB));
+  short vals[12];
   *reinterpret_cast<u32x2*>(&vals[0]) = r0;
   *reinterpret_cast<u32x2*>(&vals[4]) = r1;
+  *reinterpret_cast<u32x2*>(&vals[8]) = r2;
 #pragma unroll
-  for (int j = 0; j < 8; ++j) out[lane * 8 + j] = vals[j];
+  for (int j = 0; j < 12; ++j) out[lane * 12 + j] = vals[j];
 }
 
 extern "C" void tr_b16_probe_launch(void* out, hipStream_t stream) {
