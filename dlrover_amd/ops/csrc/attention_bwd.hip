@@ -381,6 +381,177 @@ __global__ void f32_to_bf16_kernel(const float* __restrict__ src,
     dst[i] = f2bf(src[i]);
 }
 
+// ============================================================================
+// dKV v2: identical math to fa_bwd_dkv_kernel but the q dimension is staged
+// in 32-row SUBTILES, halving every q-side LDS buffer: 80 KB -> 40 KB per
+// block, so 2-4 blocks/CU instead of 1 (the v1 kernel is latency-exposed at
+// occupancy 1 — 1393 us/call vs dq's 781 at similar work, profiles/r01g).
+// MFMA count per q-row is unchanged; only barrier/staging rounds double.
+
+#define FA_TS 32  // staged q rows per round
+
+// 64-byte-row swizzle for the [128][32] transposed buffers (the (row&7)<<4
+// key of swz2 would wrap outside a 64 B row)
+__device__ __forceinline__ int swz64(int row, int byte_col) {
+  return byte_col ^ ((row & 3) << 4);
+}
+
+__device__ __forceinline__ bf16x8 ld_bT64(const char* lds, int n0, int sub,
+                                          int k0, int quarter) {
+  const int row = n0 + sub;
+  const int bc = (k0 + quarter * 8) * 2;
+  return *reinterpret_cast<const bf16x8*>(lds + row * 64 + swz64(row, bc));
+}
+
+__device__ __forceinline__ bf16x8 ld_a64(const char* lds, int sub, int k0,
+                                         int quarter) {
+  const int bc = (k0 + quarter * 8) * 2;
+  return *reinterpret_cast<const bf16x8*>(lds + sub * 64 + swz64(sub, bc));
+}
+
+// stage a [32][128] tile row-major (+ transposed [128][32])
+__device__ __forceinline__ void stage_tile32(const short* __restrict__ src,
+                                             long long row_stride,
+                                             char* row_lds, char* tr_lds,
+                                             int tid) {
+#pragma unroll
+  for (int c = 0; c < 2; ++c) {
+    const int linear = (tid * 2 + c) * 8;
+    const int row = linear / FA_D;
+    const int col = linear % FA_D;
+    bf16x8 v8 =
+        *reinterpret_cast<const bf16x8*>(src + (long long)row * row_stride + col);
+    *reinterpret_cast<bf16x8*>(row_lds + row * 256 + swz2(row, col * 2)) = v8;
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      *reinterpret_cast<__bf16*>(
+          tr_lds + (col + i) * 64 + swz64(col + i, row * 2)) = v8[i];
+  }
+}
+
+__global__ __launch_bounds__(256) void fa_bwd_dkv_v2_kernel(
+    const short* __restrict__ q, const short* __restrict__ k,
+    const short* __restrict__ v, const short* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ dvec,
+    float* __restrict__ dk32, float* __restrict__ dv32, int B, int H,
+    int HKV, int S, float scale,
+    long long qs_b, long long qs_h, long long qs_s,
+    long long ks_b, long long ks_h, long long ks_s,
+    long long vs_b, long long vs_h, long long vs_s,
+    long long ds_b, long long ds_h, long long ds_s) {
+  __shared__ char q_lds[FA_TS * 256];      // Q rows (8 KB)
+  __shared__ char qt_lds[FA_D * 64];       // Q^T [128][32] (8 KB)
+  __shared__ char do_lds[FA_TS * 256];     // dO rows (8 KB)
+  __shared__ char dot_lds[FA_D * 64];      // dO^T (8 KB)
+  __shared__ char pt_lds[4 * 16 * 64];     // per-wave P^T [16kv][32q] (4 KB)
+  __shared__ char dst_lds[4 * 16 * 64];    // per-wave dS^T (4 KB)
+
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int quarter = lane >> 4;
+  const int sub = lane & 15;
+
+  const int kt = blockIdx.x, h = blockIdx.y, b = blockIdx.z;
+  const int g = h / (H / HKV);
+  const short* k_blk = k + (long long)b * ks_b + (long long)g * ks_h +
+                       (long long)(kt * FA_T) * ks_s;
+  const short* v_blk = v + (long long)b * vs_b + (long long)g * vs_h +
+                       (long long)(kt * FA_T) * vs_s;
+  const short* q_head = q + (long long)b * qs_b + (long long)h * qs_h;
+  const short* do_head = dout + (long long)b * ds_b + (long long)h * ds_h;
+  const long long row_base = ((long long)b * H + h) * S;
+  const int kvrow_w = wave * 16;
+
+  bf16x8 ak[4], av[4];
+#pragma unroll
+  for (int ks = 0; ks < 4; ++ks) {
+    const int ko = ks * 32 + quarter * 8;
+    ak[ks] = *reinterpret_cast<const bf16x8*>(
+        k_blk + (long long)(kvrow_w + sub) * ks_s + ko);
+    av[ks] = *reinterpret_cast<const bf16x8*>(
+        v_blk + (long long)(kvrow_w + sub) * vs_s + ko);
+  }
+
+  f32x4_t acc_dk[8], acc_dv[8];
+#pragma unroll
+  for (int n = 0; n < 8; ++n) {
+    acc_dk[n] = {0.f, 0.f, 0.f, 0.f};
+    acc_dv[n] = {0.f, 0.f, 0.f, 0.f};
+  }
+  char* pt_wave = pt_lds + wave * 16 * 64;
+  char* dst_wave = dst_lds + wave * 16 * 64;
+
+  for (int qs0 = kt * FA_T; qs0 < S; qs0 += FA_TS) {
+    stage_tile32(q_head + (long long)qs0 * qs_s, qs_s, q_lds, qt_lds, tid);
+    stage_tile32(do_head + (long long)qs0 * ds_s, ds_s, do_lds, dot_lds, tid);
+    __syncthreads();
+
+    // S^T = K_band @ Q^T ; dP^T = V_band @ dO^T   (both [16kv, 32q])
+    f32x4_t acc_st[2], acc_dpt[2];
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      acc_st[n] = {0.f, 0.f, 0.f, 0.f};
+      acc_dpt[n] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        bf16x8 bq = ld_bT(q_lds, n * 16, sub, ks * 32, quarter, 256);
+        acc_st[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ak[ks], bq,
+                                                            acc_st[n], 0, 0, 0);
+        bf16x8 bdo = ld_bT(do_lds, n * 16, sub, ks * 32, quarter, 256);
+        acc_dpt[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            av[ks], bdo, acc_dpt[n], 0, 0, 0);
+      }
+    }
+    const int kvrow0 = kt * FA_T + kvrow_w + quarter * 4;
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      const int qcol = qs0 + n * 16 + sub;
+      const float l_col = lse[row_base + qcol];
+      const float d_col = dvec[row_base + qcol];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float p = __expf(acc_st[n][r] * scale - l_col);
+        if (qcol < kvrow0 + r) p = 0.f;
+        const float ds = p * (acc_dpt[n][r] - d_col) * scale;
+        const int prow = quarter * 4 + r;
+        const int pcol = n * 16 + sub;
+        *reinterpret_cast<__bf16*>(pt_wave + prow * 64 + swz64(prow, pcol * 2)) =
+            (__bf16)p;
+        *reinterpret_cast<__bf16*>(
+            dst_wave + prow * 64 + swz64(prow, pcol * 2)) = (__bf16)ds;
+      }
+    }
+    __syncthreads();
+    // dV += P^T @ dO ; dK += dS^T @ Q   (k dimension = 32 staged q rows)
+#pragma unroll
+    for (int n = 0; n < 8; ++n) {
+      bf16x8 apt = ld_a64(pt_wave, sub, 0, quarter);
+      bf16x8 bdot = ld_bT64(dot_lds, n * 16, sub, 0, quarter);
+      acc_dv[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(apt, bdot,
+                                                          acc_dv[n], 0, 0, 0);
+      bf16x8 adst = ld_a64(dst_wave, sub, 0, quarter);
+      bf16x8 bqt = ld_bT64(qt_lds, n * 16, sub, 0, quarter);
+      acc_dk[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(adst, bqt,
+                                                          acc_dk[n], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int krow = quarter * 4 + r;
+    const long long out_off =
+        (((long long)b * HKV + g) * S + kt * FA_T + kvrow_w + krow) * FA_D;
+#pragma unroll
+    for (int n = 0; n < 8; ++n) {
+      atomicAdd(&dk32[out_off + n * 16 + sub], acc_dk[n][r]);
+      atomicAdd(&dv32[out_off + n * 16 + sub], acc_dv[n][r]);
+    }
+  }
+}
+
+
 extern "C" {
 
 void fa_bwd_pre_launch(const void* dout, const void* out, void* dvec, int B,
@@ -411,13 +582,29 @@ void fa_bwd_dkv_launch(const void* q, const void* k, const void* v,
                        void* dk32, void* dv32, int B, int H, int HKV, int S,
                        float scale, const long long* st, hipStream_t stream) {
   dim3 grid(S / FA_T, H, B);
-  hipLaunchKernelGGL(fa_bwd_dkv_kernel, grid, dim3(256), 0, stream,
-                     (const short*)q, (const short*)k, (const short*)v,
-                     (const short*)dout, (const float*)lse,
-                     (const float*)dvec, (float*)dk32, (float*)dv32, B, H,
-                     HKV, S, scale,
-                     st[0], st[1], st[2], st[3], st[4], st[5], st[6], st[7],
-                     st[8], st[9], st[10], st[11]);
+  // v2 (32-row q-subtiles, 40 KB LDS, occupancy >1) is the default;
+  // DLROVER_FA_BWD_V1=1 selects the original 80 KB kernel for A/B
+  static const bool use_v1 = []() {
+    const char* e = getenv("DLROVER_FA_BWD_V1");
+    return e != nullptr && e[0] == '1';
+  }();
+  if (use_v1) {
+    hipLaunchKernelGGL(fa_bwd_dkv_kernel, grid, dim3(256), 0, stream,
+                       (const short*)q, (const short*)k, (const short*)v,
+                       (const short*)dout, (const float*)lse,
+                       (const float*)dvec, (float*)dk32, (float*)dv32, B, H,
+                       HKV, S, scale,
+                       st[0], st[1], st[2], st[3], st[4], st[5], st[6], st[7],
+                       st[8], st[9], st[10], st[11]);
+  } else {
+    hipLaunchKernelGGL(fa_bwd_dkv_v2_kernel, grid, dim3(256), 0, stream,
+                       (const short*)q, (const short*)k, (const short*)v,
+                       (const short*)dout, (const float*)lse,
+                       (const float*)dvec, (float*)dk32, (float*)dv32, B, H,
+                       HKV, S, scale,
+                       st[0], st[1], st[2], st[3], st[4], st[5], st[6], st[7],
+                       st[8], st[9], st[10], st[11]);
+  }
 }
 
 void f32_to_bf16_launch(const void* src, void* dst, long long n,
