@@ -582,11 +582,14 @@ void fa_bwd_dkv_launch(const void* q, const void* k, const void* v,
                        void* dk32, void* dv32, int B, int H, int HKV, int S,
                        float scale, const long long* st, hipStream_t stream) {
   dim3 grid(S / FA_T, H, B);
-  // v2 (32-row q-subtiles, 40 KB LDS, occupancy >1) is the default;
-  // DLROVER_FA_BWD_V1=1 selects the original 80 KB kernel for A/B
+  // v1 (64-row tiles, 80 KB LDS) measured FASTER than the 32-row-subtile
+  // v2 (1394 vs 1515 us/call): the subtile variant stays VGPR-occupancy-
+  // limited, so halving LDS bought nothing and the doubled staging/barrier
+  // rounds cost ~9%. DLROVER_FA_BWD_V2=1 keeps v2 selectable for re-testing
+  // after a register-pressure rework (round 2).
   static const bool use_v1 = []() {
-    const char* e = getenv("DLROVER_FA_BWD_V1");
-    return e != nullptr && e[0] == '1';
+    const char* e = getenv("DLROVER_FA_BWD_V2");
+    return !(e != nullptr && e[0] == '1');
   }();
   if (use_v1) {
     hipLaunchKernelGGL(fa_bwd_dkv_kernel, grid, dim3(256), 0, stream,
