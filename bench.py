@@ -39,6 +39,8 @@ def parse_args():
     # K=55 covers one in-window save plus the injected failure restore
     p.add_argument("--ckpt-interval", type=int, default=50)
     p.add_argument("--no-ckpt", action="store_true")
+    p.add_argument("--act-ckpt", action="store_true",
+                   help="recompute blocks in backward (activation ckpt)")
     p.add_argument("--ckpt-scope", default="full", choices=["full", "model"])
     p.add_argument("--lr", type=float, default=1e-4)
     return p.parse_args()
@@ -77,6 +79,8 @@ def build_model(args, device):
     )
     if args.model == "tiny":
         args.seq = min(args.seq, cfg.max_seq_len)
+    if getattr(args, "act_ckpt", False):
+        cfg.activation_checkpointing = True
     with device:
         model = LlamaForCausalLM(cfg)
     model = model.to(device)

@@ -49,3 +49,26 @@ def test_nanogpt_step():
     loss = model(ids, ids.clone())
     loss.backward()
     assert loss.isfinite()
+
+
+def test_activation_checkpointing_grad_equivalence():
+    """cfg.activation_checkpointing recomputes blocks in backward; the
+    gradients must match the stored-activation path exactly."""
+    import torch
+    from dlrover_amd.models import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig.tiny()
+    m1 = LlamaForCausalLM(cfg)
+    cfg2 = LlamaConfig.tiny()
+    cfg2.activation_checkpointing = True
+    m2 = LlamaForCausalLM(cfg2)
+    m2.load_state_dict(m1.state_dict())
+    ids = torch.randint(0, cfg.vocab_size, (2, 16))
+    for m in (m1, m2):
+        m.train()
+        loss = m(ids, ids.clone())
+        loss.backward()
+    for (n1, p1), (n2, p2) in zip(m1.named_parameters(), m2.named_parameters()):
+        assert n1 == n2
+        assert torch.allclose(p1.grad, p2.grad, rtol=1e-5, atol=1e-7), n1
