@@ -61,4 +61,4 @@ def test_bench_under_torchrun(tmp_path):
     assert out.returncode == 0, out.stderr[-3000:]
     r = _parse_bench(out.stdout)
     assert r["n_gpus"] == 2
-    assert r["config"]["global_batch"] == 2  # weak scaling
+    assert r["config"]["global_batch"] == 2 * 2  # ws=2 x default batch 2 (weak scaling)
