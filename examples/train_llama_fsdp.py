@@ -33,6 +33,8 @@ def main():
     p.add_argument("--ckpt-dir", default="/tmp/dlrover_amd_ckpt/llama")
     p.add_argument("--progress-file", default="")
     p.add_argument("--lr", type=float, default=1e-4)
+    p.add_argument("--act-ckpt", action="store_true",
+                   help="recompute transformer blocks in backward")
     args = p.parse_args()
 
     from dlrover_amd.models import LlamaConfig, LlamaForCausalLM
@@ -62,6 +64,7 @@ def main():
     )
     if args.model == "tiny":
         args.seq = min(args.seq, cfg.max_seq_len)
+    cfg.activation_checkpointing = args.act_ckpt
     with device:
         model = LlamaForCausalLM(cfg)
     model = model.to(device)
