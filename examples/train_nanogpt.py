@@ -99,7 +99,8 @@ def main():
                     json.dumps(
                         {"step": step, "loss": round(loss.item(), 4),
                          "incarnation": incarnation, "resumed_from": start_step,
-                         "world": dist.get_world_size()}
+                         "world": dist.get_world_size(),
+                         "device": str(device)}
                     )
                     + "\n"
                 )

@@ -18,7 +18,7 @@ ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 def _run_cli(tmp_path, extra_env=None, steps=12, ckpt_interval=3, max_restarts=2,
-             timeout=420):
+             timeout=420, nproc=2):
     progress = tmp_path / "progress.jsonl"
     ckpt_dir = tmp_path / "ckpt"
     env = dict(os.environ)
@@ -37,7 +37,7 @@ def _run_cli(tmp_path, extra_env=None, steps=12, ckpt_interval=3, max_restarts=2
         "dlrover_amd.trainer.elastic_run",
         "--standalone",
         "--nproc-per-node",
-        "2",
+        str(nproc),
         "--max-restarts",
         str(max_restarts),
         "--monitor-interval",
