@@ -43,3 +43,31 @@ def test_nanogpt_rccl_sigkill_recovery_gpu(tmp_path):
     # the restarted worker resumed from the last committed step, not step 0
     resumed = [r for r in rows if r.get("incarnation") == 1]
     assert resumed and resumed[0].get("resumed_from", 0) >= 3, resumed[:2]
+
+
+@pytest.mark.timeout(540)
+def test_nanogpt_hiptimer_metrics_gpu(tmp_path):
+    """--hiptimer preloads libhiptimer into the RCCL worker; the per-rank
+    Prometheus file must show real kernel launches, per-communicator
+    traffic, and hang=0 (config #5's detection plumbing, live on HW)."""
+    import glob
+    import os
+
+    from dlrover_amd import xpu_timer
+
+    proc, progress, ckpt_dir = _run_cli(
+        tmp_path, steps=6, ckpt_interval=3, nproc=1,
+        extra_env={"DLROVER_HIPTIMER": "1"},
+    )
+    assert proc.returncode == 0, (
+        f"stdout:\n{proc.stdout[-3000:]}\nstderr:\n{proc.stderr[-5000:]}"
+    )
+    # the agent derives the metrics dir from the job name it generated;
+    # find it by glob
+    candidates = glob.glob("/tmp/hiptimer_*/hiptimer_0.prom")
+    newest = max(candidates, key=os.path.getmtime)
+    m = xpu_timer.parse_metrics_file(newest)
+    assert m.get("hiptimer_launched_total", 0) > 100, newest
+    assert m.get("XPU_TIMER_COMMON_HANG") == 0
+    comm_keys = [k for k in m if k.startswith("hiptimer_comm_calls")]
+    assert comm_keys, f"no communicator metrics in {newest}"
