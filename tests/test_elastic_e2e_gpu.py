@@ -57,7 +57,7 @@ def test_nanogpt_hiptimer_metrics_gpu(tmp_path):
 
     proc, progress, ckpt_dir = _run_cli(
         tmp_path, steps=6, ckpt_interval=3, nproc=1,
-        extra_env={"DLROVER_HIPTIMER": "1"},
+        extra_env={"DLROVER_HIPTIMER": "1", "HIPTIMER_DUMP_INTERVAL": "1"},
     )
     assert proc.returncode == 0, (
         f"stdout:\n{proc.stdout[-3000:]}\nstderr:\n{proc.stderr[-5000:]}"
