@@ -270,8 +270,10 @@ class SharedMemoryHandler:
             if self._stager is None:
                 self._stager = _DeviceStager()
             self._stager.snapshot(gpu_tensors)  # blocking: HBM->HBM, ms-class
-            blocking = time.perf_counter() - t0
             done_evt = self._stager.drain_async(host_view, self._pinned)
+            # chunked staging drains synchronously (live sources) — the
+            # honest blocking time then includes the PCIe drain
+            blocking = time.perf_counter() - t0
 
             def _finish():
                 done_evt.synchronize()
@@ -405,13 +407,17 @@ class _DeviceStager:
                 )
                 evt.record(self._stream)
             return evt
-        # chunked fallback: serialize tensor copies through the small buffer
+        # chunked fallback: serialize tensor copies through the small buffer.
+        # The sources are the LIVE tensors (no full device snapshot exists),
+        # so this path must complete before training resumes — synchronize
+        # before returning (blocking bounded by PCIe, as documented). The
+        # non-chunked path above is the one that overlaps with training.
         with torch.cuda.stream(self._stream):
             for tm, t in self._pending:
                 src = t.contiguous().view(-1).view(torch.uint8)
                 n = tm.nbytes
-                stage = self._buf[:n] if n <= self._buf.numel() else src
                 if n <= self._buf.numel():
+                    stage = self._buf[:n]
                     stage.copy_(src)
                     dst[tm.offset : tm.offset + n].copy_(stage, non_blocking=pinned)
                 else:
@@ -422,5 +428,6 @@ class _DeviceStager:
                             self._buf[: c1 - c0], non_blocking=pinned
                         )
             evt.record(self._stream)
+        self._stream.synchronize()
         self._pending = []
         return evt
