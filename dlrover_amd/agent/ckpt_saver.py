@@ -235,7 +235,8 @@ class AsyncCheckpointSaver:
             )
             try:
                 persist_shm_to_storage(
-                    handler, event, self.storage, self.checkpoint_dir, self.expected_shards
+                    handler, event, self.storage, self.checkpoint_dir,
+                    meta.extra.get("expected_shards") or self.expected_shards,
                 )
                 self._persisted_steps[local_rank] = meta.step
             except Exception:  # noqa: BLE001

@@ -177,6 +177,10 @@ class CheckpointEngine:
                     "global_rank": _global_rank(),
                     "world_size": _world_size(),
                     "shard_name": self._shard_file_name(_global_rank()),
+                    # the commit gate for THIS snapshot — the agent's static
+                    # max_nodes-derived count can exceed the live world and
+                    # would block the failure-path commit forever
+                    "expected_shards": self.expected_shards(),
                 }
                 blocking = self.shm_handler.save_state_dict(
                     step, state_dict, extra=extra, block=block
