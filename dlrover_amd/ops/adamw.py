@@ -186,8 +186,11 @@ class FusedAdamW(torch.optim.Optimizer):
             g.update({k: v for k, v in sg.items() if k != "params"})
         self._graph = None  # state tensors moved: any captured graph is stale
 
-    def zero_grad(self, set_to_none: bool = False):
-        # graph capture needs stable grad storage: zero in place by default
+    def zero_grad(self, set_to_none: bool = True):
+        # torch semantics by default: dropping grads lets backward ASSIGN the
+        # first accumulation instead of add_-ing into zeroed storage (measured
+        # ~2% of an 8B step in fills + fan-in adds). Graph capture is the one
+        # case that needs stable grad storage — zero in place there.
         for group in self.param_groups:
             for p in group["params"]:
                 if p.grad is not None:
