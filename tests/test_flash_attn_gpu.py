@@ -155,3 +155,36 @@ def test_rmsnorm_add_gpu_matches_ref():
     torch.testing.assert_close(y.float(), y2, rtol=2e-2, atol=2e-2)
     torch.testing.assert_close(x.grad.float(), x2.grad, rtol=5e-2, atol=5e-1)
     torch.testing.assert_close(w.grad.float(), w2.grad, rtol=5e-2, atol=5e-1)
+
+
+def test_chunked_staging_roundtrip(monkeypatch, tmp_path):
+    """The chunked _DeviceStager fallback (payload too big for a device
+    staging buffer) must produce a byte-identical snapshot — it drains the
+    LIVE tensors synchronously, so mutating them right after save_state_dict
+    returns must NOT affect the stored snapshot."""
+    import os
+    import torch
+
+    monkeypatch.setenv("DLROVER_CKPT_FORCE_CHUNKED", "1")
+    from dlrover_amd.trainer.flash_checkpoint.shm_handler import (
+        SharedMemoryHandler,
+    )
+
+    h = SharedMemoryHandler(f"test_chunked_{os.getpid()}")
+    try:
+        t1 = torch.randn(3 << 20, device="cuda")
+        t2 = torch.randn(1000, device="cuda", dtype=torch.bfloat16)
+        sd = {"a": t1, "b": {"c": t2}, "step": 5}
+        want1, want2 = t1.clone(), t2.clone()
+        blocking = h.save_state_dict(5, sd, block=False)
+        # chunked mode is synchronous: mutations after return are not seen
+        t1.add_(100.0)
+        t2.add_(7.0)
+        h.wait_drained()
+        out = h.load_state_dict(device=torch.device("cuda:0"))
+        assert out["step"] == 5
+        torch.testing.assert_close(out["a"], want1)
+        torch.testing.assert_close(out["b"]["c"], want2)
+        assert blocking > 0
+    finally:
+        h.unlink()
