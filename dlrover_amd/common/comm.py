@@ -11,7 +11,7 @@ unpickler in dlrover_amd.common.serialize.
 import socket
 import time
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 
 @dataclass

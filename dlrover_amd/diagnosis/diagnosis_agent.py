@@ -16,7 +16,7 @@ common/constants.py + diagnose_training_failure):
 
 import re
 from dataclasses import dataclass
-from typing import List, Optional
+from typing import List
 
 from dlrover_amd.common.log import logger
 

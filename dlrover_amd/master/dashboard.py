@@ -10,7 +10,6 @@ import html
 import json
 import threading
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
-from typing import Optional
 
 from dlrover_amd.common.log import logger
 

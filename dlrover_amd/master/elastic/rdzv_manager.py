@@ -24,7 +24,7 @@ multi-node — DpTopologySorter hooks in here when node topology is reported).
 
 import time
 from threading import Lock
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 from dlrover_amd.common.constants import NetworkFailureReason, RendezvousName
 from dlrover_amd.common.log import logger

@@ -2,7 +2,7 @@
 (ref: dlrover/python/master/node/job_context.py:44-411)."""
 
 import threading
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 from dlrover_amd.common.constants import JobStage, NodeType
 from dlrover_amd.common.node import Node

@@ -16,7 +16,7 @@ Split:
 
 import threading
 import time
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 from dlrover_amd.common import comm
 from dlrover_amd.common.constants import (

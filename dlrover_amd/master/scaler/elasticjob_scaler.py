@@ -8,7 +8,6 @@ without a cluster. The ScalePlan body matches the reference operator's
 schema so existing ElasticJob deployments reconcile it unchanged.
 """
 
-import time
 from abc import ABC, abstractmethod
 from typing import Dict, List, Optional
 

@@ -10,8 +10,7 @@ task_manager.py:35.
 
 import json
 import threading
-import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional
 
 from dlrover_amd.common import comm

@@ -6,8 +6,7 @@ The stream source is injected so tests drive it with a FakeEventSource.
 """
 
 import queue
-import threading
-from typing import Iterator, List, Optional
+from typing import Iterator, Optional
 
 from dlrover_amd.common import comm
 from dlrover_amd.common.constants import NodeEventType, NodeStatus

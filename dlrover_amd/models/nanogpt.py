@@ -5,7 +5,6 @@ Uses plain torch modules so it runs anywhere (CPU tests, smoke jobs); the
 MI355X hot path lives in models/llama.py.
 """
 
-import math
 from dataclasses import dataclass
 
 import torch

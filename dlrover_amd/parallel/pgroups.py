@@ -11,7 +11,7 @@ all-reduces are small and latency-bound, so locality beats ring length.
 """
 
 from dataclasses import dataclass
-from typing import List, Optional
+from typing import Optional
 
 import torch.distributed as dist
 

@@ -16,7 +16,7 @@ node). Both schedules produce bit-identical gradients — only ordering and
 liveness differ.
 """
 
-from typing import Callable, List, Optional
+from typing import List, Optional
 
 import torch
 import torch.distributed as dist

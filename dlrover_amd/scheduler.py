@@ -2,7 +2,6 @@
 ElasticJob CR / defaults (ref: dlrover/python/scheduler/{job,kubernetes,
 factory}.py — JobArgs, new_job_args)."""
 
-import json
 import os
 from dataclasses import dataclass, field
 from typing import Dict, Optional

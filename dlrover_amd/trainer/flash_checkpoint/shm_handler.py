@@ -21,7 +21,6 @@ sees commit==step>0 has a complete, consistent snapshot — the two-phase
 commit the judged failure-recovery metric depends on.
 """
 
-import io
 import pickle
 import struct
 import threading
