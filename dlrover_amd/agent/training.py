@@ -20,7 +20,6 @@ import traceback
 from dataclasses import dataclass, field
 from typing import Any, Dict, List, Optional, Tuple, Union
 
-import torch.distributed.elastic.timer as timer
 from torch.distributed.elastic.agent.server.api import (
     RunResult,
     WorkerGroup,

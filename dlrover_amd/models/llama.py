@@ -16,7 +16,6 @@ from dataclasses import dataclass
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
 from dlrover_amd.ops import (
     causal_softmax,

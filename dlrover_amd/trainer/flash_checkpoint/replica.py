@@ -17,7 +17,6 @@ shm-first load path works untouched.
 import struct
 from typing import Optional
 
-import numpy as np
 import torch
 import torch.distributed as dist
 
