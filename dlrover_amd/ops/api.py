@@ -211,23 +211,16 @@ class _RoPE(torch.autograd.Function):
     def forward(ctx, x, pos, cos, sin):
         ctx.save_for_backward(pos, cos, sin)
         if _use_hip(x):
-            # one copy, not two: contiguous() of a strided view already
-            # produces a fresh tensor the kernel may mutate in place
-            out = x.contiguous()
-            if out is x:
-                out = x.clone()
-            hip_ops().rope_apply(out, pos.int(), cos, sin, False)
-            return out
+            # out-of-place kernel reads the strided view (e.g. a q/k slice
+            # of the fused QKV projection) directly — no clone pass
+            return hip_ops().rope_rotate_oop(x, pos.int(), cos, sin, False)
         return rope_ref(x, pos, cos, sin)
 
     @staticmethod
     def backward(ctx, dy):
         pos, cos, sin = ctx.saved_tensors
         if _use_hip(dy):
-            dx = dy.contiguous()
-            if dx is dy:
-                dx = dy.clone()
-            hip_ops().rope_apply(dx, pos.int(), cos, sin, True)
+            dx = hip_ops().rope_rotate_oop(dy, pos.int(), cos, sin, True)
             return dx, None, None, None
         # inverse rotation
         return rope_ref(dy, pos, cos, -sin), None, None, None

@@ -18,6 +18,9 @@ void rmsnorm_bwd_launch(const void*, const void*, const void*, const void*,
                         int, void*);
 void rope_launch(void*, const void*, const void*, const void*, long long,
                  long long, int, int, int, void*);
+void rope_oop_launch(const void*, void*, const void*, const void*,
+                     const void*, long long, long long, int, int, long long,
+                     int, void*);
 void swiglu_fwd_launch(const void*, void*, long long, int, void*);
 void swiglu_bwd_launch(const void*, const void*, void*, long long, int, void*);
 void adamw_launch(void*, const void*, void*, void*, void*, long long, float,
@@ -136,6 +139,42 @@ void rope_apply(at::Tensor& x, const at::Tensor& pos, const at::Tensor& cos,
   rope_launch(x.data_ptr(), pos.data_ptr(), cos.data_ptr(), sin.data_ptr(),
               n_tokens, (long long)pos.numel(), n_heads, head_dim,
               backward ? 1 : 0, cur_stream());
+}
+
+at::Tensor rope_rotate_oop(const at::Tensor& x, const at::Tensor& pos,
+                           const at::Tensor& cos, const at::Tensor& sin,
+                           bool backward) {
+  // x: [..., n_tokens?, n_heads, head_dim] with heads*head_dim contiguous
+  // per token row; the token stride may exceed n_heads*head_dim (a view out
+  // of the fused QKV projection). Returns a fresh CONTIGUOUS tensor.
+  check_bf16(x, "x");
+  TORCH_CHECK(pos.scalar_type() == at::kInt && pos.is_cuda(),
+              "pos must be int32 on GPU");
+  const int head_dim = (int)x.size(-1);
+  const int n_heads = (int)x.size(-2);
+  TORCH_CHECK(head_dim % 8 == 0, "head_dim must be a multiple of 8");
+  TORCH_CHECK(x.stride(-1) == 1 && x.stride(-2) == head_dim,
+              "rope: head_dim/heads must be contiguous within a token row");
+  const long long n_tokens = x.numel() / ((long long)n_heads * head_dim);
+  // flatten every leading dim into tokens: requires uniform token stride
+  long long tok_stride = n_heads * (long long)head_dim;
+  if (x.dim() >= 3) {
+    tok_stride = x.stride(-3);
+    long long rows = x.size(-3);
+    for (int d = (int)x.dim() - 4; d >= 0; --d) {
+      TORCH_CHECK(x.stride(d) == rows * tok_stride,
+                  "rope: leading dims must be uniformly strided");
+      rows *= x.size(d);
+    }
+  }
+  TORCH_CHECK(pos.numel() > 0 && n_tokens % pos.numel() == 0,
+              "pos length must divide the token count");
+  auto out = at::empty(x.sizes(), x.options());
+  rope_oop_launch(x.data_ptr(), out.data_ptr(), pos.data_ptr(),
+                  cos.data_ptr(), sin.data_ptr(), n_tokens,
+                  (long long)pos.numel(), n_heads, head_dim, tok_stride,
+                  backward ? 1 : 0, cur_stream());
+  return out;
 }
 
 at::Tensor swiglu_fwd(const at::Tensor& gate_up) {
@@ -370,6 +409,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_add_fwd", &rmsnorm_add_fwd,
         "fused residual-add + RMSNorm forward -> (h, y, invrms)");
   m.def("rope_apply", &rope_apply, "in-place RoPE rotate-half (bf16)");
+  m.def("rope_rotate_oop", &rope_rotate_oop,
+        "out-of-place RoPE: strided source -> contiguous output");
   m.def("swiglu_fwd", &swiglu_fwd, "fused SwiGLU forward (bf16)");
   m.def("swiglu_bwd", &swiglu_bwd, "fused SwiGLU backward (bf16)");
   m.def("adamw_step", &adamw_step, "fused AdamW with fp32 master weights");

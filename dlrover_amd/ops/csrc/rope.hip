@@ -60,4 +60,60 @@ void rope_launch(void* x, const void* pos, const void* cos_tab,
                      n_pos, n_heads, head_dim, backward ? -1.f : 1.f);
 }
 
+// out-of-place variant: reads a STRIDED source (token rows src_tok_stride
+// elements apart, heads/head_dim contiguous within a row — e.g. q/k views
+// straight out of the fused QKV projection) and writes a fresh contiguous
+// tensor. Replaces the clone-then-rotate-in-place path: the separate
+// strided clone ran at 1.3 TB/s and cost ~1% of the 8B step.
+__global__ void rope_oop_kernel(
+    const short* __restrict__ src, short* __restrict__ dst,
+    const int* __restrict__ pos, const float* __restrict__ cos_tab,
+    const float* __restrict__ sin_tab, long long n_tokens, long long n_pos,
+    int n_heads, int head_dim, long long src_tok_stride, float sin_sign) {
+  const int half = head_dim >> 1;
+  const int vecs_per_head = half >> 2;
+  const long long total = n_tokens * n_heads * vecs_per_head;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    const int v = (int)(i % vecs_per_head);
+    const long long th = i / vecs_per_head;
+    const int h = (int)(th % n_heads);
+    const long long t = th / n_heads;
+    const int d0 = v * 4;
+    const short* sbase = src + t * src_tok_stride + (long long)h * head_dim;
+    short* dbase = dst + (t * n_heads + h) * (long long)head_dim;
+    short4_t v1 = *reinterpret_cast<const short4_t*>(sbase + d0);
+    short4_t v2 = *reinterpret_cast<const short4_t*>(sbase + half + d0);
+    const int p = pos[t % n_pos];
+    const float* cr = cos_tab + (long long)p * half + d0;
+    const float* sr = sin_tab + (long long)p * half + d0;
+    float4_t c = *reinterpret_cast<const float4_t*>(cr);
+    float4_t s = *reinterpret_cast<const float4_t*>(sr);
+    short4_t o1, o2;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float x1 = bf2f(v1[j]);
+      const float x2 = bf2f(v2[j]);
+      const float sj = s[j] * sin_sign;
+      o1[j] = f2bf(x1 * c[j] - x2 * sj);
+      o2[j] = f2bf(x2 * c[j] + x1 * sj);
+    }
+    *reinterpret_cast<short4_t*>(dbase + d0) = o1;
+    *reinterpret_cast<short4_t*>(dbase + half + d0) = o2;
+  }
+}
+
+void rope_oop_launch(const void* src, void* dst, const void* pos,
+                     const void* cos_tab, const void* sin_tab,
+                     long long n_tokens, long long n_pos, int n_heads,
+                     int head_dim, long long src_tok_stride, int backward,
+                     hipStream_t stream) {
+  const long long total = n_tokens * n_heads * (head_dim >> 3);
+  hipLaunchKernelGGL(rope_oop_kernel, dim3(grid_capped(total, 256)),
+                     dim3(256), 0, stream, (const short*)src, (short*)dst,
+                     (const int*)pos, (const float*)cos_tab,
+                     (const float*)sin_tab, n_tokens, n_pos, n_heads,
+                     head_dim, src_tok_stride, backward ? -1.f : 1.f);
+}
+
 }  // extern "C"
