@@ -206,6 +206,23 @@ def swiglu(gate_up: torch.Tensor) -> torch.Tensor:
     return _SwiGLU.apply(gate_up)
 
 
+def _rope_addressable(x: torch.Tensor) -> torch.Tensor:
+    """The oop rope kernel addresses [tokens(strided), heads, head_dim] with
+    heads*head_dim contiguous and a uniform token stride; anything else
+    (rare) goes through one contiguous copy."""
+    hd = x.size(-1)
+    if x.stride(-1) != 1 or x.stride(-2) != hd:
+        return x.contiguous()
+    if x.dim() >= 3:
+        tok_stride = x.stride(-3)
+        rows = x.size(-3)
+        for d in range(x.dim() - 4, -1, -1):
+            if x.stride(d) != rows * tok_stride:
+                return x.contiguous()
+            rows *= x.size(d)
+    return x
+
+
 class _RoPE(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, pos, cos, sin):
@@ -213,14 +230,18 @@ class _RoPE(torch.autograd.Function):
         if _use_hip(x):
             # out-of-place kernel reads the strided view (e.g. a q/k slice
             # of the fused QKV projection) directly — no clone pass
-            return hip_ops().rope_rotate_oop(x, pos.int(), cos, sin, False)
+            return hip_ops().rope_rotate_oop(
+                _rope_addressable(x), pos.int(), cos, sin, False
+            )
         return rope_ref(x, pos, cos, sin)
 
     @staticmethod
     def backward(ctx, dy):
         pos, cos, sin = ctx.saved_tensors
         if _use_hip(dy):
-            dx = hip_ops().rope_rotate_oop(dy, pos.int(), cos, sin, True)
+            dx = hip_ops().rope_rotate_oop(
+                _rope_addressable(dy), pos.int(), cos, sin, True
+            )
             return dx, None, None, None
         # inverse rotation
         return rope_ref(dy, pos, cos, -sin), None, None, None

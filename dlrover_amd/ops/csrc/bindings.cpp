@@ -147,7 +147,8 @@ at::Tensor rope_rotate_oop(const at::Tensor& x, const at::Tensor& pos,
   // x: [..., n_tokens?, n_heads, head_dim] with heads*head_dim contiguous
   // per token row; the token stride may exceed n_heads*head_dim (a view out
   // of the fused QKV projection). Returns a fresh CONTIGUOUS tensor.
-  check_bf16(x, "x");
+  TORCH_CHECK(x.is_cuda(), "x must be on GPU");
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16, "x must be bf16");
   TORCH_CHECK(pos.scalar_type() == at::kInt && pos.is_cuda(),
               "pos must be int32 on GPU");
   const int head_dim = (int)x.size(-1);
