@@ -71,11 +71,18 @@ class ListIdx(int):
 
 
 def traverse_state_dict(value: Any, path: Tuple = ()):
-    """Yield (path, leaf) pairs; dicts and lists/tuples are traversed."""
+    """Yield (path, leaf) pairs; dicts and lists/tuples are traversed.
+    EMPTY containers are yielded as leaves — otherwise an unstepped
+    optimizer's {"state": {}} would vanish in the roundtrip and break
+    load_state_dict on restore."""
     if isinstance(value, dict):
+        if not value:
+            yield path, value
         for k, v in value.items():
             yield from traverse_state_dict(v, path + (k,))
     elif isinstance(value, (list, tuple)):
+        if not value:
+            yield path, value
         for i, v in enumerate(value):
             yield from traverse_state_dict(v, path + (ListIdx(i),))
     else:
