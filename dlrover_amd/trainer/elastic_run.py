@@ -31,8 +31,13 @@ from dlrover_amd.utils.transport import wait_for_server
 def parse_nnodes(val: str) -> Tuple[int, int]:
     if ":" in val:
         lo, hi = val.split(":")
-        return int(lo), int(hi)
+        lo_i, hi_i = int(lo), int(hi)
+        if lo_i < 1 or hi_i < lo_i:
+            raise ValueError(f"--nnodes {val}: need 1 <= MIN <= MAX")
+        return lo_i, hi_i
     n = int(val)
+    if n < 1:
+        raise ValueError(f"--nnodes {val}: need >= 1 node")
     return n, n
 
 

@@ -157,3 +157,32 @@ def test_sharding_client_against_local_master(tmp_path):
         assert sorted(seen) == list(range(12)), seen
     finally:
         master.stop()
+
+
+def test_dlrover_run_arg_parsing():
+    """dlrover-run accepts the torchrun-superset surface (both --kebab and
+    --snake spellings) and splits script args correctly."""
+    from dlrover_amd.trainer.elastic_run import parse_args, parse_nnodes
+
+    args = parse_args(
+        [
+            "--nnodes", "2:4", "--nproc_per_node", "8", "--max_restarts", "5",
+            "--standalone", "--network_check", "--node_unit", "2",
+            "--hiptimer", "train.py", "--steps", "100", "--lr", "1e-4",
+        ]
+    )
+    assert args.nnodes == "2:4" and parse_nnodes(args.nnodes) == (2, 4)
+    assert parse_nnodes("3") == (3, 3)
+    assert args.nproc_per_node == 8
+    assert args.max_restarts == 5
+    assert args.standalone and args.network_check and args.hiptimer
+    assert args.node_unit == 2
+    assert args.training_script == "train.py"
+    assert args.training_script_args == ["--steps", "100", "--lr", "1e-4"]
+
+
+def test_dlrover_run_nnodes_rejects_garbage():
+    from dlrover_amd.trainer.elastic_run import parse_nnodes
+
+    with pytest.raises(ValueError):
+        parse_nnodes("4:2")  # min > max
