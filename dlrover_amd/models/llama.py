@@ -43,6 +43,8 @@ class LlamaConfig:
     # "auto": hand-written flash kernel on GPU when D==128 and S%64==0,
     # composite (hipBLASLt GEMM + fused softmax) otherwise
     attn_impl: str = "auto"
+    # qkv projection bias (Qwen2-style checkpoints; Llama-3 uses none)
+    attn_bias: bool = False
     # recompute each block in backward instead of saving activations
     # (non-reentrant torch.utils.checkpoint — composes with FSDP2); trades
     # ~30% step time for ~n_layers x less activation memory, the right move
@@ -70,6 +72,22 @@ class LlamaConfig:
             n_kv_heads=2,
             max_seq_len=128,
             rope_base=10000.0,
+        )
+
+    @classmethod
+    def qwen2_7b(cls, max_seq_len: int = 8192) -> "LlamaConfig":
+        """Qwen2-7B geometry: Llama architecture + qkv bias + tied-free
+        151k vocab (the HIP kernel stack is identical)."""
+        return cls(
+            vocab_size=152064,
+            hidden_size=3584,
+            intermediate_size=18944,
+            n_layers=28,
+            n_heads=28,
+            n_kv_heads=4,
+            max_seq_len=max_seq_len,
+            rope_base=1000000.0,
+            attn_bias=True,
         )
 
     @classmethod
@@ -101,7 +119,7 @@ class Attention(nn.Module):
         self.cfg = cfg
         d, hd = cfg.hidden_size, cfg.head_dim
         self.qkv_proj = nn.Linear(
-            d, (cfg.n_heads + 2 * cfg.n_kv_heads) * hd, bias=False
+            d, (cfg.n_heads + 2 * cfg.n_kv_heads) * hd, bias=cfg.attn_bias
         )
         self.o_proj = nn.Linear(cfg.n_heads * hd, d, bias=False)
 

@@ -72,3 +72,22 @@ def test_activation_checkpointing_grad_equivalence():
     for (n1, p1), (n2, p2) in zip(m1.named_parameters(), m2.named_parameters()):
         assert n1 == n2
         assert torch.allclose(p1.grad, p2.grad, rtol=1e-5, atol=1e-7), n1
+
+
+def test_qwen2_style_config_trains():
+    """Qwen2 geometry = Llama arch + qkv bias; one CPU step must run and the
+    bias must exist and receive gradient."""
+    import torch
+    from dlrover_amd.models import LlamaConfig, LlamaForCausalLM
+
+    cfg = LlamaConfig.tiny()
+    cfg.attn_bias = True
+    m = LlamaForCausalLM(cfg)
+    assert m.blocks[0].attn.qkv_proj.bias is not None
+    ids = torch.randint(0, cfg.vocab_size, (2, 16))
+    loss = m(ids, ids.clone())
+    loss.backward()
+    assert m.blocks[0].attn.qkv_proj.bias.grad is not None
+    # geometry sanity of the full preset
+    q = LlamaConfig.qwen2_7b()
+    assert q.attn_bias and q.head_dim == 128 and q.n_kv_heads == 4
