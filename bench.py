@@ -32,7 +32,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=110, help="timed steps (K)")
     p.add_argument("--warmup", type=int, default=3, help="untimed steps (W)")
-    p.add_argument("--model", default="llama3_8b", choices=["llama3_8b", "small_1b", "tiny"])
+    p.add_argument("--model", default="llama3_8b", choices=["llama3_8b", "qwen2_7b", "small_1b", "tiny"])
     p.add_argument("--batch", type=int, default=2, help="per-GPU micro batch")
     p.add_argument("--seq", type=int, default=4096)
     # BASELINE.json config #2 names "flash-checkpoint every 50 steps";
