@@ -21,10 +21,38 @@ the same line. vs_baseline divides by the reference's 95% goodput headline.
 import argparse
 import json
 import os
+import sys
 import time
 
 import torch
 import torch.distributed as dist
+
+_T0 = time.perf_counter()
+
+
+def hb(msg: str):
+    """Per-phase heartbeat on stderr (stdout stays the one JSON line).
+    A crash mid-run then localizes itself in the driver's stderr tail."""
+    if os.getenv("DLROVER_BENCH_QUIET", "") == "1":
+        return
+    r = os.environ.get("RANK", "0")
+    print(f"[bench hb r{r} +{time.perf_counter() - _T0:.1f}s] {msg}",
+          file=sys.stderr, flush=True)
+
+
+def self_launch(args):
+    """`bench.py --gpus N` run bare (no RANK in env) must really measure N
+    ranks: exec torchrun with ourselves as the target (the driver may launch
+    either way). Ref behavior: launcher owns process spawn
+    (elastic_run.py:246-643)."""
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", f"--nproc-per-node={args.gpus}",
+        "--master-addr=127.0.0.1", "--master-port=29573",
+        sys.argv[0],
+    ] + sys.argv[1:]
+    hb(f"self-launching {args.gpus} ranks via torch.distributed.run")
+    os.execv(sys.executable, cmd)
 
 
 def parse_args():
@@ -54,22 +82,34 @@ def setup_dist(args):
         world = int(os.environ["WORLD_SIZE"])
         local_rank = int(os.environ.get("LOCAL_RANK", rank))
     else:
+        if args.gpus > 1:
+            self_launch(args)  # execs torchrun; does not return
         rank, world, local_rank = 0, 1, 0
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29571")
         os.environ.setdefault("RANK", "0")
         os.environ.setdefault("WORLD_SIZE", "1")
         os.environ.setdefault("LOCAL_RANK", "0")
+    if args.gpus > 1 and world != args.gpus:
+        raise RuntimeError(
+            f"--gpus {args.gpus} but WORLD_SIZE={world}: refusing to "
+            "silently measure a different world size"
+        )
     on_gpu = torch.cuda.is_available()
     backend = "nccl" if on_gpu else "gloo"  # nccl == RCCL on ROCm
     if on_gpu:
-        torch.cuda.set_device(local_rank)
+        # modulo: lets N ranks share one device for multi-rank RCCL
+        # validation on a 1-GPU box (the 8-GPU node maps 1:1)
+        dev = local_rank % torch.cuda.device_count()
+        torch.cuda.set_device(dev)
         dist.init_process_group(
             backend=backend, rank=rank, world_size=world,
-            device_id=torch.device(f"cuda:{local_rank}"),
+            device_id=torch.device(f"cuda:{dev}"),
         )
+        local_rank = dev
     else:
         dist.init_process_group(backend=backend, rank=rank, world_size=world)
+    hb(f"dist ready: rank={rank} world={world} backend={backend}")
     return rank, world, local_rank, on_gpu
 
 
@@ -119,8 +159,10 @@ def main():
 
         hip_ops()  # never bench a silent eager fallback
 
+    hb(f"building {args.model} on {device}")
     model, cfg = build_model(args, device)
     n_params = sum(p.numel() for p in model.parameters())
+    hb(f"model built: {n_params / 1e9:.2f}B params")
     # FSDP degenerates to 1 shard at world_size 1 but still runs its
     # all-gather copy-in/out machinery (~7% of step, measured: profiles/r01d
     # __amd_rocclr_copyBuffer + chunk_cat) — shard only when there is
@@ -168,8 +210,10 @@ def main():
         return sd
 
     # ---- warmup (also seeds optimizer state so checkpoints are full-size)
-    for _ in range(args.warmup):
+    for w in range(args.warmup):
+        hb(f"warmup step {w + 1}/{args.warmup}")
         train_step()
+    hb("warmup done; sizing ckpt segment")
     if cp is not None:
         # size the shm segment + staging outside the timed window (the
         # reference also excludes first-export spin-up, ~20 s: BASELINE.md)
@@ -187,12 +231,14 @@ def main():
     save_blockings = []
     restore_s = None
     restore_at = args.steps // 2
+    hb(f"timed window begin: K={args.steps}")
     t_begin = time.perf_counter()
     for k in range(args.steps):
         t0 = time.perf_counter()
         loss = train_step()
         sync()
         useful += time.perf_counter() - t0
+        hb(f"step {k + 1}/{args.steps}")
         if (cp is not None and (k + 1) % args.ckpt_interval == 0
                 and k + 1 < args.steps):
             # no save on the final step: its async drain would sit in the
@@ -202,6 +248,7 @@ def main():
             sd["step"] = k + 1
             blocking = cp.engine.save_to_memory(k + 1, sd, block=False)
             save_blockings.append(blocking)
+            hb(f"flash save @ step {k + 1}: blocking {blocking * 1e3:.1f} ms")
         if cp is not None and k + 1 == restore_at:
             # simulated failure recovery: reload model+optimizer from shm
             cp.engine.shm_handler.wait_drained()
@@ -212,12 +259,14 @@ def main():
             assert sd is not None, "no checkpoint in shm to restore from"
             sync()
             restore_s = time.perf_counter() - t0
+            hb(f"restore @ step {k + 1}: {restore_s:.2f} s")
     if cp is not None:
         cp.engine.shm_handler.wait_drained()
     sync()
     dist.barrier()
     sync()
     t_total = time.perf_counter() - t_begin
+    hb(f"timed window done: {t_total:.2f} s")
 
     # ---- aggregate across ranks: total = max, useful = min (conservative)
     stats = torch.tensor([t_total, useful], dtype=torch.float64)
