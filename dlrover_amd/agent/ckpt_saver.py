@@ -56,6 +56,7 @@ def persist_shm_to_storage(
     if meta is None:
         logger.warning("shm %s holds no committed snapshot", handler.name)
         return False
+    step = meta.step
     if meta.step != event.step:
         logger.warning(
             "shm %s holds step %s, event asked for %s — persisting what we have",
@@ -63,8 +64,11 @@ def persist_shm_to_storage(
             meta.step,
             event.step,
         )
-    step = meta.step
-    path = event.path or os.path.join(checkpoint_dir, str(step))
+        # event.path is named for event.step; the tracker commit below uses
+        # meta.step — derive the directory from meta.step so they agree
+        path = os.path.join(checkpoint_dir, str(step))
+    else:
+        path = event.path or os.path.join(checkpoint_dir, str(step))
     global_rank = meta.extra.get("global_rank", event.global_rank)
     state = handler.load_state_dict()
     storage.safe_makedirs(path)

@@ -128,8 +128,13 @@ class FusedAdamW(torch.optim.Optimizer):
                 return loss
             torch.cuda.synchronize()
             self._graph = torch.cuda.CUDAGraph()
+            # capture records the kernels WITHOUT executing them — the
+            # step counters bumped inside _one_step() then describe work
+            # that has not run yet. Replay immediately (capture-then-replay)
+            # so the capture iteration's gradients are actually applied.
             with torch.cuda.graph(self._graph):
                 self._one_step()
+            self._graph.replay()
             self._graph_ptrs = ptrs
             self._lr_at_capture = lr
             return loss

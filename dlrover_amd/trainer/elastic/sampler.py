@@ -56,6 +56,15 @@ class ElasticDistributedSampler(Sampler):
         # skip what the job already consumed this epoch, then shard round-robin
         start = self.completed_num
         indices = indices[start:]
+        # after an elastic world-size change completed_num is generally NOT a
+        # multiple of the new num_replicas: pad the remainder back to one so
+        # every rank yields exactly the same count (unequal per-rank lengths
+        # hang DDP at epoch end). Mirrors the reference sampler's re-pad.
+        rem = len(indices) % self.num_replicas
+        if rem:
+            pad = self.num_replicas - rem
+            pool = indices if indices else list(range(len(self.dataset)))
+            indices += pool[:pad]
         return iter(indices[self.rank :: self.num_replicas])
 
     def __len__(self) -> int:

@@ -183,7 +183,29 @@ class SharedMemoryHandler:
         """Attach read-only (agent side). Returns False if absent."""
         if self._shm is None:
             self._shm = attach_shared_memory(self.name)
+            self._attached_ro = self._shm is not None
         return self._shm is not None
+
+    def _refresh_attachment(self):
+        """A writer that OUTGREW the segment unlinks and re-creates it
+        (ensure_size) — a reader's cached mapping then points at the dead
+        segment and would silently read stale snapshots forever. Re-attach
+        whenever the live file's size no longer matches our mapping."""
+        if not getattr(self, "_attached_ro", False) or self._shm is None:
+            return
+        import os as _os
+
+        try:
+            live = _os.stat(f"/dev/shm/{self.name}").st_size
+        except OSError:
+            live = -1
+        if live != self._shm.size:
+            try:
+                self._shm.close()
+            except BufferError:
+                pass
+            self._shm = attach_shared_memory(self.name)
+            self._attached_ro = self._shm is not None
 
     def _buf_addr(self) -> int:
         return ctypes_addr(self._shm.buf)
@@ -214,6 +236,7 @@ class SharedMemoryHandler:
         struct.pack_into("<q", self._shm.buf, 0, step)
 
     def committed_step(self) -> int:
+        self._refresh_attachment()
         if self._shm is None and not self.attach():
             return 0
         return struct.unpack_from("<q", self._shm.buf, 0)[0]
@@ -229,6 +252,7 @@ class SharedMemoryHandler:
         self._shm.buf[16 : 16 + len(blob)] = blob
 
     def read_meta(self) -> Optional[SegmentMeta]:
+        self._refresh_attachment()
         if self._shm is None and not self.attach():
             return None
         step = self.committed_step()

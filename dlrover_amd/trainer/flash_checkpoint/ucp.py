@@ -123,9 +123,13 @@ def load_resharded(engine, model, optimizer, path: str) -> Optional[dict]:
                 continue
             dst = _to_local(live[name].data)
             if tags.get(name, "replicated") == "shard0":
+                # cache keyed by FULL dim-0 (rank-uniform): every rank takes
+                # the same hit/miss sequence, so the all_gather stays
+                # collective-consistent
                 offs = offsets_cache.get(saved.shape[0])
                 if offs is None:
                     offs = gather_new_offsets(dst.shape[0])
+                    offsets_cache[saved.shape[0]] = offs
                 my = _slice_for_rank(saved, offs, rank, dst.shape[0])
             else:
                 my = saved
@@ -143,10 +147,13 @@ def load_resharded(engine, model, optimizer, path: str) -> Optional[dict]:
             if idx >= len(params):
                 continue
             local = _to_local(params[idx])
+            # the all_gather must be UNCONDITIONAL per state index: gating it
+            # on v.shape[0] != local.shape[0] (a per-rank fact — one rank can
+            # hold all rows while others hold 0) deadlocks the collective
+            offs = gather_new_offsets(local.shape[0])
             sliced = {}
             for k, v in st.items():
                 if torch.is_tensor(v) and v.dim() >= 1 and v.shape[0] != local.shape[0]:
-                    offs = gather_new_offsets(local.shape[0])
                     sliced[k] = _slice_for_rank(v, offs, rank, local.shape[0]).clone()
                 else:
                     sliced[k] = v
