@@ -12,6 +12,9 @@ What is measured (all real, nothing modeled):
   - goodput % = (sum of pure train-step seconds, min over ranks) /
     (timed wall seconds, max over ranks) — the same accounting the reference
     quotes 95% goodput with (README.md:61, flash_checkpoint.md:38).
+    Per-step timing waits the COMPUTE stream only, so the async ckpt drain
+    on its side stream is never booked as useful time; saves are skipped
+    when the window has no remaining steps to hide their drain under.
 
 value = goodput %; config carries ckpt_save_blocking_s / ckpt_restore_s /
 tokens_per_s so the save/restore seconds of the metric name are reported on
@@ -197,6 +200,13 @@ def main():
         if on_gpu:
             torch.cuda.synchronize()
 
+    def sync_compute():
+        # per-step timing must wait the COMPUTE stream only: a device-wide
+        # synchronize would also wait the checkpoint drain running on its
+        # side stream, booking checkpoint time as useful training time
+        if on_gpu:
+            torch.cuda.current_stream().synchronize()
+
     def train_step():
         ids, labels = fresh_batch()
         loss = model(ids, labels)
@@ -220,8 +230,15 @@ def main():
         # step must be > 0: the shm commit word treats 0 as "empty"
         sd = ckpt_state()
         sd["step"] = 1
+        t0 = time.perf_counter()
         cp.engine.save_to_memory(1, sd)
         cp.engine.wait_saving()
+        # measured full save+drain seconds: used to decide whether a later
+        # in-window save's async drain still has training steps to hide under
+        drain_estimate = time.perf_counter() - t0
+        hb(f"ckpt segment sized; full drain {drain_estimate:.2f} s")
+    else:
+        drain_estimate = 0.0
     sync()
     dist.barrier()
     sync()
@@ -232,18 +249,24 @@ def main():
     restore_s = None
     restore_at = args.steps // 2
     hb(f"timed window begin: K={args.steps}")
+    step_est = None  # running mean of pure step seconds
     t_begin = time.perf_counter()
     for k in range(args.steps):
         t0 = time.perf_counter()
         loss = train_step()
-        sync()
-        useful += time.perf_counter() - t0
+        sync_compute()
+        dt = time.perf_counter() - t0
+        useful += dt
+        step_est = dt if step_est is None else 0.7 * step_est + 0.3 * dt
         hb(f"step {k + 1}/{args.steps}")
+        remaining_cover = (args.steps - (k + 1)) * (step_est or 0.0)
         if (cp is not None and (k + 1) % args.ckpt_interval == 0
-                and k + 1 < args.steps):
-            # no save on the final step: its async drain would sit in the
-            # timed window with no steps left to overlap (steady-state
-            # training always has future steps to hide the drain under)
+                and k + 1 < args.steps
+                and remaining_cover > drain_estimate):
+            # saves only when the async drain has future steps to hide
+            # under — steady-state training always does; only a driver
+            # window shorter than one drain doesn't (the drain would sit
+            # exposed in the timed window as pure checkpoint stall)
             sd = ckpt_state()
             sd["step"] = k + 1
             blocking = cp.engine.save_to_memory(k + 1, sd, block=False)
