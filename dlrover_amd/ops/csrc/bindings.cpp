@@ -31,6 +31,7 @@ void causal_softmax_bwd_launch(void*, const void*, long long, int, float,
 void cross_entropy_launch(void*, const void*, void*, long long, int, int,
                           float, int, void*);
 void mfma16_probe_launch(const void*, const void*, void*, void*);
+void mfma32_probe_launch(const void*, const void*, void*, void*);
 void tr_b16_probe_launch(void*, void*);
 void flash_attn_fwd_launch(const void*, const void*, const void*, void*,
                            void*, int, int, int, int, float,
@@ -385,6 +386,17 @@ at::Tensor mfma16_probe(const at::Tensor& A, const at::Tensor& B) {
   return C;
 }
 
+at::Tensor mfma32_probe(const at::Tensor& A, const at::Tensor& B) {
+  check_bf16(A, "A");
+  check_bf16(B, "B");
+  TORCH_CHECK(A.sizes() == at::IntArrayRef({32, 16}) &&
+              B.sizes() == at::IntArrayRef({16, 32}),
+              "probe wants A[32,16], B[16,32]");
+  auto C = at::empty({32, 32}, A.options().dtype(at::kFloat));
+  mfma32_probe_launch(A.data_ptr(), B.data_ptr(), C.data_ptr(), cur_stream());
+  return C;
+}
+
 at::Tensor cross_entropy_fwd_bwd(at::Tensor& logits, const at::Tensor& targets,
                                  int64_t ignore_index, double grad_scale,
                                  bool compute_grad) {
@@ -429,6 +441,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     return out;
   }, "ds_read_b64_tr_b16 semantics probe");
   m.def("mfma16_probe", &mfma16_probe, "MFMA 16x16x32 bf16 layout self-test");
+  m.def("mfma32_probe", &mfma32_probe, "MFMA 32x32x16 bf16 layout self-test");
   m.def("flash_attn_fwd", &flash_attn_fwd,
         "flash attention forward (bf16, causal, GQA, D=128) -> (out, lse)");
   m.def("flash_attn_bwd", &flash_attn_bwd,

@@ -41,6 +41,50 @@ extern "C" void mfma16_probe_launch(const void* A, const void* B, void* C,
                      (const short*)A, (const short*)B, (float*)C);
 }
 
+// ---- 32x32x16 MFMA layout probe ---------------------------------------------
+//
+// C[32,32] = A[32,16] @ B[16,32] with one v_mfma_f32_32x32x16_bf16 per wave.
+// Assumed layouts (guide §3 gives C/D measured; A/B by analogy with the
+// 16x16x32 family — THIS PROBE verifies them on hardware before the FA v3
+// kernel builds on them):
+//   A[32,16]: lane l holds A[l%32][(l/32)*8 + i], i in [0,8)
+//   B[16,32]: lane l holds B[(l/32)*8 + i][l%32]
+//   C[32,32]: lane l reg r -> C[(r&3) + 8*(r>>2) + 4*(l>>5)][l&31]
+//             (guide §3, measured m74/m101), r in [0,16)
+__global__ void mfma32_probe_kernel(const short* __restrict__ A,
+                                    const short* __restrict__ B,
+                                    float* __restrict__ C) {
+  const int lane = threadIdx.x & 63;
+  const int half = lane >> 5;  // which 8-wide k-slice
+  const int sub = lane & 31;
+  bf16x8 a, b;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    short abits = A[sub * 16 + half * 8 + i];
+    short bbits = B[(half * 8 + i) * 32 + sub];
+    a[i] = *reinterpret_cast<__bf16*>(&abits);
+    b[i] = *reinterpret_cast<__bf16*>(&bbits);
+  }
+  typedef __attribute__((ext_vector_type(16))) float f32x16_;
+  f32x16_ acc;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) acc[r] = 0.f;
+  acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+      *reinterpret_cast<bf16x8*>(&a), *reinterpret_cast<bf16x8*>(&b), acc, 0,
+      0, 0);
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int row = (r & 3) + 8 * (r >> 2) + 4 * half;
+    C[row * 32 + sub] = acc[r];
+  }
+}
+
+extern "C" void mfma32_probe_launch(const void* A, const void* B, void* C,
+                                    hipStream_t stream) {
+  hipLaunchKernelGGL(mfma32_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     (const short*)A, (const short*)B, (float*)C);
+}
+
 // ---- ds_read_b64_tr_b16 semantics probe -------------------------------------
 //
 // Hardware-measured semantics (two probe rounds on MI355X):

@@ -38,6 +38,9 @@ def main():
     use_gpu = torch.cuda.is_available()
     local_rank = int(os.getenv("LOCAL_RANK", "0"))
     if use_gpu:
+        # modulo: N ranks can share one device (multi-rank RCCL validation
+        # on a 1-GPU box; on the 8-GPU node the mapping is 1:1)
+        local_rank = local_rank % torch.cuda.device_count()
         torch.cuda.set_device(local_rank)
     # short collective timeout so a dead peer surfaces as a worker failure
     # quickly (elastic scale-down path); DLROVER_PG_TIMEOUT in seconds

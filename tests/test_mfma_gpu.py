@@ -18,6 +18,18 @@ def test_mfma16_probe_matches_matmul():
     torch.testing.assert_close(C, ref, rtol=2e-2, atol=2e-2)
 
 
+def test_mfma32_probe_matches_matmul():
+    """v_mfma_f32_32x32x16_bf16 A/B/C layouts (FA v3 builds on these)."""
+    from dlrover_amd.ops.api import hip_ops
+
+    torch.manual_seed(1)
+    A = torch.randn(32, 16, device="cuda", dtype=torch.bfloat16)
+    B = torch.randn(16, 32, device="cuda", dtype=torch.bfloat16)
+    C = hip_ops().mfma32_probe(A, B)
+    ref = A.float() @ B.float()
+    torch.testing.assert_close(C, ref, rtol=2e-2, atol=2e-2)
+
+
 def test_tr_b16_probe():
     """ds_read_b64_tr_b16 hardware semantics (round-2 FA groundwork).
 
