@@ -230,13 +230,17 @@ def main():
         # step must be > 0: the shm commit word treats 0 as "empty"
         sd = ckpt_state()
         sd["step"] = 1
-        t0 = time.perf_counter()
         cp.engine.save_to_memory(1, sd)
         cp.engine.wait_saving()
-        # measured full save+drain seconds: used to decide whether a later
-        # in-window save's async drain still has training steps to hide under
+        # second save now that the segment exists+is pinned: measures the
+        # STEADY-STATE drain (the first one includes one-time shm creation
+        # and 100-GB-class page pinning), used to decide whether a later
+        # in-window save's async drain still has steps to hide under
+        t0 = time.perf_counter()
+        cp.engine.save_to_memory(1, ckpt_state() | {"step": 1})
+        cp.engine.wait_saving()
         drain_estimate = time.perf_counter() - t0
-        hb(f"ckpt segment sized; full drain {drain_estimate:.2f} s")
+        hb(f"ckpt segment sized; steady-state drain {drain_estimate:.2f} s")
     else:
         drain_estimate = 0.0
     sync()

@@ -32,6 +32,7 @@ void cross_entropy_launch(void*, const void*, void*, long long, int, int,
                           float, int, void*);
 void mfma16_probe_launch(const void*, const void*, void*, void*);
 void mfma32_probe_launch(const void*, const void*, void*, void*);
+void tr_b16_probe3_launch(void*, void*);
 void tr_b16_probe_launch(void*, void*);
 void flash_attn_fwd_launch(const void*, const void*, const void*, void*,
                            void*, int, int, int, int, float,
@@ -442,6 +443,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   }, "ds_read_b64_tr_b16 semantics probe");
   m.def("mfma16_probe", &mfma16_probe, "MFMA 16x16x32 bf16 layout self-test");
   m.def("mfma32_probe", &mfma32_probe, "MFMA 32x32x16 bf16 layout self-test");
+  m.def("tr_b16_probe3", []() {
+    auto out = at::empty({64, 12}, at::TensorOptions()
+                                      .dtype(at::kShort)
+                                      .device(at::kCUDA));
+    tr_b16_probe3_launch(out.data_ptr(), cur_stream());
+    return out;
+  }, "tr_b16 cooperative tile-sourcing probe (raw index bits)");
   m.def("flash_attn_fwd", &flash_attn_fwd,
         "flash attention forward (bf16, causal, GQA, D=128) -> (out, lse)");
   m.def("flash_attn_bwd", &flash_attn_bwd,

@@ -131,3 +131,45 @@ extern "C" void tr_b16_probe_launch(void* out, hipStream_t stream) {
   hipLaunchKernelGGL(tr_b16_probe_kernel, dim3(1), dim3(64), 0, stream,
                      (short*)out);
 }
+
+// ---- tr_b16 probe r3: COOPERATIVE tile sourcing ----------------------------
+//
+// Round-1 established: per-lane bits 1-2 select the column of a 4x4 tile
+// (rows at +0,+8,+16,+24 bytes), tile-high bits are NOT taken per-lane, and
+// with mixed high bits the HW sources them "cooperatively". AITER/HK issue
+// tr reads with per-lane-VARYING tile addresses productively (192 reads per
+// loop, not 4x that), so the cooperative mode must be structured. This probe
+// maps it exactly: LDS word i holds raw bits i, so each returned value
+// identifies (tile, row, col) it came from.
+//   read A: tile base varies per LANE   (addr = lane*32)       — full map
+//   read B: tile base varies per QUAD   (addr = (lane>>2)*32 + (lane&3)*2)
+//   read C: base uniform per 16-group, column per lane, offset immediate
+__global__ void tr_b16_probe3_kernel(short* __restrict__ out) {
+  __shared__ short lds[2048];
+  for (int i = threadIdx.x; i < 2048; i += blockDim.x) lds[i] = (short)i;
+  __syncthreads();
+  const int lane = threadIdx.x & 63;
+  typedef __attribute__((ext_vector_type(2))) unsigned int u32x2;
+  const unsigned base = (unsigned)(uintptr_t)&lds[0];
+  const unsigned addrA = base + (unsigned)lane * 32u;
+  const unsigned addrB = base + (unsigned)(lane >> 2) * 32u + (lane & 3) * 2u;
+  const unsigned addrC = base + (unsigned)(lane >> 4) * 512u + (lane & 3) * 2u;
+  u32x2 rA, rB, rC;
+  asm volatile("ds_read_b64_tr_b16 %0, %3\n\t"
+               "ds_read_b64_tr_b16 %1, %4\n\t"
+               "ds_read_b64_tr_b16 %2, %5 offset:64\n\t"
+               "s_waitcnt lgkmcnt(0)"
+               : "=v"(rA), "=v"(rB), "=v"(rC)
+               : "v"(addrA), "v"(addrB), "v"(addrC));
+  short vals[12];
+  *reinterpret_cast<u32x2*>(&vals[0]) = rA;
+  *reinterpret_cast<u32x2*>(&vals[4]) = rB;
+  *reinterpret_cast<u32x2*>(&vals[8]) = rC;
+#pragma unroll
+  for (int j = 0; j < 12; ++j) out[lane * 12 + j] = vals[j];
+}
+
+extern "C" void tr_b16_probe3_launch(void* out, hipStream_t stream) {
+  hipLaunchKernelGGL(tr_b16_probe3_kernel, dim3(1), dim3(64), 0, stream,
+                     (short*)out);
+}

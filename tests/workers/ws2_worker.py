@@ -18,7 +18,9 @@ def main():
     mode = sys.argv[1]
     rank = int(os.environ["RANK"])
     world = int(os.environ["WORLD_SIZE"])
-    dev = torch.device("cuda:0")
+    # distinct devices when available (e.g. CPX partition: 1 MI355X -> 8
+    # logical XCD devices); RCCL refuses two ranks on ONE device
+    dev = torch.device(f"cuda:{rank % torch.cuda.device_count()}")
     torch.cuda.set_device(dev)
     dist.init_process_group("nccl", rank=rank, world_size=world,
                             device_id=dev)
