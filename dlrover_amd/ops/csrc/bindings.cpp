@@ -387,6 +387,30 @@ at::Tensor mfma16_probe(const at::Tensor& A, const at::Tensor& B) {
   return C;
 }
 
+// Direct async copies between device tensors and FOREIGN host memory
+// (hipHostRegister'ed shm mappings). torch's copy_ cannot be trusted here:
+// it may not recognize the registered pointer as pinned and fall back to a
+// synchronous null-stream hipMemcpy, which serializes with ALL compute on
+// the default stream (measured: each 104-GB ckpt drain added its full ~2 s
+// to the training wall instead of overlapping). These run on the CURRENT
+// torch stream — call inside `with torch.cuda.stream(side)`.
+void memcpy_d2h_async(int64_t dst_addr, const at::Tensor& src) {
+  TORCH_CHECK(src.is_cuda() && src.is_contiguous(), "src must be cuda+contig");
+  auto err = hipMemcpyAsync((void*)dst_addr, src.data_ptr(), src.nbytes(),
+                            hipMemcpyDeviceToHost,
+                            (hipStream_t)cur_stream());
+  TORCH_CHECK(err == hipSuccess, "hipMemcpyAsync D2H: ", hipGetErrorString(err));
+}
+
+void memcpy_h2d_async(at::Tensor& dst, int64_t src_addr, int64_t nbytes) {
+  TORCH_CHECK(dst.is_cuda() && dst.is_contiguous(), "dst must be cuda+contig");
+  TORCH_CHECK(nbytes <= (int64_t)dst.nbytes(), "overflow");
+  auto err = hipMemcpyAsync(dst.data_ptr(), (const void*)src_addr, nbytes,
+                            hipMemcpyHostToDevice,
+                            (hipStream_t)cur_stream());
+  TORCH_CHECK(err == hipSuccess, "hipMemcpyAsync H2D: ", hipGetErrorString(err));
+}
+
 at::Tensor mfma32_probe(const at::Tensor& A, const at::Tensor& B) {
   check_bf16(A, "A");
   check_bf16(B, "B");
@@ -443,6 +467,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   }, "ds_read_b64_tr_b16 semantics probe");
   m.def("mfma16_probe", &mfma16_probe, "MFMA 16x16x32 bf16 layout self-test");
   m.def("mfma32_probe", &mfma32_probe, "MFMA 32x32x16 bf16 layout self-test");
+  m.def("memcpy_d2h_async", &memcpy_d2h_async,
+        "hipMemcpyAsync device tensor -> registered host address");
+  m.def("memcpy_h2d_async", &memcpy_h2d_async,
+        "hipMemcpyAsync registered host address -> device tensor");
   m.def("tr_b16_probe3", []() {
     auto out = at::empty({64, 12}, at::TensorOptions()
                                       .dtype(at::kShort)

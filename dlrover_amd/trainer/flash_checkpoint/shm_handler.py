@@ -435,9 +435,21 @@ class _DeviceStager:
         dst = torch.from_numpy(host_view)
         if not self._chunked:
             with torch.cuda.stream(self._stream):
-                dst[: self._payload].copy_(
-                    self._buf[: self._payload], non_blocking=pinned
-                )
+                if pinned:
+                    # direct hipMemcpyAsync into the REGISTERED shm mapping:
+                    # torch's copy_ does not recognize foreign-pinned memory
+                    # and degrades to a synchronous null-stream hipMemcpy
+                    # that serializes with training compute (measured: each
+                    # drain added its full ~2 s to the step wall)
+                    from dlrover_amd.ops.api import hip_ops
+
+                    hip_ops().memcpy_d2h_async(
+                        host_view.ctypes.data, self._buf[: self._payload]
+                    )
+                else:
+                    dst[: self._payload].copy_(
+                        self._buf[: self._payload], non_blocking=False
+                    )
                 evt.record(self._stream)
             return evt
         # chunked fallback: serialize tensor copies through the small buffer.

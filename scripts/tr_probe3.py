@@ -9,6 +9,8 @@ Reads:
   C: addr = (lane>>4)*512 + (lane&3)*2, offset:64    (tile per GROUP+imm)
 """
 
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch  # noqa: F401
 
 from dlrover_amd.ops.api import hip_ops
