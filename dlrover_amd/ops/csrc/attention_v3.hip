@@ -246,14 +246,13 @@ __global__ __launch_bounds__(512) void flash_attn_fwd_v3_kernel(
           unsigned w1 = pack_bf16(p[r0 + 2], p[r0 + 3]);
           unsigned w2 = pack_bf16(p[r0 + 4], p[r0 + 5]);
           unsigned w3 = pack_bf16(p[r0 + 6], p[r0 + 7]);
+          // swap(vdst, vsrc) exchanges vdst's lanes 32-63 with vsrc's 0-31:
+          //   .x (new w0): h0 lanes own kv(+0,1); h1 lanes partner kv(+8,9)
+          //   .y (new w2): h0 lanes partner kv(+4,5); h1 lanes own kv(+12,13)
+          // -> frag order [x0, x1, y0, y1] is the SAME for both halves
           u32x2_t s0 = __builtin_amdgcn_permlane32_swap(w0, w2, false, false);
           u32x2_t s1 = __builtin_amdgcn_permlane32_swap(w1, w3, false, false);
-          u32x4_t frag;
-          if (half == 0) {
-            frag = u32x4_t{s0.x, s1.x, s0.y, s1.y};
-          } else {
-            frag = u32x4_t{s0.y, s1.y, s0.x, s1.x};
-          }
+          u32x4_t frag = u32x4_t{s0.x, s1.x, s0.y, s1.y};
           bp[n * 2 + ks2] = *reinterpret_cast<bf16x8*>(&frag);
         }
       }
@@ -284,47 +283,32 @@ __global__ __launch_bounds__(512) void flash_attn_fwd_v3_kernel(
     }
   }
 
-  // ---- epilogue: normalize and store O (+ logsumexp) ----
+  // ---- epilogue: normalize, bounce through LDS, coalesced b128 stores ----
+  // O^T register layout: lane owns q = q_row_w, d = n*32+(r&3)+8*(r>>2)+4h —
+  // direct global stores would be 2-byte scatters. The kv-loop LDS is free
+  // now: 4 waves at a time write their 32x128 band into an 8 KB swizzled
+  // slab, then read it back row-major and store b128.
   const float inv = l_run > 0.f ? 1.f / l_run : 0.f;
   short* o_blk = out + (long long)b * os_b + (long long)h * os_h +
                  (long long)(qt * FA3_QB) * os_s;
-  // O^T layout: lane owns q = q_row_w, d = n*32 + (r&3)+8*(r>>2)+4*half.
-  // Bounce through LDS (k_lds is free) so the global store is b128
-  // row-major: wave w uses k_lds[w&1] + (w>>1)*... -> 8 KB per wave, the
-  // 32 KB k_lds+vt_lds area holds 4 waves' bands at a time; do it in two
-  // rounds to stay in bounds.
-  char* bounce = (wave < 4 ? k_lds[0] : vt_lds[0]);
-  char* my_slab = bounce + (wave & 3) * (FA3_QW * FA3_D * 2 / 4) * 0;  // see below
-  (void)my_slab;
-  // simpler: two rounds over waves; each round 4 waves write 8 KB slabs
+  char* slab = (char*)k_lds + (wave & 3) * (FA3_QW * FA3_D * 2);
   for (int round = 0; round < 2; ++round) {
     __syncthreads();
     if ((wave >> 2) == round) {
-      char* slab = (wave < 4 ? k_lds[0] : vt_lds[0]) +
-                   ((wave & 3) ? 0 : 0);
-      slab = (round == 0 ? k_lds[0] : k_lds[0]);  // base
-      slab = k_lds[0] + (wave & 3) * (FA3_QW * FA3_D * 2);
 #pragma unroll
       for (int n = 0; n < 4; ++n) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int d = n * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
-          *reinterpret_cast<__bf16*>(
-              slab + qcol * 256 + swz3(qcol, d * 2)) =
+          *reinterpret_cast<__bf16*>(slab + qcol * 256 + swz3(qcol, d * 2)) =
               (__bf16)(acc_o[n][r] * inv);
         }
       }
     }
     __syncthreads();
     if ((wave >> 2) == round) {
-      char* slab = k_lds[0] + (wave & 3) * (FA3_QW * FA3_D * 2);
-      // row-major b128 stores: thread's lane covers 16 elems of its band
-      const int row = (lane * 16) / FA3_D + half * 0;  // 0..3 per 8 lanes
-      // each lane stores 2 x b128 of row `lrow`, cols lcol..lcol+16
-      const int lrow = (lane * 16) / FA3_D;        // needs 32 rows: use all
-      (void)row;
-      // 64 lanes x 16 elems = 1024 elems per pass; band = 32*128 = 4096
-      // -> 4 passes
+      // 64 lanes x 16 elems = 2 rows of 128 per pass; 32-row band -> 4
+      // passes of 2 x b128 per lane
 #pragma unroll
       for (int pass = 0; pass < 4; ++pass) {
         const int elem0 = (pass * 64 + lane) * 16;
@@ -338,7 +322,6 @@ __global__ __launch_bounds__(512) void flash_attn_fwd_v3_kernel(
         *reinterpret_cast<bf16x8*>(dst) = v0;
         *reinterpret_cast<bf16x8*>(dst + 8) = v1;
       }
-      (void)lrow;
     }
   }
   if (half == 0 && lse != nullptr) {

@@ -37,6 +37,9 @@ void tr_b16_probe_launch(void*, void*);
 void flash_attn_fwd_launch(const void*, const void*, const void*, void*,
                            void*, int, int, int, int, float,
                            const long long*, void*);
+void flash_attn_fwd_v3_launch(const void*, const void*, const void*, void*,
+                              void*, int, int, int, int, float,
+                              const long long*, void*);
 void flash_attn_fwd_v2_launch(const void*, const void*, const void*, void*,
                               void*, int, int, int, int, float,
                               const long long*, void*);
@@ -293,7 +296,18 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q,
     const char* e = getenv("DLROVER_FA_V2");
     return e != nullptr && e[0] == '1';
   }();
-  if (use_v2) {
+  // V3 (default when shapes allow): 32x32x16 MFMA, swapped operands,
+  // in-register online softmax — attention_v3.hip. DLROVER_FA_V3=0 falls
+  // back to v1 (16x16x32 + LDS-staged P).
+  static const bool v3_enabled = []() {
+    const char* e = getenv("DLROVER_FA_V3");
+    return e == nullptr || e[0] != '0';
+  }();
+  if (v3_enabled && !use_v2 && S % 256 == 0) {
+    flash_attn_fwd_v3_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                             out_bshd.data_ptr(), lse.data_ptr(), B, H, HKV,
+                             S, (float)scale, st, cur_stream());
+  } else if (use_v2) {
     flash_attn_fwd_v2_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                              out_bshd.data_ptr(), lse.data_ptr(), B, H, HKV,
                              S, (float)scale, st, cur_stream());

@@ -27,6 +27,7 @@ sources = [
     os.path.join(CSRC, "cross_entropy.hip"),
     os.path.join(CSRC, "mfma_probe.hip"),
     os.path.join(CSRC, "attention.hip"),
+    os.path.join(CSRC, "attention_v3.hip"),
     os.path.join(CSRC, "attention_bwd.hip"),
 ]
 
