@@ -338,10 +338,89 @@ class ElasticTrainingAgent(LocalElasticAgent):
                         "(does not consume the restart budget)",
                         num_waiting,
                     )
-                    self._restart_workers(self._worker_group)
+                    self._membership_change_restart()
                 continue
 
             raise RuntimeError(f"unknown worker state {state}")
+
+    # -- membership change (ref: training.py:1687 -> graceful stop :779,
+    # -- UCP persist :1548-1651, orphan sweep :800) ---------------------------
+
+    def _membership_change_restart(self):
+        """World size is about to change. Order matters:
+        1. persist every local rank's committed shm snapshot to DISK — the
+           new world's UCP reshard (flash_checkpoint/ucp.load_resharded)
+           needs ALL old shards on storage, and the old shm layout dies
+           with the old processes;
+        2. stop workers gracefully (pcontext SIGTERM -> grace -> SIGKILL);
+        3. sweep orphaned descendants (dataloader workers etc. holding HIP
+           contexts would pin GPU memory into the new incarnation);
+        4. re-rendezvous + start the new worker group.
+        """
+        self._save_ckpt_to_storage()
+        descendants = self._worker_descendants()
+        self._stop_workers(self._worker_group)
+        self._sweep_orphans(descendants)
+        self._restart_workers(self._worker_group)
+
+    def _worker_descendants(self) -> List[int]:
+        """Exact PIDs of worker processes and their live descendants,
+        recorded BEFORE the stop (never kill by pattern — only these)."""
+        try:
+            roots = [int(p) for p in dict(self._pcontext.pids()).values()]
+        except Exception:  # noqa: BLE001
+            return []
+        children: Dict[int, List[int]] = {}
+        try:
+            for pid_dir in os.listdir("/proc"):
+                if not pid_dir.isdigit():
+                    continue
+                try:
+                    with open(f"/proc/{pid_dir}/status") as f:
+                        for line in f:
+                            if line.startswith("PPid:"):
+                                ppid = int(line.split()[1])
+                                children.setdefault(ppid, []).append(int(pid_dir))
+                                break
+                except OSError:
+                    continue
+        except OSError:
+            return roots
+        out: List[int] = []
+        stack = list(roots)
+        while stack:
+            p = stack.pop()
+            out.append(p)
+            stack.extend(children.get(p, []))
+        return out
+
+    def _sweep_orphans(self, pids: List[int], grace: float = 3.0):
+        """After _stop_workers, kill any still-alive recorded descendant.
+        HIP contexts make this MORE important than on the reference's
+        stack: an orphaned dataloader holding a context pins GPU memory
+        (SURVEY §7 hard-part b)."""
+        import signal as _signal
+
+        alive = [p for p in pids if os.path.exists(f"/proc/{p}")]
+        if not alive:
+            return
+        for p in alive:
+            try:
+                os.kill(p, _signal.SIGTERM)
+            except OSError:
+                pass
+        deadline = time.time() + grace
+        while time.time() < deadline and any(
+            os.path.exists(f"/proc/{p}") for p in alive
+        ):
+            time.sleep(0.2)
+        for p in alive:
+            if os.path.exists(f"/proc/{p}"):
+                logger.warning("orphan worker descendant pid=%s: SIGKILL", p)
+                try:
+                    os.kill(p, _signal.SIGKILL)
+                except OSError:
+                    pass
 
     def _dump_worker_py_stacks(self):
         """Before a hang-driven restart kills the workers, capture where each
