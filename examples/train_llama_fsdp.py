@@ -68,8 +68,13 @@ def main():
     with device:
         model = LlamaForCausalLM(cfg)
     model = model.to(device)
+    use_fsdp = use_gpu or os.getenv("DLROVER_FSDP_CPU", "") == "1"
     if use_gpu:
         model = model.bfloat16()
+    if use_fsdp:
+        # fully_shard even at world 1: DTensor shard0 tags keep checkpoints
+        # UCP-reshardable in BOTH scale directions (grow needs the ws-1 ckpt
+        # tagged shard0, not replicated)
         from torch.distributed.fsdp import fully_shard
 
         for blk in model.blocks:
@@ -111,6 +116,8 @@ def main():
                     "step": step, "loss": round(float(loss.item()), 4),
                     "step_s": round(time.perf_counter() - t0, 3),
                     "incarnation": incarnation, "resumed_from": start_step,
+                    "world": dist.get_world_size(),
+                    "device": str(device),
                 }) + "\n")
         if kill_at and step == kill_at and incarnation == 0 and rank == 0:
             print(f"[train] injecting SIGKILL at step {step}", flush=True)
