@@ -115,6 +115,7 @@ def main():
                 f.write(json.dumps({
                     "step": step, "loss": round(float(loss.item()), 4),
                     "step_s": round(time.perf_counter() - t0, 3),
+                    "ts": round(time.time(), 3),
                     "incarnation": incarnation, "resumed_from": start_step,
                     "world": dist.get_world_size(),
                     "device": str(device),

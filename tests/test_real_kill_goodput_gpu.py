@@ -1,0 +1,35 @@
+"""Real-SIGKILL goodput measurement path on hardware:
+the kill, agent detection, worker respawn, RCCL re-init and checkpoint
+restore all land inside the measured window."""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.mark.timeout(540)
+def test_real_kill_goodput_small(tmp_path):
+    out = tmp_path / "rk.json"
+    proc = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "scripts", "bench_real_kill.py"),
+         "--model", "small_1b", "--seq", "2048", "--batch", "1",
+         "--steps", "24", "--kill-at", "12", "--ckpt-interval", "4",
+         "--out", str(out)],
+        cwd=ROOT, capture_output=True, text=True, timeout=480,
+    )
+    assert proc.returncode == 0, proc.stderr[-4000:]
+    r = json.loads(out.read_text())
+    assert r["steps"] == 24
+    assert r["incarnations"] == [0, 1], r
+    assert r["resumed_from"] >= 4, r
+    assert 0 < r["value"] <= 100.0
+    # recovery (death -> trained-again) must be bounded: agent detect (~1s
+    # monitor) + respawn + RCCL init + 1B model build + restore
+    assert r["recovery_s"] < 120, r
