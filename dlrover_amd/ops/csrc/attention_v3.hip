@@ -26,6 +26,7 @@
 //
 // The reference (DLRover) has no attention kernel of its own — this is the
 // MI355X-native hot path (BASELINE.json north star).
+#include <cstdlib>
 #include "kern_common.h"
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
@@ -51,6 +52,38 @@ __device__ __forceinline__ unsigned pack_bf16(float lo, float hi) {
   return *reinterpret_cast<unsigned*>(&v);
 }
 
+// A-operand fragment X^T[d = n*32 + (l&31)][row = rb + half*8 + i] read from
+// a ROW-major swz3-swizzled [rows][128] bf16 LDS tile via two cooperative
+// ds_read_b64_tr_b16 gathers. Semantics probe-pinned (profiles/r02a): the
+// 16-lane group's addresses form a gather table — lane l supplies the word
+// address for (row rb + ((l&15)>>2), d-quad base+(l&3)) and receives, as
+// element j, its OWN d column of row rb + j (slots 4j + ((l>>2)&3)).
+__device__ __forceinline__ bf16x8 tr_colT_frag(const char* lds, int rb,
+                                               int n, int lane, int half) {
+  typedef __attribute__((ext_vector_type(2))) unsigned int u32x2_;
+  const int j = (lane & 15) >> 2;
+  const int dq = n * 8 + ((lane & 16) >> 2) + (lane & 3);
+  const int r0 = rb + half * 8 + j;
+  const int r1 = r0 + 4;
+  const unsigned base = (unsigned)(uintptr_t)lds;
+  const unsigned a0 = base + r0 * 256 + ((dq * 8) ^ ((r0 & 7) << 4));
+  const unsigned a1 = base + r1 * 256 + ((dq * 8) ^ ((r1 & 7) << 4));
+  u32x2_ w0, w1;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=v"(w0), "=v"(w1)
+      : "v"(a0), "v"(a1));
+  u32x4_t frag = u32x4_t{w0.x, w0.y, w1.x, w1.y};
+  return *reinterpret_cast<bf16x8*>(&frag);
+}
+
+// TRV=false: V staged TRANSPOSED (vt_lds, per-element scatter writes — the
+//   measured bank-conflict source) and PV A-operand read b128.
+// TRV=true ("v4"): V staged ROW-major like K (conflict-free b128 writes) and
+//   the V^T A-operand gathered with ds_read_b64_tr_b16 (tr_colT_frag).
+template <bool TRV>
 __global__ __launch_bounds__(512) void flash_attn_fwd_v3_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ out,
@@ -125,12 +158,19 @@ __global__ __launch_bounds__(512) void flash_attn_fwd_v3_kernel(
       const int col = st_col0 + c * 8;
       *reinterpret_cast<bf16x8*>(kb + st_row0 * 256 + swz3(st_row0, col * 2)) =
           st_k[c];
-      // V transposed: element (row, col+i) -> vt[col+i][row]
+      if (TRV) {
+        // row-major like K: conflict-free b128 store; the transpose happens
+        // on READ via the tr_b16 cooperative gather
+        *reinterpret_cast<bf16x8*>(vb + st_row0 * 256 +
+                                   swz3(st_row0, col * 2)) = st_v[c];
+      } else {
+        // V transposed: element (row, col+i) -> vt[col+i][row]
 #pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        const int trow = col + i;
-        *reinterpret_cast<__bf16*>(vb + trow * 128 + swz3(trow, st_row0 * 2)) =
-            st_v[c][i];
+        for (int i = 0; i < 8; ++i) {
+          const int trow = col + i;
+          *reinterpret_cast<__bf16*>(vb + trow * 128 +
+                                     swz3(trow, st_row0 * 2)) = st_v[c][i];
+        }
       }
     }
   };
@@ -264,7 +304,9 @@ __global__ __launch_bounds__(512) void flash_attn_fwd_v3_kernel(
         for (int ks = 0; ks < 4; ++ks) {  // kv k-slices of 16
           // A = V^T[d = n*32 + (lane&31)][kv = ks*16 + half*8 + i]
           bf16x8 av;
-          {
+          if (TRV) {
+            av = tr_colT_frag(vt_cur, ks * 16, n, lane, half);
+          } else {
             const int row = n * 32 + qcol;
             const int byte_col = (ks * 16 + half * 8) * 2;
             av = *reinterpret_cast<const bf16x8*>(
@@ -336,10 +378,26 @@ extern "C" void flash_attn_fwd_v3_launch(const void* q, const void* k,
                                          float scale, const long long* strides,
                                          hipStream_t stream) {
   dim3 grid(S / FA3_QB, H, B);
-  hipLaunchKernelGGL(flash_attn_fwd_v3_kernel, grid, dim3(512), 0, stream,
-                     (const short*)q, (const short*)k, (const short*)v,
-                     (short*)out, (float*)lse, B, H, HKV, S, scale,
-                     strides[0], strides[1], strides[2], strides[3],
-                     strides[4], strides[5], strides[6], strides[7],
-                     strides[8], strides[9], strides[10], strides[11]);
+  // v4 (tr-read V, conflict-free staging) unless DLROVER_FA_TRV=0
+  static const bool trv = []() {
+    const char* e = getenv("DLROVER_FA_TRV");
+    return e == nullptr || e[0] != '0';
+  }();
+  if (trv) {
+    hipLaunchKernelGGL((flash_attn_fwd_v3_kernel<true>), grid, dim3(512), 0,
+                       stream, (const short*)q, (const short*)k,
+                       (const short*)v, (short*)out, (float*)lse, B, H, HKV,
+                       S, scale, strides[0], strides[1], strides[2],
+                       strides[3], strides[4], strides[5], strides[6],
+                       strides[7], strides[8], strides[9], strides[10],
+                       strides[11]);
+  } else {
+    hipLaunchKernelGGL((flash_attn_fwd_v3_kernel<false>), grid, dim3(512), 0,
+                       stream, (const short*)q, (const short*)k,
+                       (const short*)v, (short*)out, (float*)lse, B, H, HKV,
+                       S, scale, strides[0], strides[1], strides[2],
+                       strides[3], strides[4], strides[5], strides[6],
+                       strides[7], strides[8], strides[9], strides[10],
+                       strides[11]);
+  }
 }
