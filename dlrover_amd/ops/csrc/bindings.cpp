@@ -48,6 +48,12 @@ void fa_bwd_pre_launch(const void*, const void*, void*, int, int, int,
 void fa_bwd_dq_launch(const void*, const void*, const void*, const void*,
                       const void*, const void*, void*, int, int, int, int,
                       float, const long long*, void*);
+void fa_bwd_dq_v3_launch(const void*, const void*, const void*, const void*,
+                         const void*, const void*, void*, int, int, int, int,
+                         float, const long long*, void*);
+void fa_bwd_dkv_v3_launch(const void*, const void*, const void*, const void*,
+                          const void*, const void*, void*, void*, int, int,
+                          int, int, float, const long long*, void*);
 void fa_bwd_dkv_launch(const void*, const void*, const void*, const void*,
                        const void*, const void*, void*, void*, int, int, int,
                        int, float, const long long*, void*);
@@ -369,16 +375,34 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> flash_attn_bwd(
         quse = qv;
         pack_strides(stq, quse, 0);
       }
-      fa_bwd_dq_launch(quse.data_ptr(), k.data_ptr(), v.data_ptr(),
-                       dout.data_ptr(), lse.data_ptr(), dvec.data_ptr(),
-                       dq_bshd.data_ptr(), B, H, HKV, S, (float)scale, stq,
-                       cur_stream());
+      // v3 backward (32x32 MFMA swapped rebuild) when shapes allow;
+      // DLROVER_FA_BWD_V3=0 reverts to the 16x16 kernels
+      static const bool bwd_v3 = []() {
+        const char* e = getenv("DLROVER_FA_BWD_V3");
+        return e == nullptr || e[0] != '0';
+      }();
       auto dk32 = at::zeros({B, HKV, S, D}, k.options().dtype(at::kFloat));
       auto dv32 = at::zeros({B, HKV, S, D}, v.options().dtype(at::kFloat));
-      fa_bwd_dkv_launch(quse.data_ptr(), k.data_ptr(), v.data_ptr(),
-                        dout.data_ptr(), lse.data_ptr(), dvec.data_ptr(),
-                        dk32.data_ptr(), dv32.data_ptr(), B, H, HKV, S,
-                        (float)scale, stq, cur_stream());
+      if (bwd_v3 && S % 256 == 0) {
+        fa_bwd_dq_v3_launch(quse.data_ptr(), k.data_ptr(), v.data_ptr(),
+                            dout.data_ptr(), lse.data_ptr(), dvec.data_ptr(),
+                            dq_bshd.data_ptr(), B, H, HKV, S, (float)scale,
+                            stq, cur_stream());
+        fa_bwd_dkv_v3_launch(quse.data_ptr(), k.data_ptr(), v.data_ptr(),
+                             dout.data_ptr(), lse.data_ptr(),
+                             dvec.data_ptr(), dk32.data_ptr(),
+                             dv32.data_ptr(), B, H, HKV, S, (float)scale,
+                             stq, cur_stream());
+      } else {
+        fa_bwd_dq_launch(quse.data_ptr(), k.data_ptr(), v.data_ptr(),
+                         dout.data_ptr(), lse.data_ptr(), dvec.data_ptr(),
+                         dq_bshd.data_ptr(), B, H, HKV, S, (float)scale, stq,
+                         cur_stream());
+        fa_bwd_dkv_launch(quse.data_ptr(), k.data_ptr(), v.data_ptr(),
+                          dout.data_ptr(), lse.data_ptr(), dvec.data_ptr(),
+                          dk32.data_ptr(), dv32.data_ptr(), B, H, HKV, S,
+                          (float)scale, stq, cur_stream());
+      }
       auto dk = at::empty({B, HKV, S, D}, k.options());
       auto dv = at::empty({B, HKV, S, D}, v.options());
       f32_to_bf16_launch(dk32.data_ptr(), dk.data_ptr(), dk32.numel(),

@@ -29,6 +29,7 @@ sources = [
     os.path.join(CSRC, "attention.hip"),
     os.path.join(CSRC, "attention_v3.hip"),
     os.path.join(CSRC, "attention_bwd.hip"),
+    os.path.join(CSRC, "attention_bwd_v3.hip"),
 ]
 
 setup(
