@@ -34,6 +34,7 @@
 #include <chrono>
 #include <deque>
 #include <map>
+#include <set>
 #include <mutex>
 #include <string>
 #include <vector>
@@ -104,6 +105,16 @@ struct PendingOp {
   double bytes;
   double enqueue_ts;
   const char* name;  // owned by the HIP runtime (kernel symbol) or static
+  double busbytes;   // collective bus-bytes (bytes x op/world factor)
+  double flops;      // GEMM flops for TFLOPS attribution
+};
+
+// one retired op in the bounded kernel-trace ring (timeline dump)
+struct TraceEvent {
+  const char* name;
+  int cat;
+  double ts;      // host enqueue time (s)
+  double dur_ms;  // GPU duration
 };
 
 struct CatStats {
@@ -157,12 +168,21 @@ class Manager {
 
   void record_end(hipStream_t stream, Category cat, double bytes,
                   hipEvent_t start, hipEvent_t stop,
-                  const char* name = nullptr) {
+                  const char* name = nullptr, double busbytes = 0,
+                  double flops = 0) {
     if (!enabled_ || start == nullptr) return;
     (void)hipEventRecord(stop, stream);
     std::lock_guard<std::mutex> g(q_mu_);
-    pending_.push_back({start, stop, cat, bytes, now(), name});
+    pending_.push_back({start, stop, cat, bytes, now(), name, busbytes, flops});
   }
+
+  // stable owned name strings (per-GEMM-shape labels outlive the call)
+  const char* intern(const std::string& s) {
+    std::lock_guard<std::mutex> g(intern_mu_);
+    return interned_.insert(s).first->c_str();
+  }
+
+  void count_gemm_flops(double flops) { atomic_add(gemm_flops_, flops); }
 
   // ---- RCCL communicator introspection (ref behavior: the reference's
   // nccl_parser extracts comm world size/rank + per-comm traffic so the
@@ -187,6 +207,12 @@ class Manager {
     std::lock_guard<std::mutex> g(comm_mu_);
     auto it = comms_.find(c);
     if (it != comms_.end()) it->second.alive = false;  // keep stats visible
+  }
+
+  int comm_nranks(void* c) {
+    std::lock_guard<std::mutex> g(comm_mu_);
+    auto it = comms_.find(c);
+    return it != comms_.end() ? it->second.nranks : 0;
   }
 
   void count_comm_call(void* c, double bytes) {
@@ -251,6 +277,7 @@ class Manager {
         dump_metrics();
         last_dump = t;
       }
+      maybe_dump_timeline();
     }
   }
 
@@ -282,6 +309,18 @@ class Manager {
           e.first += w;
           e.second += (double)ms * w;
         }
+        if (op.cat == CAT_COMM && op.busbytes > 0) {
+          atomic_add(comm_busbytes_, op.busbytes);
+          comm_busms_ += (double)ms;
+        }
+        if (op.cat == CAT_GEMM) {
+          gemm_ms_timed_ += (double)ms * w;
+        }
+        // bounded kernel-trace ring (timeline dump, ref manager.h:50-62)
+        trace_ring_[trace_head_ % kTraceRing] = {
+            op.name ? op.name : kCatNames[op.cat], op.cat, op.enqueue_ts,
+            (double)ms};
+        trace_head_ += 1;
       }
       last_completion_ = now();
       {
@@ -379,12 +418,65 @@ class Manager {
                 kv.second.alive ? 1 : 0, kv.second.bytes);
       }
     }
+    // derived throughput metrics
+    {
+      double bus = comm_busbytes_.load();
+      double bms = comm_busms_;
+      fprintf(f, "hiptimer_comm_busbytes_total %.0f\n", bus);
+      fprintf(f, "hiptimer_comm_busbw_gbs %.3f\n",
+              bms > 0 ? bus / (bms / 1e3) / 1e9 : 0.0);
+      double gf = gemm_flops_.load();
+      double gms = gemm_ms_timed_;
+      fprintf(f, "hiptimer_gemm_flops_total %.0f\n", gf);
+      fprintf(f, "hiptimer_gemm_tflops %.2f\n",
+              gms > 0 ? gf / (gms / 1e3) / 1e12 : 0.0);
+    }
     fprintf(f, "hiptimer_device_alloc_bytes %.0f\n", (double)alloc_bytes_.load());
     fprintf(f, "hiptimer_device_free_total %ld\n", free_count_.load());
     fprintf(f, "hiptimer_host_alloc_bytes %.0f\n",
             (double)host_alloc_bytes_.load());
     fclose(f);
     rename(tmp, path);
+  }
+
+  // timeline: dump the trace ring as chrome-trace JSON (perfetto-loadable)
+  // when <metrics_dir>/dump_timeline_<rank> appears, or automatically on the
+  // first hang detection (ref: KernelTraceManager ring + gen_trace_timeline)
+  void maybe_dump_timeline() {
+    char flag[512];
+    snprintf(flag, sizeof(flag), "%s/dump_timeline_%d", metrics_dir_.c_str(),
+             rank_);
+    double since;
+    bool hang_now = is_hang(&since);
+    bool flagged = access(flag, F_OK) == 0;
+    if (!flagged && !(hang_now && !hang_dumped_)) return;
+    if (hang_now) hang_dumped_ = true;
+    if (flagged) unlink(flag);
+    dump_timeline();
+  }
+
+  void dump_timeline() {
+    char path[512];
+    snprintf(path, sizeof(path), "%s/timeline_%d.json", metrics_dir_.c_str(),
+             rank_);
+    FILE* f = fopen(path, "w");
+    if (!f) return;
+    fprintf(f, "{\"traceEvents\":[\n");
+    long n = trace_head_ < (long)kTraceRing ? trace_head_ : (long)kTraceRing;
+    long first = trace_head_ - n;
+    for (long i = 0; i < n; ++i) {
+      const TraceEvent& e = trace_ring_[(first + i) % kTraceRing];
+      std::string nm = e.name ? std::string(e.name).substr(0, 160) : "?";
+      for (auto& c : nm)
+        if (c == '"' || c == '\\' || c == '\n') c = '_';
+      fprintf(f,
+              "%s{\"name\":\"%s\",\"cat\":\"%s\",\"ph\":\"X\","
+              "\"ts\":%.1f,\"dur\":%.1f,\"pid\":%d,\"tid\":%d}",
+              i ? ",\n" : "", nm.c_str(), kCatNames[e.cat], e.ts * 1e6,
+              e.dur_ms * 1e3, rank_, e.cat);
+    }
+    fprintf(f, "\n]}\n");
+    fclose(f);
   }
 
   bool enabled_ = false;
@@ -411,6 +503,16 @@ class Manager {
   std::atomic<long> free_count_{0};
   std::atomic<long> host_alloc_bytes_{0};
   std::atomic<double> last_completion_{0};
+  std::mutex intern_mu_;
+  std::set<std::string> interned_;
+  std::atomic<double> gemm_flops_{0};
+  double gemm_ms_timed_ = 0;        // poller thread only
+  std::atomic<double> comm_busbytes_{0};
+  double comm_busms_ = 0;           // poller thread only
+  static constexpr size_t kTraceRing = 8192;
+  TraceEvent trace_ring_[kTraceRing] = {};
+  long trace_head_ = 0;             // poller thread only
+  bool hang_dumped_ = false;
 };
 
 struct Scoped {
@@ -419,12 +521,16 @@ struct Scoped {
   Category cat;
   double bytes;
   const char* name = nullptr;
-  Scoped(hipStream_t s, Category c, double b, const char* n = nullptr)
-      : stream(s), cat(c), bytes(b), name(n) {
+  double busbytes = 0;
+  double flops = 0;
+  Scoped(hipStream_t s, Category c, double b, const char* n = nullptr,
+         double bus = 0, double fl = 0)
+      : stream(s), cat(c), bytes(b), name(n), busbytes(bus), flops(fl) {
     Manager::inst().record_begin(s, c, b, &start, &stop);
   }
   void finish() {
-    Manager::inst().record_end(stream, cat, bytes, start, stop, name);
+    Manager::inst().record_end(stream, cat, bytes, start, stop, name,
+                               busbytes, flops);
   }
 };
 
@@ -496,6 +602,25 @@ hipError_t hipExtModuleLaunchKernel(hipFunction_t f, unsigned gx, unsigned gy,
 
 // hipblasLtMatmul(handle, desc, alpha, A, Adesc, B, Bdesc, beta, C, Cdesc,
 //                 D, Ddesc, algo, workspace, wsSize, stream)
+// layout attribute query (hipblaslt.h): BATCH_COUNT=0, ROWS=4, COLS=5
+static void layout_dims(void* layout, long long* rows, long long* cols,
+                        int* batch) {
+  using get_t = int (*)(void*, int, void*, size_t, size_t*);
+  static get_t get = (get_t)real("hipblasLtMatrixLayoutGetAttribute");
+  *rows = *cols = 0;
+  *batch = 1;
+  if (get == nullptr || layout == nullptr) return;
+  unsigned long long r = 0, c = 0;
+  int32_t b = 1;
+  size_t written = 0;
+  get(layout, 4, &r, sizeof(r), &written);
+  get(layout, 5, &c, sizeof(c), &written);
+  get(layout, 0, &b, sizeof(b), &written);
+  *rows = (long long)r;
+  *cols = (long long)c;
+  *batch = b > 0 ? b : 1;
+}
+
 int hipblasLtMatmul(void* handle, void* matmulDesc, const void* alpha,
                     const void* A, void* Adesc, const void* B, void* Bdesc,
                     const void* beta, const void* C, void* Cdesc, void* D,
@@ -505,7 +630,22 @@ int hipblasLtMatmul(void* handle, void* matmulDesc, const void* alpha,
                        const void*, void*, const void*, const void*, void*,
                        void*, void*, const void*, void*, size_t, hipStream_t);
   static fn_t fn = (fn_t)real("hipblasLtMatmul");
-  hiptimer::Scoped sc(stream, CAT_GEMM, 0);
+  // per-GEMM m/n/k/batch -> flops + a per-shape label (ref hook.cc:253-322)
+  long long m = 0, n = 0, ar = 0, ac = 0;
+  int batch = 1, abatch = 1;
+  layout_dims(Ddesc, &m, &n, &batch);
+  layout_dims(Adesc, &ar, &ac, &abatch);
+  const long long kdim = (ar == m) ? ac : ar;
+  const double flops = 2.0 * m * n * kdim * batch;
+  const char* label = nullptr;
+  if (m > 0 && n > 0 && kdim > 0) {
+    char buf[96];
+    snprintf(buf, sizeof(buf), "gemm_m%lld_n%lld_k%lld_b%d", m, n, kdim,
+             batch);
+    label = Manager::inst().intern(buf);
+    Manager::inst().count_gemm_flops(flops);
+  }
+  hiptimer::Scoped sc(stream, CAT_GEMM, 0, label, 0, flops);
   int rc = fn(handle, matmulDesc, alpha, A, Adesc, B, Bdesc, beta, C, Cdesc, D,
               Ddesc, algo, workspace, workspaceSizeInBytes, stream);
   sc.finish();
@@ -514,15 +654,41 @@ int hipblasLtMatmul(void* handle, void* matmulDesc, const void* alpha,
 
 // ---- RCCL collectives (librccl exports nccl* names) --------------------------
 
-#define HIPTIMER_NCCL_COLL(NAME, COUNT_EXPR)                                   \
+// ncclDataType_t element sizes (nccl.h ordering; RCCL matches upstream)
+static int nccl_dtype_bytes(int dt) {
+  switch (dt) {
+    case 0: case 1: return 1;               // int8/uint8
+    case 2: case 3: return 4;               // int32/uint32
+    case 4: case 5: return 8;               // int64/uint64
+    case 6: return 2;                       // float16
+    case 7: return 4;                       // float32
+    case 8: return 8;                       // float64
+    case 9: return 2;                       // bfloat16
+    default: return 1;                      // fp8 variants / unknown
+  }
+}
+
+// bus-bytes per DLRover/nccl-tests busbw convention (node_check utils.py):
+// allreduce 2(n-1)/n x data; all-gather/reduce-scatter (n-1)/n x total data
+static double busbw_factor_allreduce(int n) {
+  return n > 0 ? 2.0 * (n - 1) / n : 0.0;
+}
+static double busbw_factor_ag_rs(int n) {
+  return n > 0 ? (double)(n - 1) / n : 0.0;
+}
+
+#define HIPTIMER_NCCL_COLL(NAME, BUS_FACTOR)                                   \
   ncclResult_t NAME(const void* sendbuff, void* recvbuff, size_t count,        \
                     ncclDataType_t dt, ncclRedOp_t op, ncclComm_t comm,        \
                     hipStream_t stream) {                                      \
     using fn_t = ncclResult_t (*)(const void*, void*, size_t, ncclDataType_t,  \
                                   ncclRedOp_t, ncclComm_t, hipStream_t);       \
     static fn_t fn = (fn_t)real(#NAME);                                        \
-    Manager::inst().count_comm_call(comm, (double)(COUNT_EXPR));               \
-    hiptimer::Scoped sc(stream, CAT_COMM, (double)(COUNT_EXPR));               \
+    const double bytes = (double)count * nccl_dtype_bytes((int)dt);            \
+    const int nr = Manager::inst().comm_nranks(comm);                          \
+    Manager::inst().count_comm_call(comm, bytes);                              \
+    hiptimer::Scoped sc(stream, CAT_COMM, bytes, #NAME,                        \
+                        bytes * (BUS_FACTOR));                                 \
     ncclResult_t rc = fn(sendbuff, recvbuff, count, dt, op, comm, stream);     \
     sc.finish();                                                               \
     return rc;                                                                 \
@@ -566,8 +732,8 @@ ncclResult_t ncclCommAbort(ncclComm_t comm) {
   return fn(comm);
 }
 
-HIPTIMER_NCCL_COLL(ncclAllReduce, count)
-HIPTIMER_NCCL_COLL(ncclReduce, count)
+HIPTIMER_NCCL_COLL(ncclAllReduce, busbw_factor_allreduce(nr))
+HIPTIMER_NCCL_COLL(ncclReduce, 1.0)
 
 ncclResult_t ncclAllGather(const void* sendbuff, void* recvbuff,
                            size_t sendcount, ncclDataType_t dt, ncclComm_t comm,
@@ -575,8 +741,11 @@ ncclResult_t ncclAllGather(const void* sendbuff, void* recvbuff,
   using fn_t = ncclResult_t (*)(const void*, void*, size_t, ncclDataType_t,
                                 ncclComm_t, hipStream_t);
   static fn_t fn = (fn_t)real("ncclAllGather");
-  Manager::inst().count_comm_call(comm, (double)sendcount);
-  hiptimer::Scoped sc(stream, CAT_COMM, (double)sendcount);
+  const int nr = Manager::inst().comm_nranks(comm);
+  const double bytes = (double)sendcount * nccl_dtype_bytes((int)dt) * (nr > 0 ? nr : 1);
+  Manager::inst().count_comm_call(comm, bytes);
+  hiptimer::Scoped sc(stream, CAT_COMM, bytes, "ncclAllGather",
+                      bytes * busbw_factor_ag_rs(nr));
   ncclResult_t rc = fn(sendbuff, recvbuff, sendcount, dt, comm, stream);
   sc.finish();
   return rc;
@@ -589,8 +758,11 @@ ncclResult_t ncclReduceScatter(const void* sendbuff, void* recvbuff,
   using fn_t = ncclResult_t (*)(const void*, void*, size_t, ncclDataType_t,
                                 ncclRedOp_t, ncclComm_t, hipStream_t);
   static fn_t fn = (fn_t)real("ncclReduceScatter");
-  Manager::inst().count_comm_call(comm, (double)recvcount);
-  hiptimer::Scoped sc(stream, CAT_COMM, (double)recvcount);
+  const int nr = Manager::inst().comm_nranks(comm);
+  const double bytes = (double)recvcount * nccl_dtype_bytes((int)dt) * (nr > 0 ? nr : 1);
+  Manager::inst().count_comm_call(comm, bytes);
+  hiptimer::Scoped sc(stream, CAT_COMM, bytes, "ncclReduceScatter",
+                      bytes * busbw_factor_ag_rs(nr));
   ncclResult_t rc = fn(sendbuff, recvbuff, recvcount, dt, op, comm, stream);
   sc.finish();
   return rc;
@@ -602,8 +774,9 @@ ncclResult_t ncclBroadcast(const void* sendbuff, void* recvbuff, size_t count,
   using fn_t = ncclResult_t (*)(const void*, void*, size_t, ncclDataType_t,
                                 int, ncclComm_t, hipStream_t);
   static fn_t fn = (fn_t)real("ncclBroadcast");
-  Manager::inst().count_comm_call(comm, (double)count);
-  hiptimer::Scoped sc(stream, CAT_COMM, (double)count);
+  const double bytes = (double)count * nccl_dtype_bytes((int)dt);
+  Manager::inst().count_comm_call(comm, bytes);
+  hiptimer::Scoped sc(stream, CAT_COMM, bytes, "ncclBroadcast", bytes);
   ncclResult_t rc = fn(sendbuff, recvbuff, count, dt, root, comm, stream);
   sc.finish();
   return rc;
@@ -615,8 +788,9 @@ ncclResult_t ncclSend(const void* sendbuff, size_t count, ncclDataType_t dt,
       ncclResult_t (*)(const void*, size_t, ncclDataType_t, int, ncclComm_t,
                        hipStream_t);
   static fn_t fn = (fn_t)real("ncclSend");
-  Manager::inst().count_comm_call(comm, (double)count);
-  hiptimer::Scoped sc(stream, CAT_COMM, (double)count);
+  const double bytes = (double)count * nccl_dtype_bytes((int)dt);
+  Manager::inst().count_comm_call(comm, bytes);
+  hiptimer::Scoped sc(stream, CAT_COMM, bytes, "ncclSend", bytes);
   ncclResult_t rc = fn(sendbuff, count, dt, peer, comm, stream);
   sc.finish();
   return rc;
@@ -627,8 +801,9 @@ ncclResult_t ncclRecv(void* recvbuff, size_t count, ncclDataType_t dt, int peer,
   using fn_t = ncclResult_t (*)(void*, size_t, ncclDataType_t, int, ncclComm_t,
                                 hipStream_t);
   static fn_t fn = (fn_t)real("ncclRecv");
-  Manager::inst().count_comm_call(comm, (double)count);
-  hiptimer::Scoped sc(stream, CAT_COMM, (double)count);
+  const double bytes = (double)count * nccl_dtype_bytes((int)dt);
+  Manager::inst().count_comm_call(comm, bytes);
+  hiptimer::Scoped sc(stream, CAT_COMM, bytes, "ncclRecv", bytes);
   ncclResult_t rc = fn(recvbuff, count, dt, peer, comm, stream);
   sc.finish();
   return rc;
