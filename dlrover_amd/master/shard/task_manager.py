@@ -87,7 +87,11 @@ class DatasetManager:
         for shard in self.splitter.create_shards():
             self._todo.append(Task(self._task_id, shard, self.splitter.epoch))
             self._task_id += 1
-        self.splitter.epoch += 1
+        # sub-epoch splitters (table variant) only finish an epoch after
+        # their LAST refill; streaming never advances epochs
+        done_hook = getattr(self.splitter, "epoch_complete_after_refill", None)
+        if done_hook is None or done_hook():
+            self.splitter.epoch += 1
 
     def get_task(self, node_id: int) -> Optional[Task]:
         with self._lock:
