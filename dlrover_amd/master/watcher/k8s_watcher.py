@@ -93,6 +93,56 @@ class K8sEventSource:  # pragma: no cover - needs a cluster
             yield event["type"], event["object"].to_dict()
 
 
+class ScalePlanWatcher:
+    """User-submitted ScalePlan CRs -> ResourcePlan for the job manager
+    (ref: K8sScalePlanWatcher, k8s_watcher.py:354 — only ADDED manual-scale
+    plans for THIS job are honored, each CR consumed once by uid).
+
+    The event source is pluggable: FakeEventSource in the sim harness, a
+    real custom-object watch stream on a cluster.
+    """
+
+    def __init__(self, job_name: str, namespace: str = "default", source=None):
+        self.job_name = job_name
+        self.namespace = namespace
+        self._source = source or FakeEventSource()
+        self._used_uids: set = set()
+
+    def watch(self):
+        from dlrover_amd.master.auto_scale import ResourcePlan
+
+        for event_type, crd in self._source.stream():
+            if event_type != "ADDED" or not crd:
+                continue
+            if crd.get("kind") != "ScalePlan":
+                continue
+            meta = crd.get("metadata", {})
+            labels = meta.get("labels", {})
+            if labels.get("elasticjob.dlrover/name", labels.get("job")) not in (
+                None,
+                self.job_name,
+            ):
+                continue
+            if labels.get("scale-type", "manual") != "manual":
+                continue
+            uid = meta.get("uid") or meta.get("name")
+            if uid in self._used_uids:
+                continue
+            self._used_uids.add(uid)
+            spec = crd.get("spec", {})
+            specs = spec.get("replicaResourceSpecs", {})
+            worker = specs.get("worker", specs.get("Worker", {}))
+            plan = ResourcePlan(
+                node_count=int(worker.get("replicas", 0)),
+                node_resource={
+                    k: float(str(v).rstrip("mMiGg"))
+                    for k, v in worker.get("resource", {}).items()
+                },
+                comment=f"scaleplan/{meta.get('name', uid)}",
+            )
+            yield plan
+
+
 class PodWatcher:
     """Yields NodeEvents to DistributedJobManager._watch_events."""
 
