@@ -121,3 +121,57 @@ def test_events_exporter(tmp_path, monkeypatch):
     rows = [json.loads(l) for l in open(tmp_path / "ev.jsonl")]
     assert [r["type"] for r in rows] == ["instant", "begin", "end"]
     assert rows[2]["duration_s"] > 0
+
+
+def test_scaleplan_watcher_converts_manual_plans():
+    """User-submitted ScalePlan CR -> ResourcePlan (ref k8s_watcher.py:354);
+    only ADDED manual plans for this job, each uid once."""
+    from dlrover_amd.master.watcher.k8s_watcher import (
+        FakeEventSource,
+        ScalePlanWatcher,
+    )
+
+    src = FakeEventSource()
+    crd = {
+        "kind": "ScalePlan",
+        "metadata": {
+            "name": "sp1",
+            "uid": "u1",
+            "labels": {"elasticjob.dlrover/name": "jobA",
+                       "scale-type": "manual"},
+        },
+        "spec": {"replicaResourceSpecs": {
+            "worker": {"replicas": 6, "resource": {"cpu": "8", "memory": "1024"}}
+        }},
+    }
+    src.push("ADDED", crd)
+    src.push("ADDED", crd)  # duplicate uid -> ignored
+    src.push("MODIFIED", dict(crd, metadata={"name": "sp1", "uid": "u2",
+                                             "labels": crd["metadata"]["labels"]}))
+    other = {
+        "kind": "ScalePlan",
+        "metadata": {"name": "spB", "uid": "u3",
+                     "labels": {"elasticjob.dlrover/name": "jobB"}},
+        "spec": {"replicaResourceSpecs": {"worker": {"replicas": 2}}},
+    }
+    src.push("ADDED", other)  # different job -> ignored
+    w = ScalePlanWatcher("jobA", source=src)
+    plans = list(w.watch())
+    assert len(plans) == 1
+    assert plans[0].node_count == 6
+    assert plans[0].node_resource["cpu"] == 8.0
+    assert plans[0].comment == "scaleplan/sp1"
+
+
+def test_crd_manifests_parse():
+    import os
+
+    import yaml
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for f in ("elasticjob-crd.yaml", "scaleplan-crd.yaml",
+              "example-elasticjob.yaml"):
+        docs = list(yaml.safe_load_all(open(os.path.join(root, "deploy/k8s", f))))
+        assert docs and docs[0].get("kind") in (
+            "CustomResourceDefinition", "ElasticJob",
+        ), f
