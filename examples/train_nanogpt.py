@@ -83,6 +83,11 @@ def main():
             import time as _t
 
             _t.sleep(3600)
+        if (os.getenv("DLROVER_TEST_TIMELINE_FLAG") == "1"
+                and step == max(2, args.steps * 3 // 5) and rank == 0):
+            # ask the preloaded hiptimer to dump its kernel-trace ring
+            mdir = os.getenv("HIPTIMER_METRICS_DIR", "/tmp/hiptimer")
+            open(os.path.join(mdir, f"dump_timeline_{rank}"), "w").close()
         loss = model(ids, ids.clone())
         opt.zero_grad()
         loss.backward()

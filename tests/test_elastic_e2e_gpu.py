@@ -77,6 +77,45 @@ def test_nanogpt_hiptimer_metrics_gpu(tmp_path):
             comm_keys = [k for k in m if k.startswith("hiptimer_comm_calls")]
             if (m.get("hiptimer_launched_total", 0) > 100
                     and m.get("XPU_TIMER_COMMON_HANG") == 0 and comm_keys):
+                # round-2 depth: per-GEMM TFLOPS attribution and honest
+                # collective busbw must be exported
+                assert "hiptimer_gemm_tflops" in m, sorted(m)[:40]
+                assert "hiptimer_comm_busbw_gbs" in m, sorted(m)[:40]
+                gemm_shapes = [k for k in m if k.startswith(
+                    "hiptimer_kernel_count") and "gemm_m" in k]
+                assert gemm_shapes, sorted(m)[-40:]
                 return
         time.sleep(1.0)
     raise AssertionError(f"no satisfying hiptimer metrics; last parsed: {last}")
+
+
+@pytest.mark.timeout(540)
+def test_hiptimer_timeline_dump_gpu(tmp_path):
+    """Flag-file-triggered kernel-trace ring dump loads as chrome trace
+    (perfetto-compatible; ref manager.h:50-62 + gen_trace_timeline)."""
+    import glob
+    import json
+    import os
+    import time
+
+    proc, progress, ckpt_dir = _run_cli(
+        tmp_path, steps=25, ckpt_interval=30, nproc=1,
+        extra_env={
+            "DLROVER_HIPTIMER": "1",
+            "HIPTIMER_DUMP_INTERVAL": "1",
+            "DLROVER_TEST_TIMELINE_FLAG": "1",
+        },
+    )
+    assert proc.returncode == 0, proc.stderr[-4000:]
+    # the worker (train_nanogpt) touches the dump flag mid-run when
+    # DLROVER_TEST_TIMELINE_FLAG is set; find the dumped timeline
+    deadline = time.time() + 15
+    while time.time() < deadline:
+        for path in glob.glob("/tmp/hiptimer_*/timeline_*.json"):
+            data = json.load(open(path))
+            evs = data.get("traceEvents", [])
+            if len(evs) > 50:
+                assert {"name", "ph", "ts", "dur"} <= set(evs[0])
+                return
+        time.sleep(1.0)
+    raise AssertionError("no timeline dump found")
