@@ -553,6 +553,13 @@ def launch_agent(
                 )
             )
             collector = xpu_timer.HiptimerCollector(metrics_dir, client).start()
+            # HTTP scrape endpoint (ref daemon :18889); port 0 = pick free,
+            # DLROVER_PROM_PORT pins it for real Prometheus scrape configs
+            try:
+                prom_port = int(os.getenv("DLROVER_PROM_PORT", "0"))
+                xpu_timer.PrometheusExporter(metrics_dir, port=prom_port).start()
+            except OSError as e:
+                logger.warning("prometheus endpoint unavailable: %s", e)
 
     if config.network_check:
         from dlrover_amd.agent.node_check_agent import run_network_check

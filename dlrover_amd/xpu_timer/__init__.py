@@ -133,3 +133,68 @@ class HiptimerCollector:
                             )
             except Exception:  # noqa: BLE001
                 logger.exception("hiptimer collector iteration failed")
+
+
+class PrometheusExporter:
+    """HTTP scrape endpoint serving the node's aggregated hiptimer metrics
+    (ref: xpu_timer daemon LocalPrometheusService on :18889,
+    server/server.cc:33 — ours is an in-agent stdlib HTTP server over the
+    per-rank .prom files, same exposition format so any Prometheus scraper
+    or the master's SimpleMetricMonitor consumes it unchanged)."""
+
+    def __init__(self, metrics_dir: str, port: int = 18889, host: str = "0.0.0.0"):
+        from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+
+        metrics_dir_ = metrics_dir
+
+        class _H(BaseHTTPRequestHandler):
+            protocol_version = "HTTP/1.1"
+
+            def log_message(self, fmt, *args):  # noqa: N802
+                pass
+
+            def do_GET(self):  # noqa: N802
+                if self.path not in ("/metrics", "/"):
+                    self.send_response(404)
+                    self.send_header("Content-Length", "0")
+                    self.end_headers()
+                    return
+                lines = []
+                for path in sorted(
+                    glob.glob(os.path.join(metrics_dir_, "hiptimer_*.prom"))
+                ):
+                    try:
+                        rank = os.path.basename(path).split("_")[1].split(".")[0]
+                        for raw in open(path):
+                            raw = raw.strip()
+                            if not raw:
+                                continue
+                            key, _, val = raw.rpartition(" ")
+                            if "{" in key:
+                                key = key.replace("{", '{rank="%s",' % rank, 1)
+                            else:
+                                key = '%s{rank="%s"}' % (key, rank)
+                            lines.append(f"{key} {val}")
+                    except OSError:
+                        continue
+                body = ("\n".join(lines) + "\n").encode()
+                self.send_response(200)
+                self.send_header("Content-Type", "text/plain; version=0.0.4")
+                self.send_header("Content-Length", str(len(body)))
+                self.end_headers()
+                self.wfile.write(body)
+
+        self._server = ThreadingHTTPServer((host, port), _H)
+        self.port = self._server.server_address[1]
+        self._thread = threading.Thread(
+            target=self._server.serve_forever, name="hiptimer-prom", daemon=True
+        )
+
+    def start(self) -> "PrometheusExporter":
+        self._thread.start()
+        logger.info("hiptimer Prometheus endpoint on :%s/metrics", self.port)
+        return self
+
+    def stop(self):
+        self._server.shutdown()
+        self._server.server_close()

@@ -102,3 +102,29 @@ def test_rccl_env_bandwidth_expectations(monkeypatch):
     monkeypatch.setenv("NCCL_MIN_NCHANNELS", "4")
     env = rccl_env.effective_env()
     assert env["NCCL_MIN_NCHANNELS"] == "4"
+
+
+def test_prometheus_exporter_serves_aggregated_metrics(tmp_path):
+    """HTTP scrape endpoint (ref daemon :18889): per-rank .prom files are
+    merged with a rank label in standard exposition format."""
+    import urllib.request
+
+    from dlrover_amd import xpu_timer
+
+    (tmp_path / "hiptimer_0.prom").write_text(
+        "XPU_TIMER_COMMON_HANG 0\nhiptimer_gemm_tflops 123.45\n"
+        'hiptimer_op_count{cat="gemm"} 7\n'
+    )
+    (tmp_path / "hiptimer_1.prom").write_text("XPU_TIMER_COMMON_HANG 1\n")
+    exp = xpu_timer.PrometheusExporter(str(tmp_path), port=0,
+                                       host="127.0.0.1").start()
+    try:
+        body = urllib.request.urlopen(
+            f"http://127.0.0.1:{exp.port}/metrics", timeout=10
+        ).read().decode()
+        assert 'XPU_TIMER_COMMON_HANG{rank="0"} 0' in body
+        assert 'XPU_TIMER_COMMON_HANG{rank="1"} 1' in body
+        assert 'hiptimer_gemm_tflops{rank="0"} 123.45' in body
+        assert 'hiptimer_op_count{rank="0",cat="gemm"} 7' in body
+    finally:
+        exp.stop()
