@@ -371,23 +371,57 @@ class ShardedCheckpointEngine(CheckpointEngine):
 
         return itertools.chain(model.named_parameters(), model.named_buffers())
 
+    @staticmethod
+    def _derived_param_names(model, optimizer) -> dict:
+        """bf16 params whose exact value is RE-DERIVABLE from the optimizer's
+        fp32 master copy (FusedAdamW writes param = bf16(master) every step):
+        storing them in the snapshot is 16 GB of pure redundancy on the 8B
+        model — ~15% of checkpoint size, drain time and restore seconds.
+        Returns {param_name: flat_param_index}."""
+        import os
+
+        if optimizer is None or os.getenv("DLROVER_CKPT_DERIVED", "1") == "0":
+            return {}
+        state = getattr(optimizer, "state", None)
+        if not isinstance(state, dict):
+            return {}
+        id2name = {id(p): n for n, p in model.named_parameters()}
+        out = {}
+        idx = 0
+        for group in getattr(optimizer, "param_groups", []):
+            for p in group["params"]:
+                st = state.get(p)
+                if (
+                    st
+                    and "master_param" in st
+                    and p.dtype == torch.bfloat16
+                    and id(p) in id2name
+                ):
+                    out[id2name[id(p)]] = idx
+                idx += 1
+        return out
+
     def gather_state_dict(self, model, optimizer):
         try:
             from torch.distributed.tensor import DTensor
         except ImportError:  # pragma: no cover
             DTensor = ()
+        derived = self._derived_param_names(model, optimizer)
         model_sd = {}
         tags = {}
         for name, t in self._named_tensors(model):
             tags[name] = (
                 "shard0" if DTensor and isinstance(t, DTensor) else "replicated"
             )
+            if name in derived:
+                continue  # re-derived from the fp32 master on restore
             model_sd[name] = _to_local(t.detach())
         opt_sd = _localize(optimizer.state_dict()) if optimizer is not None else {}
         return {
             "model": model_sd,
             "optimizer": opt_sd,
             "_sharding": tags,
+            "_derived": derived,
             "world_size": _world_size(),
         }
 
@@ -406,6 +440,29 @@ class ShardedCheckpointEngine(CheckpointEngine):
             torch.cuda.synchronize()
         if optimizer is not None and state_dict.get("optimizer"):
             optimizer.load_state_dict(state_dict["optimizer"])
+        self._rederive_params(model, optimizer, state_dict)
+
+    @staticmethod
+    def _rederive_params(model, optimizer, state_dict):
+        """Params omitted from the snapshot (gather tagged them _derived)
+        are reconstructed on DEVICE from the restored fp32 master — an
+        HBM-local cast instead of 16 GB over PCIe."""
+        derived = state_dict.get("_derived") or {}
+        if not derived or optimizer is None:
+            return
+        params = [p for g in optimizer.param_groups for p in g["params"]]
+        name2p = dict(model.named_parameters())
+        with torch.no_grad():
+            for name, idx in derived.items():
+                if name not in name2p or idx >= len(params):
+                    continue
+                st = optimizer.state.get(params[idx])
+                if not st or "master_param" not in st:
+                    logger.warning("derived param %s has no master in the "
+                                   "restored optimizer state", name)
+                    continue
+                dst = _to_local(name2p[name].data)
+                dst.copy_(st["master_param"].to(dst.dtype))
 
     def restore_into(self, model, optimizer, path: str = ""):
         """Fast-path restore: zero-copy shm views -> async H2D into the live

@@ -160,6 +160,20 @@ def load_resharded(engine, model, optimizer, path: str) -> Optional[dict]:
             new_state["state"][idx] = sliced
         optimizer.load_state_dict(new_state)
 
+    # params omitted as _derived (bf16 twins of the fp32 master): rebuild
+    # them from the freshly resharded optimizer state
+    derived = shards[0].get("_derived") or {}
+    if derived and optimizer is not None:
+        params = [p for g in optimizer.param_groups for p in g["params"]]
+        name2p = dict(model.named_parameters())
+        with torch.no_grad():
+            for name, idx in derived.items():
+                if name in name2p and idx < len(params):
+                    st = optimizer.state.get(params[idx], {})
+                    if "master_param" in st:
+                        dst = _to_local(name2p[name].data)
+                        dst.copy_(st["master_param"].to(dst.dtype))
+
     meta = {k: v for k, v in shards[0].items()
             if not isinstance(v, dict) or k == "parallel"}
     meta["step"] = shards[0].get("step", 0)
