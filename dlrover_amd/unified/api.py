@@ -84,6 +84,12 @@ class _RoleBuilder:
     def role(self, name: str) -> "_RoleBuilder":
         return self._parent.role(name)
 
+    def with_collocation(self, *names: str) -> "DLJobBuilder":
+        return self._parent.with_collocation(*names)
+
+    def node_unit(self, n: int) -> "DLJobBuilder":
+        return self._parent.node_unit(n)
+
     def build(self) -> DLJob:
         return self._parent.build()
 

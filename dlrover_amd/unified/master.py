@@ -20,6 +20,7 @@ from typing import Dict, List, Optional
 
 from dlrover_amd.common.log import logger
 from dlrover_amd.unified.api import DLJob, WorkloadDesc
+from dlrover_amd.unified.scheduler import NodeSpec, Placement, Scheduler
 
 
 @dataclass
@@ -65,7 +66,8 @@ def _worker_entry(desc_env, role, rank, world, entry_func, entry_args):
 class PrimeMaster:
     """Controller driving the execution graph with per-role failover."""
 
-    def __init__(self, job: DLJob, state_path: str = ""):
+    def __init__(self, job: DLJob, state_path: str = "",
+                 nodes: Optional[List[NodeSpec]] = None):
         self.job = job
         self.graph = DLExecutionGraph(job)
         self.state_path = state_path or f"/tmp/dlrover_amd_prime_{job.name}.json"
@@ -74,11 +76,45 @@ class PrimeMaster:
         self._monitor: Optional[threading.Thread] = None
         self.status = "INIT"
         self.exit_code: Optional[int] = None
+        self.nodes = nodes or self._default_pool()
+        self.placement: Optional[Placement] = None
+        # self-recovery (ref: manager state save/self-recover :591-644):
+        # a master restarted mid-job resumes failover budgets from disk
+        prev = self.load_state(self.state_path)
+        if prev and prev.get("status") == "RUNNING":
+            budgets = {v["name"]: v.get("restarts", 0)
+                       for v in prev.get("vertices", [])}
+            for v in self.graph.vertices:
+                v.restarts = budgets.get(v.name, 0)
+            logger.info("prime master self-recovered state: %s", budgets)
+
+    def _default_pool(self) -> List[NodeSpec]:
+        # DLROVER_PRIME_NODES="node0:8,node1:8" (gpus per node); defaults to
+        # one node sized to hold the whole graph (local backend)
+        spec = os.getenv("DLROVER_PRIME_NODES", "")
+        if spec:
+            out = []
+            for part in spec.split(","):
+                name, _, g = part.partition(":")
+                out.append(NodeSpec(name, int(g or 8)))
+            return out
+        total_gpu = sum(
+            int(r.resource.get("gpu", 0)) * r.total
+            for r in self.job.roles.values()
+        )
+        total_cpu = sum(
+            float(r.resource.get("cpu", 1)) * r.total
+            for r in self.job.roles.values()
+        )
+        return [NodeSpec("local", max(total_gpu, 8), max(total_cpu, 64.0))]
 
     # -- lifecycle (ref: manager.prepare :141 / start :189) -------------------
 
     def prepare(self):
         self.status = "SCHEDULING"
+        # placement: collocation-aware bundles bin-packed onto the pool
+        # (ref: DLExecutionGraph.create + scheduler placement groups)
+        self.placement = Scheduler(self.nodes).schedule(self.graph)
         self._save_state()
         return self
 
@@ -97,9 +133,15 @@ class PrimeMaster:
         desc = self.job.roles[v.role]
         if desc.entry_func is None:
             raise ValueError(f"role {v.role}: local backend needs entry_func")
+        env = dict(desc.env)
+        env["DLROVER_PRIME_JOB"] = self.job.name
+        if self.placement is not None and v.name in self.placement.assignments:
+            node, dev = self.placement.assignments[v.name]
+            env["NODE_ID"] = node
+            env["DEVICE_INDEX"] = str(dev)
         p = self._ctx.Process(
             target=_worker_entry,
-            args=(desc.env, v.role, v.rank, v.world_size, desc.entry_func,
+            args=(env, v.role, v.rank, v.world_size, desc.entry_func,
                   desc.entry_args),
             name=v.name,
         )
