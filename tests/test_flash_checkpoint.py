@@ -148,3 +148,65 @@ def test_restore_from_disk_after_shm_gone(tmp_path, monkeypatch):
     finally:
         cp2.close()
         cp2.engine.shm_handler.unlink()
+
+
+def test_derived_param_omission_and_rederive(tmp_path):
+    """bf16 params with an fp32 master are OMITTED from the snapshot and
+    re-derived from the restored master (the GPU saves 16 GB on 8B; the
+    logic is dtype-driven so a bf16 CPU model exercises it end-to-end)."""
+    import torch
+
+    from dlrover_amd.models import LlamaConfig, LlamaForCausalLM
+    from dlrover_amd.ops import FusedAdamW
+    from dlrover_amd.trainer.flash_checkpoint.engine import (
+        ShardedCheckpointEngine,
+    )
+
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(LlamaConfig.tiny()).bfloat16()
+    opt = FusedAdamW(model.parameters(), lr=1e-3, weight_decay=0.0)
+    ids = torch.randint(0, 512, (2, 16))
+    model(ids, ids.clone()).backward()
+    opt.step()
+    opt.zero_grad()
+
+    eng = ShardedCheckpointEngine(str(tmp_path / "ckpt"))
+    sd = eng.gather_state_dict(model, opt)
+    n_params = sum(1 for _ in model.named_parameters())
+    assert len(sd["_derived"]) == n_params  # every param has a master
+    for name in sd["_derived"]:
+        assert name not in sd["model"]  # omitted from the payload
+    # buffers (rope tables) still present
+    assert any("rope" in k for k in sd["model"])
+
+    before = {n: p.detach().clone() for n, p in model.named_parameters()}
+    eng.save_to_memory(3, sd)
+    with torch.no_grad():
+        for p in model.parameters():
+            p.add_(1.0)
+    restored = eng.restore_into(model, opt)
+    assert restored is not None
+    for n, p in model.named_parameters():
+        assert torch.equal(p.detach(), before[n]), n
+    eng.shm_handler.unlink()
+
+
+def test_derived_disabled_by_env(tmp_path, monkeypatch):
+    import torch
+
+    from dlrover_amd.models import LlamaConfig, LlamaForCausalLM
+    from dlrover_amd.ops import FusedAdamW
+    from dlrover_amd.trainer.flash_checkpoint.engine import (
+        ShardedCheckpointEngine,
+    )
+
+    monkeypatch.setenv("DLROVER_CKPT_DERIVED", "0")
+    model = LlamaForCausalLM(LlamaConfig.tiny()).bfloat16()
+    opt = FusedAdamW(model.parameters(), lr=1e-3)
+    ids = torch.randint(0, 512, (1, 16))
+    model(ids, ids.clone()).backward()
+    opt.step()
+    eng = ShardedCheckpointEngine(str(tmp_path / "ckpt"))
+    sd = eng.gather_state_dict(model, opt)
+    assert sd["_derived"] == {}
+    assert all(n in sd["model"] for n, _ in model.named_parameters())
