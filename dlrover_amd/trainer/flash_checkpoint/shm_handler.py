@@ -435,18 +435,25 @@ class _DeviceStager:
         dst = torch.from_numpy(host_view)
         if not self._chunked:
             with torch.cuda.stream(self._stream):
+                ok = False
                 if pinned:
                     # direct hipMemcpyAsync into the REGISTERED shm mapping:
                     # torch's copy_ does not recognize foreign-pinned memory
                     # and degrades to a synchronous null-stream hipMemcpy
                     # that serializes with training compute (measured: each
                     # drain added its full ~2 s to the step wall)
-                    from dlrover_amd.ops.api import hip_ops
+                    try:
+                        from dlrover_amd.ops.api import hip_ops
 
-                    hip_ops().memcpy_d2h_async(
-                        host_view.ctypes.data, self._buf[: self._payload]
-                    )
-                else:
+                        hip_ops().memcpy_d2h_async(
+                            host_view.ctypes.data, self._buf[: self._payload]
+                        )
+                        ok = True
+                    except Exception:  # noqa: BLE001 — never lose the ckpt
+                        logger.exception(
+                            "async D2H fast path failed; torch fallback"
+                        )
+                if not ok:
                     dst[: self._payload].copy_(
                         self._buf[: self._payload], non_blocking=False
                     )
