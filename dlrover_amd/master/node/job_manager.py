@@ -237,11 +237,14 @@ class DistributedJobManager(JobManager):
     """Cluster mode: relaunch via the platform scaler and watch platform
     events (ref: dist_job_manager.py)."""
 
-    def __init__(self, scaler=None, watcher=None, **kw):
+    def __init__(self, scaler=None, watcher=None, min_nodes: int = 1, **kw):
         super().__init__(**kw)
         self.scaler = scaler
         self.watcher = watcher
+        self.min_nodes = min_nodes
         self._next_node_id = 1000  # relaunched nodes get fresh ids
+        self._pending_since: dict = {}  # node id -> first PENDING sighting
+        self._group_failures: dict = {}  # group id -> [fail timestamps]
 
     def start(self):
         super().start()
@@ -251,6 +254,64 @@ class DistributedJobManager(JobManager):
             )
             t.start()
             self._threads.append(t)
+        t = threading.Thread(
+            target=self._pending_monitor, name="pending-monitor", daemon=True
+        )
+        t.start()
+        self._threads.append(t)
+
+    # -- early stop on unschedulable nodes (ref: dist_job_manager.py:386
+    # -- _early_stop_part_of_pending) -----------------------------------------
+
+    def _pending_monitor(self):
+        """A job whose nodes sit PENDING past seconds_to_wait_pending can
+        never reach min_nodes: stop early and release what it holds rather
+        than occupying partial resources forever."""
+        wait = self._config.seconds_to_wait_pending
+        while not self._stop.wait(5.0):
+            now = time.time()
+            pending = [
+                n for n in self.ctx.job_nodes().values()
+                if n.status in (NodeStatus.PENDING, NodeStatus.INITIAL)
+                and not n.eliminated
+            ]
+            seen = {n.id for n in pending}
+            for nid in list(self._pending_since):
+                if nid not in seen:
+                    del self._pending_since[nid]
+            overdue = []
+            for n in pending:
+                first = self._pending_since.setdefault(n.id, now)
+                if now - first > wait:
+                    overdue.append(n)
+            if not overdue:
+                continue
+            running = [
+                n for n in self.ctx.alive_nodes()
+                if n.status == NodeStatus.RUNNING
+            ]
+            if len(running) >= self.min_nodes:
+                # enough schedulable capacity: give up on the stragglers
+                for n in overdue:
+                    logger.warning(
+                        "node %s pending > %.0fs: eliminating (job keeps "
+                        "%s running nodes)", n, wait, len(running),
+                    )
+                    n.eliminated = True
+                    n.relaunchable = False
+                    self.ctx.update_node(n)
+                    del self._pending_since[n.id]
+                continue
+            logger.error(
+                "%s nodes pending > %.0fs and only %s/%s running: "
+                "early-stopping the job (unschedulable)",
+                len(overdue), wait, len(running), self.min_nodes,
+            )
+            self.ctx.enqueue_action(
+                JobAbortAction(node_id=-1, reason="pending timeout")
+            )
+            self.ctx.request_stop(JobExitReason.PENDING_TIMEOUT, code=1)
+            return
 
     def _watch_events(self):
         while not self._stop.is_set():
@@ -274,9 +335,54 @@ class DistributedJobManager(JobManager):
                 )
             )
             return
+        if self._maybe_group_relaunch(node, reason):
+            return
+        self._relaunch_one(node, reason)
+
+    def _relaunch_one(self, node: Node, reason: str) -> Node:
         replacement = node.new_incarnation(self._next_node_id)
         self._next_node_id += 1
         self.ctx.update_node(replacement)
         logger.info("relaunching %s as %s (%s)", node, replacement, reason)
         self.scaler.launch_node(replacement)
         self.scaler.remove_node(node)
+        return replacement
+
+    # -- node-group relaunch (ref: dist_job_manager.py:1224) -------------------
+
+    GROUP_FAIL_WINDOW = 300.0  # s
+    GROUP_FAIL_THRESHOLD = 2
+
+    def _maybe_group_relaunch(self, node: Node, reason: str) -> bool:
+        """Multiple failures inside one node GROUP (super-pod / switch
+        domain) within a window indicate fabric-level trouble: relaunch the
+        WHOLE group together so the replacement set can be placed on a
+        healthy domain, instead of trickling single pods back into the bad
+        one."""
+        if node.group is None:
+            return False
+        now = time.time()
+        hist = self._group_failures.setdefault(node.group, [])
+        hist.append(now)
+        hist[:] = [t for t in hist if now - t < self.GROUP_FAIL_WINDOW]
+        if len(hist) < self.GROUP_FAIL_THRESHOLD:
+            return False
+        peers = [
+            n for n in self.ctx.job_nodes().values()
+            if n.group == node.group and not n.eliminated
+            and n.id != node.id
+            and n.status in (NodeStatus.RUNNING, NodeStatus.PENDING,
+                             NodeStatus.FAILED)
+        ]
+        logger.warning(
+            "group %s: %s failures in %.0fs — relaunching the whole group "
+            "(%s peers + the failed node)",
+            node.group, len(hist), self.GROUP_FAIL_WINDOW, len(peers),
+        )
+        self._group_failures[node.group] = []
+        self._relaunch_one(node, f"group relaunch: {reason}")
+        for peer in peers:
+            if peer.should_relaunch():
+                peer.inc_relaunch_count()
+                self._relaunch_one(peer, f"group relaunch ({node.group})")
+        return True
