@@ -20,6 +20,12 @@ from dlrover_amd.common.node import Node, NodeResource
 
 
 class K8sApi(ABC):
+    def create_service(self, svc_spec: dict) -> bool:  # pragma: no cover
+        return True
+
+    def delete_service(self, name: str) -> bool:  # pragma: no cover
+        return True
+
     @abstractmethod
     def create_pod(self, pod_spec: dict) -> bool:
         ...
@@ -61,6 +67,7 @@ class FakeK8sApi(K8sApi):
 
     def __init__(self):
         self.pods: Dict[str, dict] = {}
+        self.services: Dict[str, dict] = {}
         self.created: List[str] = []
         self.deleted: List[str] = []
         self._lock = threading.Lock()
@@ -78,9 +85,44 @@ class FakeK8sApi(K8sApi):
             self.deleted.append(name)
         return True
 
+    def create_service(self, svc_spec: dict) -> bool:
+        with self._lock:
+            self.services[svc_spec["metadata"]["name"]] = svc_spec
+        return True
+
+    def delete_service(self, name: str) -> bool:
+        with self._lock:
+            self.services.pop(name, None)
+        return True
+
     def list_pods(self, label_selector: str) -> List[dict]:
         with self._lock:
             return list(self.pods.values())
+
+
+def build_service_spec(job_name: str, node: Node) -> dict:
+    """Stable per-pod Service (ref: pod_scaler.py:776 service-per-pod):
+    gives every worker a DNS name that survives pod relaunch, so anything
+    addressing workers by name (sidecars, debuggers, the dashboard) does
+    not chase pod IPs."""
+    name = f"{job_name}-{node.type}-{node.rank_index}"
+    return {
+        "apiVersion": "v1",
+        "kind": "Service",
+        "metadata": {
+            "name": name,
+            "labels": {"elasticjob.dlrover/name": job_name},
+        },
+        "spec": {
+            "clusterIP": "None",  # headless
+            "selector": {
+                "elasticjob.dlrover/name": job_name,
+                "elasticjob.dlrover/replica-type": node.type,
+                "elasticjob.dlrover/rank-index": str(node.rank_index),
+            },
+            "ports": [{"name": "master", "port": 22222}],
+        },
+    }
 
 
 def build_pod_spec(
@@ -145,12 +187,14 @@ class PodScaler:
         api: Optional[K8sApi] = None,
         master_addr: str = "",
         image: str = "dlrover-amd:latest",
+        service_per_pod: bool = False,
     ):
         self.job_name = job_name
         self.namespace = namespace
         self.api = api if api is not None else RealK8sApi(namespace)
         self.master_addr = master_addr
         self.image = image
+        self.service_per_pod = service_per_pod
         self._create_q: "queue.Queue[Node]" = queue.Queue()
         self._stop = threading.Event()
         self._thread = threading.Thread(
@@ -199,6 +243,10 @@ class PodScaler:
             for attempt in range(3):
                 try:
                     self.api.create_pod(spec)
+                    if self.service_per_pod:
+                        self.api.create_service(
+                            build_service_spec(self.job_name, node)
+                        )
                     node.update_status(NodeStatus.PENDING)
                     break
                 except Exception:  # noqa: BLE001

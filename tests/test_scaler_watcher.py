@@ -175,3 +175,29 @@ def test_crd_manifests_parse():
         assert docs and docs[0].get("kind") in (
             "CustomResourceDefinition", "ElasticJob",
         ), f
+
+
+def test_service_per_pod():
+    """service-per-pod gives workers relaunch-stable DNS names (ref
+    pod_scaler.py:776)."""
+    import time
+
+    from dlrover_amd.common.node import Node, NodeResource
+    from dlrover_amd.common.constants import NodeType
+    from dlrover_amd.master.scaler.pod_scaler import FakeK8sApi, PodScaler
+
+    api = FakeK8sApi()
+    sc = PodScaler("jobx", api=api, service_per_pod=True)
+    try:
+        n = Node(NodeType.WORKER, 3, rank_index=3,
+                 config_resource=NodeResource(gpu_num=8))
+        sc.launch_node(n)
+        deadline = time.time() + 10
+        while time.time() < deadline and not api.services:
+            time.sleep(0.05)
+        assert "jobx-worker-3" in api.services
+        svc = api.services["jobx-worker-3"]
+        assert svc["spec"]["selector"]["elasticjob.dlrover/rank-index"] == "3"
+        assert svc["spec"]["clusterIP"] == "None"
+    finally:
+        sc.stop()
