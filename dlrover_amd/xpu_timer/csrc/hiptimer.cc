@@ -23,6 +23,11 @@
 // off the hot path (reference targets <=0.5%: xpu_timer/README.md:20).
 
 #include <dlfcn.h>
+#include <execinfo.h>
+#include <fcntl.h>
+#include <sys/stat.h>
+#include <errno.h>
+#include <signal.h>
 #include <pthread.h>
 #include <stdio.h>
 #include <stdlib.h>
@@ -87,6 +92,8 @@ static void* real(const char* name) {
   cache[name] = fn;
   return fn;
 }
+
+void install_crash_trace(const std::string& dir, int rank);
 
 enum Category : int {
   CAT_KERNEL = 0,
@@ -254,10 +261,11 @@ class Manager {
     metrics_dir_ = dir ? dir : "/tmp/hiptimer";
     const char* rank = getenv("RANK");
     rank_ = rank ? atoi(rank) : 0;
-    char cmd[256];
-    snprintf(cmd, sizeof(cmd), "mkdir -p %s", metrics_dir_.c_str());
-    if (system(cmd) != 0) enabled_ = false;
+    if (mkdir(metrics_dir_.c_str(), 0755) != 0 && errno != EEXIST)
+      enabled_ = false;
     last_completion_ = now();
+    if (enabled_ && getenv("HIPTIMER_NO_CRASH_TRACE") == nullptr)
+      install_crash_trace(metrics_dir_, rank_);
     if (enabled_ && !no_poller_)
       pthread_create(&poller_, nullptr, &Manager::poll_entry, this);
   }
@@ -479,6 +487,8 @@ class Manager {
     fclose(f);
   }
 
+ public:
+ private:
   bool enabled_ = false;
   bool no_events_ = false;
   bool no_poller_ = false;
@@ -514,6 +524,64 @@ class Manager {
   long trace_head_ = 0;             // poller thread only
   bool hang_dumped_ = false;
 };
+
+// ---- fatal-signal backtrace (ref: common/signal_handler.cc) --------------
+// On SIGSEGV/SIGABRT/SIGBUS/SIGFPE: write a native backtrace of the faulting
+// thread to <metrics_dir>/crash_<rank>.txt (async-signal-safe: backtrace_*_fd
+// + write), then re-raise with the default handler so the exit code and core
+// behavior are unchanged. Round-1's driver bench died with an unattributed
+// GPU memory fault — this leaves a host-side trace next time.
+class SignalBacktrace {
+ public:
+  static void install(const std::string& dir, int rank) {
+    static SignalBacktrace inst;
+    inst.path_ = dir + "/crash_" + std::to_string(rank) + ".txt";
+    int sigs[] = {SIGSEGV, SIGABRT, SIGBUS, SIGFPE};
+    for (int s : sigs) {
+      struct sigaction sa;
+      memset(&sa, 0, sizeof(sa));
+      sa.sa_sigaction = &SignalBacktrace::on_signal;
+      sa.sa_flags = SA_SIGINFO | SA_RESETHAND;
+      sigaction(s, &sa, &inst.prev_[s]);
+    }
+    self() = &inst;
+  }
+
+ private:
+  static SignalBacktrace*& self() {
+    static SignalBacktrace* p = nullptr;
+    return p;
+  }
+
+  static void on_signal(int sig, siginfo_t* info, void*) {
+    SignalBacktrace* s = self();
+    if (s != nullptr) {
+      int fd = open(s->path_.c_str(), O_CREAT | O_WRONLY | O_TRUNC, 0644);
+      if (fd >= 0) {
+        char head[128];
+        int n = snprintf(head, sizeof(head),
+                         "signal %d at addr %p; native backtrace:\n", sig,
+                         info ? info->si_addr : nullptr);
+        if (n > 0) {
+          ssize_t w = write(fd, head, (size_t)n);
+          (void)w;
+        }
+        void* frames[64];
+        int depth = backtrace(frames, 64);
+        backtrace_symbols_fd(frames, depth, fd);
+        close(fd);
+      }
+    }
+    raise(sig);  // SA_RESETHAND restored the default handler
+  }
+
+  std::string path_;
+  struct sigaction prev_[64] = {};
+};
+
+void install_crash_trace(const std::string& dir, int rank) {
+  SignalBacktrace::install(dir, rank);
+}
 
 struct Scoped {
   hipEvent_t start = nullptr, stop = nullptr;

@@ -128,3 +128,40 @@ def test_prometheus_exporter_serves_aggregated_metrics(tmp_path):
         assert 'hiptimer_op_count{rank="0",cat="gemm"} 7' in body
     finally:
         exp.stop()
+
+
+def test_crash_backtrace_dump(tmp_path):
+    """Fatal signals leave a native backtrace in the metrics dir before the
+    default handler runs (ref signal_handler.cc)."""
+    import os
+    import signal
+    import subprocess
+    import sys
+
+    from dlrover_amd import xpu_timer
+
+    lib = xpu_timer.lib_path() if hasattr(xpu_timer, "lib_path") else None
+    if lib is None:
+        import dlrover_amd
+
+        lib = os.path.join(os.path.dirname(dlrover_amd.__file__),
+                           "xpu_timer", "libhiptimer.so")
+    env = dict(os.environ)
+    env.update({
+        "LD_PRELOAD": lib,
+        "HIPTIMER_METRICS_DIR": str(tmp_path),
+        "HIPTIMER_NO_POLLER": "1",
+        "RANK": "0",
+    })
+    proc = subprocess.run(
+        [sys.executable, "-c",
+         "import ctypes; lib = ctypes.CDLL(None); lib.hipFree(None); "
+         "ctypes.string_at(0)"],  # hipFree = interposed (installs handler)
+        env=env, capture_output=True, timeout=120,
+    )
+    assert proc.returncode != 0
+    crash = tmp_path / "crash_0.txt"
+    assert crash.exists(), list(tmp_path.iterdir())
+    txt = crash.read_text()
+    assert "signal 11" in txt and "backtrace" in txt
+    assert proc.returncode in (-signal.SIGSEGV, 139, -11, 1), proc.returncode
