@@ -40,6 +40,18 @@ class ElasticTrainer:
             STEP_FILE_DIR, f"global_step_{os.getenv('ELASTIC_JOB_NAME', 'job')}.json"
         )
         os.makedirs(STEP_FILE_DIR, exist_ok=True)
+        # in-process GC/dataloader tracing feeding the hang/starvation
+        # metrics (ref xpu_timer py_tracing; enabled with the profiler)
+        self.py_tracer = None
+        if os.getenv("DLROVER_PY_TRACE", "") == "1" or (
+            os.getenv("DLROVER_HIPTIMER", "") == "1"
+            and os.getenv("HIPTIMER_METRICS_DIR")
+        ):
+            from dlrover_amd.diagnosis.py_runtime_tracer import PyRuntimeTracer
+
+            self.py_tracer = PyRuntimeTracer().start()
+            if self.dataloader is not None:
+                self.dataloader = self.py_tracer.wrap_loader(self.dataloader)
 
     def _accum_steps(self) -> int:
         world = max(_world_size(), 1)

@@ -160,9 +160,12 @@ class PrometheusExporter:
                     self.end_headers()
                     return
                 lines = []
-                for path in sorted(
+                paths = sorted(
                     glob.glob(os.path.join(metrics_dir_, "hiptimer_*.prom"))
-                ):
+                ) + sorted(
+                    glob.glob(os.path.join(metrics_dir_, "pymetrics_*.prom"))
+                )
+                for path in paths:
                     try:
                         rank = os.path.basename(path).split("_")[1].split(".")[0]
                         for raw in open(path):

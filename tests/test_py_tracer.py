@@ -76,3 +76,53 @@ def test_aggregate_groups_distinct_stacks():
     agg = py_tracer.aggregate_stacks({0: a, 1: b, 2: a})
     assert "ranks 0,2 (2 rank(s))" in agg
     assert "ranks 1 (1 rank(s))" in agg
+
+
+def test_py_runtime_tracer_gc_and_dataloader(tmp_path):
+    """In-process GC pause + dataloader wait tracing (ref py_tracing_manager
+    ring counters) exported as Prometheus text."""
+    import gc
+    import time
+
+    from dlrover_amd.diagnosis.py_runtime_tracer import PyRuntimeTracer
+
+    tr = PyRuntimeTracer(metrics_dir=str(tmp_path), interval=600)
+    tr.start()
+    try:
+        for _ in range(3):
+            gc.collect()
+
+        def slow_batches():
+            for i in range(4):
+                time.sleep(0.01)
+                yield i
+
+        wrapped = tr.wrap_loader(slow_batches())
+        assert list(wrapped) == [0, 1, 2, 3]
+        m = tr.metrics()
+        assert m['py_gc_collections{gen="2"}'] >= 3
+        assert m["py_dataloader_batches"] == 4
+        assert m["py_dataloader_wait_ms"] >= 30
+    finally:
+        tr.stop()
+    prom = (tmp_path / "pymetrics_0.prom").read_text()
+    assert "py_dataloader_batches 4" in prom
+
+
+def test_py_metrics_served_by_prometheus_exporter(tmp_path):
+    import urllib.request
+
+    from dlrover_amd import xpu_timer
+    from dlrover_amd.diagnosis.py_runtime_tracer import PyRuntimeTracer
+
+    tr = PyRuntimeTracer(metrics_dir=str(tmp_path), interval=600)
+    tr.dump()
+    exp = xpu_timer.PrometheusExporter(str(tmp_path), port=0,
+                                       host="127.0.0.1").start()
+    try:
+        body = urllib.request.urlopen(
+            f"http://127.0.0.1:{exp.port}/metrics", timeout=10
+        ).read().decode()
+        assert 'py_dataloader_batches{rank="0"}' in body
+    finally:
+        exp.stop()
