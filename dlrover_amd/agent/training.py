@@ -48,6 +48,7 @@ from dlrover_amd.common.constants import (
     TrainingExceptionLevel,
 )
 from dlrover_amd.common.log import logger
+from dlrover_amd.common.events import agent_events
 from dlrover_amd.common.multi_process import IPCServer
 from dlrover_amd.diagnosis.actions import DiagnosisActionType, action_from_wire
 
@@ -227,7 +228,8 @@ class ElasticTrainingAgent(LocalElasticAgent):
         saver = AsyncCheckpointSaver.get_ckpt_saver()
         if saver is not None:
             try:
-                saver.save_shm_to_storage()
+                with agent_events().duration("failure_ckpt_persist", {}):
+                    saver.save_shm_to_storage()
             except Exception:  # noqa: BLE001
                 logger.exception("failure-path checkpoint persist failed")
 
@@ -236,7 +238,10 @@ class ElasticTrainingAgent(LocalElasticAgent):
     def _invoke_run(self, role: str = "default") -> RunResult:
         spec = self._worker_group.spec
         self._start_heartbeats()
-        self._initialize_workers(self._worker_group)
+        with agent_events().duration(
+            "rendezvous", {"round": 0, "role": role}
+        ):
+            self._initialize_workers(self._worker_group)
         monitor_interval = spec.monitor_interval
         rdzv_handler = spec.rdzv_handler
 
@@ -300,7 +305,11 @@ class ElasticTrainingAgent(LocalElasticAgent):
                         "restarting workers (%s restarts left)",
                         self._remaining_restarts,
                     )
-                    self._restart_workers(self._worker_group)
+                    with agent_events().duration(
+                        "failure_restart",
+                        {"restarts_left": self._remaining_restarts},
+                    ):
+                        self._restart_workers(self._worker_group)
                     continue
                 self._stop_workers(self._worker_group)
                 self._worker_group.state = WorkerState.FAILED
@@ -357,11 +366,12 @@ class ElasticTrainingAgent(LocalElasticAgent):
            contexts would pin GPU memory into the new incarnation);
         4. re-rendezvous + start the new worker group.
         """
-        self._save_ckpt_to_storage()
-        descendants = self._worker_descendants()
-        self._stop_workers(self._worker_group)
-        self._sweep_orphans(descendants)
-        self._restart_workers(self._worker_group)
+        with agent_events().duration("membership_restart", {}):
+            self._save_ckpt_to_storage()
+            descendants = self._worker_descendants()
+            self._stop_workers(self._worker_group)
+            self._sweep_orphans(descendants)
+            self._restart_workers(self._worker_group)
 
     def _worker_descendants(self) -> List[int]:
         """Exact PIDs of worker processes and their live descendants,
