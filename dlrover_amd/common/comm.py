@@ -296,12 +296,14 @@ class DataLoaderConfig(Message):
     batch_size: int = 0
     num_workers: int = 0
     pin_memory: bool = True
+    version: int = 0  # trainers apply only configs newer than theirs
 
 
 @dataclass
 class OptimizerConfig(Message):
     optimizer_name: str = ""
     learning_rate: float = 0.0
+    version: int = 0
 
 
 @dataclass

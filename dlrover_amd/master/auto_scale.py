@@ -106,23 +106,76 @@ class JobAutoScaler:
 
 
 class SimpleStrategyGenerator:
-    """Dataloader/optimizer hyperparam suggestions from node stats
-    (ref: simple_strategy_generator.py — NOT TP/PP tuning)."""
+    """Dataloader/optimizer hyperparam tuning from node stats (ref:
+    simple_strategy_generator.py — versioned concrete suggestions, NOT
+    TP/PP tuning): grow the dataloader batch into free GPU memory bounded
+    by an activation-memory estimate, and rescale the learning rate by the
+    square-root batch rule. Trainers apply a config only when its version
+    is newer than the one they run."""
+
+    # headroom to keep free so a suggestion cannot OOM (ref keeps 2400 MB;
+    # sized up for 288 GB HBM3E parts)
+    MIN_FREE_MB = 8192
 
     def __init__(self, perf: PerfMonitor):
         self.perf = perf
+        self._versions: dict = {}
 
-    def generate_parallel_config(self, node_id: int = 0) -> comm.ParallelConfig:
-        cfg = comm.ParallelConfig()
+    @staticmethod
+    def _activation_mb(batch_size: int, model: dict) -> float:
+        """Per-step activation memory of a Llama-style block stack
+        (same estimator family as the reference: linear term per token +
+        quadratic attention term)."""
+        s = model.get("seq_len", 4096)
+        layers = model.get("n_layers", 32)
+        heads = model.get("n_heads", 32)
+        embd = model.get("hidden_size", 4096)
+        bytes_per = 2  # bf16 activations
+        lin = 34 * batch_size * s * embd * bytes_per
+        att = 5 * batch_size * s * s * heads * bytes_per / 1024
+        return (lin + att) * layers / (1 << 20)
+
+    def generate_parallel_config(
+        self,
+        node_id: int = 0,
+        current: Optional[comm.ParallelConfig] = None,
+        model: Optional[dict] = None,
+    ) -> comm.ParallelConfig:
+        cfg = current or comm.ParallelConfig()
         stats = self.perf.node_resource(node_id)
         if stats is None or not stats.gpu_stats:
             return cfg
-        # heuristic from the reference: if GPU memory is underused, suggest a
-        # larger dataloader batch (the trainer decides whether to apply it)
-        used = max((g.get("used_mb", 0) for g in stats.gpu_stats), default=0)
-        total = max((g.get("total_mb", 1) for g in stats.gpu_stats), default=1)
-        util = used / max(total, 1)
-        if util < 0.5:
-            cfg.dataloader.batch_size = 0  # 0 = "caller may double"
-            cfg.dataloader.num_workers = 4
-        return cfg
+        free_mb = min(
+            (g.get("total_mb", 0) - g.get("used_mb", 0))
+            for g in stats.gpu_stats
+        )
+        if free_mb <= self.MIN_FREE_MB:
+            return cfg
+        bs = cfg.dataloader.batch_size or 1
+        act = self._activation_mb(bs, model or {"seq_len": 4096})
+        if act <= 0:
+            return cfg
+        grown = int(bs + bs * (free_mb - self.MIN_FREE_MB) / act)
+        if grown <= bs:
+            return cfg
+        new_bs = min(grown, bs * 4)  # bounded growth per suggestion
+        lr = cfg.optimizer.learning_rate
+        out = comm.ParallelConfig(
+            dataloader=comm.DataLoaderConfig(
+                dataloader_name=cfg.dataloader.dataloader_name,
+                batch_size=new_bs,
+                num_workers=max(cfg.dataloader.num_workers, 4),
+                version=cfg.dataloader.version + 1,
+            ),
+            optimizer=comm.OptimizerConfig(
+                optimizer_name=cfg.optimizer.optimizer_name,
+                # square-root LR scaling with batch growth
+                learning_rate=(lr * (new_bs / bs) ** 0.5) if lr else lr,
+                version=cfg.optimizer.version + 1,
+            ),
+        )
+        logger.info(
+            "strategy: batch %s -> %s (free %.0f MB, act %.0f MB), lr x%.3f",
+            bs, new_bs, free_mb, act, (new_bs / bs) ** 0.5,
+        )
+        return out

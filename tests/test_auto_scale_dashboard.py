@@ -102,3 +102,44 @@ def test_dashboard_endpoints(ctx):
     finally:
         dash.stop()
         master.stop()
+
+
+def test_strategy_generator_versioned_growth():
+    """Concrete versioned batch/lr suggestions from free GPU memory bounded
+    by the activation-memory estimate (ref simple_strategy_generator)."""
+    from dlrover_amd.common import comm
+    from dlrover_amd.master.auto_scale import SimpleStrategyGenerator
+
+    class FakePerf:
+        def __init__(self, free_mb):
+            self._free = free_mb
+
+        def node_resource(self, node_id):
+            class S:
+                pass
+
+            s = S()
+            s.gpu_stats = [{"total_mb": 294912,
+                            "used_mb": 294912 - self._free}]
+            return s
+
+    cur = comm.ParallelConfig(
+        dataloader=comm.DataLoaderConfig(batch_size=2, version=3),
+        optimizer=comm.OptimizerConfig(learning_rate=1e-4, version=3),
+    )
+    gen = SimpleStrategyGenerator(FakePerf(free_mb=200000))
+    out = gen.generate_parallel_config(0, current=cur,
+                                       model={"seq_len": 4096,
+                                              "n_layers": 32,
+                                              "n_heads": 32,
+                                              "hidden_size": 4096})
+    assert out.dataloader.batch_size > 2
+    assert out.dataloader.batch_size <= 8  # bounded growth
+    assert out.dataloader.version == 4
+    assert out.optimizer.learning_rate > 1e-4  # sqrt scaling applied
+
+    # no headroom -> unchanged config, same version
+    gen2 = SimpleStrategyGenerator(FakePerf(free_mb=1000))
+    out2 = gen2.generate_parallel_config(0, current=cur)
+    assert out2.dataloader.batch_size == 2
+    assert out2.dataloader.version == 3
