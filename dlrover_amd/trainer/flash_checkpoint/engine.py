@@ -57,7 +57,15 @@ class CheckpointEvent:
 
 
 def _local_rank() -> int:
-    return int(os.getenv("LOCAL_RANK", "0"))
+    # LOCAL_RANK when a launcher set it; otherwise fall back to the GLOBAL
+    # rank — on a single host without a launcher every process would
+    # otherwise claim shm segment _0 and the savers would race (observed:
+    # two standalone savers persisting the same segment, one os.replace
+    # losing the other's tmp file)
+    lr = os.getenv("LOCAL_RANK")
+    if lr is not None:
+        return int(lr)
+    return int(os.getenv("RANK", "0"))
 
 
 def _global_rank() -> int:
