@@ -152,7 +152,9 @@ def main():
 
     maybe_bind_from_env()
     device = torch.device(f"cuda:{local_rank}" if on_gpu else "cpu")
-    torch.manual_seed(1234 + rank)
+    # model init must be IDENTICAL across ranks (FSDP shards the local
+    # weights as-is); per-rank seeds are set after build for the data
+    torch.manual_seed(1234)
 
     from dlrover_amd.ops import FusedAdamW
     from dlrover_amd.trainer.flash_checkpoint import FsdpShardCheckpointer, StorageType
@@ -179,6 +181,7 @@ def main():
 
         model = DDP(model)
     opt = FusedAdamW(model.parameters(), lr=args.lr, weight_decay=0.1)
+    torch.manual_seed(1234 + rank)  # per-rank DATA streams
 
     ckpt_dir = os.path.join(os.getcwd(), "gpurun_out", "bench_ckpt")
     cp = None
