@@ -266,10 +266,21 @@ def main():
         useful += dt
         step_est = dt if step_est is None else 0.7 * step_est + 0.3 * dt
         hb(f"step {k + 1}/{args.steps}")
-        remaining_cover = (args.steps - (k + 1)) * (step_est or 0.0)
-        if (cp is not None and (k + 1) % args.ckpt_interval == 0
-                and k + 1 < args.steps
-                and remaining_cover > drain_estimate):
+        save_due = (cp is not None and (k + 1) % args.ckpt_interval == 0
+                    and k + 1 < args.steps)
+        if save_due:
+            # drain-cover margin must be RANK-UNIFORM: save_to_memory runs
+            # readiness/step collectives, so per-rank timing noise deciding
+            # differently would deadlock the world (min over ranks = every
+            # rank can cover its drain)
+            margin = (args.steps - (k + 1)) * (step_est or 0.0) - drain_estimate
+            if world > 1:
+                t = torch.tensor([margin], dtype=torch.float64,
+                                 device=device if on_gpu else "cpu")
+                dist.all_reduce(t, op=dist.ReduceOp.MIN)
+                margin = t.item()
+            save_due = margin > 0
+        if save_due:
             # saves only when the async drain has future steps to hide
             # under — steady-state training always does; only a driver
             # window shorter than one drain doesn't (the drain would sit
