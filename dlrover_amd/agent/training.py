@@ -216,11 +216,29 @@ class ElasticTrainingAgent(LocalElasticAgent):
                         DiagnosisActionType.RELAUNCH_WORKER,
                     ):
                         self._restart_requested.set()
+                    elif action.action_type == DiagnosisActionType.DUMP_TIMELINE:
+                        self._trigger_timeline_dump()
                 except Exception:  # noqa: BLE001
                     logger.warning("heartbeat failed:\n%s", traceback.format_exc())
 
         self._hb_thread = threading.Thread(target=loop, daemon=True, name="agent-hb")
         self._hb_thread.start()
+
+    def _trigger_timeline_dump(self):
+        """Touch the per-rank hiptimer dump flags: each preloaded worker
+        writes its kernel-trace ring as perfetto-loadable JSON (ref:
+        xpu_timer_dump_timeline fan-out, dump_timeline.py:70)."""
+        mdir = os.environ.get("HIPTIMER_METRICS_DIR", "")
+        if not mdir:
+            return
+        try:
+            # shared flag: every preloaded local rank dumps once per touch
+            # (mtime edge-triggered in hiptimer — global ranks unknown here)
+            with open(os.path.join(mdir, "dump_timeline_all"), "w") as f:
+                f.write(str(time.time()))
+            logger.info("timeline dump requested for all local ranks")
+        except OSError as e:
+            logger.warning("timeline dump trigger failed: %s", e)
 
     # -- failure-path checkpoint persist (ref: training.py:1533) --------------------
 

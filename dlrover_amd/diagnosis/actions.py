@@ -17,6 +17,9 @@ class DiagnosisActionType:
     RESTART_WORKER = "restart_worker"
     RELAUNCH_WORKER = "relaunch_worker"
     JOB_ABORT = "job_abort"
+    # ask every agent to dump its workers' hiptimer kernel-trace rings
+    # (ref: py_xpu_timer dump_timeline fan-out across hosts)
+    DUMP_TIMELINE = "dump_timeline"
 
 
 @dataclass

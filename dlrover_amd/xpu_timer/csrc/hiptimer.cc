@@ -451,13 +451,23 @@ class Manager {
   // when <metrics_dir>/dump_timeline_<rank> appears, or automatically on the
   // first hang detection (ref: KernelTraceManager ring + gen_trace_timeline)
   void maybe_dump_timeline() {
-    char flag[512];
+    char flag[512], all_flag[512];
     snprintf(flag, sizeof(flag), "%s/dump_timeline_%d", metrics_dir_.c_str(),
              rank_);
+    snprintf(all_flag, sizeof(all_flag), "%s/dump_timeline_all",
+             metrics_dir_.c_str());
     double since;
     bool hang_now = is_hang(&since);
     bool flagged = access(flag, F_OK) == 0;
-    if (!flagged && !(hang_now && !hang_dumped_)) return;
+    // shared fan-out flag (every local rank dumps once per touch): edge-
+    // triggered on mtime so no rank has to unlink it out from under peers
+    bool all_flagged = false;
+    struct stat st;
+    if (stat(all_flag, &st) == 0 && st.st_mtime != last_all_flag_mtime_) {
+      last_all_flag_mtime_ = st.st_mtime;
+      all_flagged = true;
+    }
+    if (!flagged && !all_flagged && !(hang_now && !hang_dumped_)) return;
     if (hang_now) hang_dumped_ = true;
     if (flagged) unlink(flag);
     dump_timeline();
@@ -523,6 +533,7 @@ class Manager {
   TraceEvent trace_ring_[kTraceRing] = {};
   long trace_head_ = 0;             // poller thread only
   bool hang_dumped_ = false;
+  long last_all_flag_mtime_ = 0;
 };
 
 // ---- fatal-signal backtrace (ref: common/signal_handler.cc) --------------
