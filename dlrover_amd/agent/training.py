@@ -74,6 +74,7 @@ class ElasticLaunchConfig:
     redirects: str = ""
     numa_affinity: bool = False
     training_port: int = 0
+    tee: str = "0"
 
     def auto_configure(self):
         """Fill nproc from visible GPUs (ref: auto_configure_params :345)."""
@@ -184,7 +185,14 @@ class ElasticTrainingAgent(LocalElasticAgent):
         logs_specs: Optional[DefaultLogsSpecs] = None,
         exit_barrier_timeout: float = 300,
     ):
-        logs_specs = logs_specs or DefaultLogsSpecs(log_dir=config.log_dir or None)
+        if logs_specs is None:
+            from torch.distributed.elastic.multiprocessing import Std
+
+            logs_specs = DefaultLogsSpecs(
+                log_dir=config.log_dir or None,
+                redirects=Std.from_str(str(config.redirects or "0")),
+                tee=Std.from_str(str(config.tee or "0")),
+            )
         super().__init__(
             spec,
             logs_specs=logs_specs,

@@ -75,6 +75,13 @@ def parse_args(argv: Optional[List[str]] = None):
                         "(kernel/GEMM/RCCL timing + hang detection)")
     p.add_argument("--checkpoint-dir", default="/tmp/dlrover_amd_ckpt")
     p.add_argument("--log-dir", default=None)
+    p.add_argument("--redirects", default="0",
+                   help="redirect worker std streams to files "
+                        "(0|1|2|3 = none|stdout|stderr|both, torchrun-style)")
+    p.add_argument("--tee", default="0",
+                   help="tee worker std streams to console AND files")
+    p.add_argument("--run-id", "--run_id", default="",
+                   help="rendezvous run id (defaults to the job name)")
     p.add_argument("--service-type", default=CommServiceType.TCP)
     p.add_argument("training_script", help="training program (.py or executable)")
     p.add_argument("training_script_args", nargs=argparse.REMAINDER)
@@ -196,6 +203,9 @@ def run(args) -> int:
         checkpoint_dir=args.checkpoint_dir,
         log_dir=args.log_dir,
         numa_affinity=args.numa_affinity,
+        redirects=args.redirects,
+        tee=args.tee,
+        run_id=args.run_id or os.getenv("ELASTIC_JOB_NAME", "dlrover"),
     )
 
     script = args.training_script
