@@ -244,6 +244,9 @@ class MasterClient:
     def report_global_step(self, step: int, timestamp: float = 0.0):
         self.report(comm.GlobalStep(step=step, timestamp=timestamp or time.time()))
 
+    def report_model_info(self, **kw):
+        return self.report(comm.ModelInfo(**kw))
+
     def report_diagnosis_data(self, data_cls: str, content: str, node_rank: int = -1):
         self.report(
             comm.DiagnosisReportData(

@@ -307,6 +307,21 @@ class OptimizerConfig(Message):
 
 
 @dataclass
+class ModelInfo(Message):
+    """Trainer-reported model card (ref: stats/training_metrics model info):
+    feeds the strategy generator's activation-memory estimate and the
+    dashboard."""
+
+    model_name: str = ""
+    params: int = 0
+    n_layers: int = 0
+    n_heads: int = 0
+    hidden_size: int = 0
+    seq_len: int = 0
+    dtype: str = ""
+
+
+@dataclass
 class ParallelConfigRequest(Message):
     pass
 

@@ -59,6 +59,11 @@ class _Handler(BaseHTTPRequestHandler):
             "nodes_waiting": rdzv.num_nodes_waiting(),
             "global_step": master.perf_monitor.completed_global_step,
             "steps_per_sec": round(master.perf_monitor.running_speed(), 4),
+            "model": (
+                master.perf_monitor.model_info.__dict__
+                if getattr(master.perf_monitor, "model_info", None)
+                else None
+            ),
         }
 
     @staticmethod

@@ -14,6 +14,8 @@ from dlrover_amd.common import comm
 
 
 class PerfMonitor:
+    model_info = None
+
     def __init__(self, window: int = 32):
         self._lock = threading.Lock()
         self._samples: Deque[Tuple[float, int]] = deque(maxlen=window)
@@ -26,6 +28,11 @@ class PerfMonitor:
             self._samples.append((timestamp, step))
             self._last_step = max(self._last_step, step)
             self._last_step_time = timestamp
+
+    def report_model_info(self, info: comm.ModelInfo):
+        """Trainer-reported model card (ref: stats/job_collector): used by
+        the strategy generator's activation estimate and the dashboard."""
+        self.model_info = info
 
     def report_resource(self, node_id: int, stats: comm.ResourceStats):
         with self._lock:

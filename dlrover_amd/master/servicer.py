@@ -150,6 +150,9 @@ class MasterServicer:
         if isinstance(msg, comm.DiagnosisReportData):
             self.master.diagnosis_manager.collect_data(msg)
             return None
+        if isinstance(msg, comm.ModelInfo):
+            self.master.perf_monitor.report_model_info(msg)
+            return None
         if isinstance(msg, comm.TrainingStatusRequest):
             return comm.TrainingStatusReply(status=self.master.job_manager.training_status())
         raise ValueError(f"unhandled report message {type(msg).__name__}")

@@ -52,6 +52,29 @@ class ElasticTrainer:
             self.py_tracer = PyRuntimeTracer().start()
             if self.dataloader is not None:
                 self.dataloader = self.py_tracer.wrap_loader(self.dataloader)
+        self._report_model_info()
+
+    def _report_model_info(self):
+        """Best-effort model card to the master (ref: stats/job_collector);
+        feeds the strategy generator and the dashboard."""
+        if not os.getenv("DLROVER_MASTER_ADDR"):
+            return
+        try:
+            from dlrover_amd.agent.master_client import MasterClient
+
+            mod = self.model.module if hasattr(self.model, "module") else self.model
+            cfg = getattr(mod, "cfg", None)
+            MasterClient.singleton_instance().report_model_info(
+                model_name=type(mod).__name__,
+                params=sum(p.numel() for p in mod.parameters()),
+                n_layers=getattr(cfg, "n_layers", 0),
+                n_heads=getattr(cfg, "n_heads", 0),
+                hidden_size=getattr(cfg, "hidden_size", 0),
+                seq_len=getattr(cfg, "max_seq_len", 0),
+                dtype=str(next(mod.parameters()).dtype).replace("torch.", ""),
+            )
+        except Exception:  # noqa: BLE001 — reporting must never break training
+            logger.debug("model info report skipped", exc_info=True)
 
     def _accum_steps(self) -> int:
         world = max(_world_size(), 1)
