@@ -66,7 +66,30 @@ class JobMaster:
     # -- servicer hooks ------------------------------------------------------------
 
     def paral_config(self) -> comm.ParallelConfig:
-        return comm.ParallelConfig()
+        """Versioned hyperparam suggestion from runtime stats (ref:
+        get_paral_config servicer:424 + SimpleStrategyGenerator): uses the
+        trainer-reported model card for the activation-memory bound."""
+        from dlrover_amd.master.auto_scale import SimpleStrategyGenerator
+
+        info = getattr(self.perf_monitor, "model_info", None)
+        model = None
+        if info is not None:
+            model = {
+                "seq_len": info.seq_len or 4096,
+                "n_layers": info.n_layers or 32,
+                "n_heads": info.n_heads or 32,
+                "hidden_size": info.hidden_size or 4096,
+            }
+        try:
+            cfg = SimpleStrategyGenerator(self.perf_monitor).generate_parallel_config(
+                0, current=self._last_paral_config, model=model
+            )
+            self._last_paral_config = cfg
+            return cfg
+        except Exception:  # noqa: BLE001 — suggestions must never fail a poll
+            return self._last_paral_config or comm.ParallelConfig()
+
+    _last_paral_config = None
 
     def elastic_run_config(self) -> Dict[str, str]:
         return dict(self._elastic_run_configs)
