@@ -355,6 +355,15 @@ class Manager {
     {
       std::lock_guard<std::mutex> g(q_mu_);
       outstanding = pending_.size();
+      if (!pending_.empty()) {
+        const PendingOp& f = pending_.front();
+        oldest_name_ = f.name ? f.name : kCatNames[f.cat];
+        oldest_cat_ = f.cat;
+        oldest_enqueue_ = f.enqueue_ts;
+      } else {
+        oldest_name_ = nullptr;
+      }
+      outstanding_ = outstanding;
     }
     double idle = now() - last_completion_;
     if (outstanding > 0 && idle > hang_secs_) {
@@ -378,6 +387,17 @@ class Manager {
     // collector/diagnostician logic carries over (XPU_TIMER_COMMON_HANG)
     fprintf(f, "XPU_TIMER_COMMON_HANG %d\n", hang);
     fprintf(f, "hiptimer_hang_since_seconds %.3f\n", since);
+    fprintf(f, "hiptimer_outstanding_ops %zu\n", outstanding_);
+    if (oldest_name_ != nullptr) {
+      // the op the device has been sitting on: a stuck ncclAllReduce names
+      // the wedged collective directly (ref hang dossier quality bar)
+      std::string nm = std::string(oldest_name_).substr(0, 120);
+      for (auto& c : nm)
+        if (c == '"' || c == '\\' || c == '\n') c = '_';
+      fprintf(f,
+              "hiptimer_oldest_pending{name=\"%s\",cat=\"%s\"} %.3f\n",
+              nm.c_str(), kCatNames[oldest_cat_], now() - oldest_enqueue_);
+    }
     fprintf(f, "hiptimer_wall_seconds %.3f\n", now());
     fprintf(f, "hiptimer_launched_total %ld\n", launched_.load());
     fprintf(f, "hiptimer_sample_interval %d\n", sample_);
@@ -534,6 +554,10 @@ class Manager {
   long trace_head_ = 0;             // poller thread only
   bool hang_dumped_ = false;
   long last_all_flag_mtime_ = 0;
+  size_t outstanding_ = 0;        // poller thread only
+  const char* oldest_name_ = nullptr;
+  int oldest_cat_ = 0;
+  double oldest_enqueue_ = 0;
 };
 
 // ---- fatal-signal backtrace (ref: common/signal_handler.cc) --------------
