@@ -89,3 +89,17 @@ def test_replica_backup_and_gather(tmp_path):
             p.join(timeout=240)
         outcomes = dict(results)
     assert all(outcomes.get(r) == "ok" for r in range(WS)), outcomes
+
+
+def test_backup_peer_configurable_groups():
+    from dlrover_amd.trainer.flash_checkpoint.replica import backup_peer
+
+    # pairs (default): ring within groups of 2
+    assert [backup_peer(r, 4, 2) for r in range(4)] == [1, 0, 3, 2]
+    # groups of 4: ring 0->1->2->3->0
+    assert [backup_peer(r, 8, 4) for r in range(8)] == [1, 2, 3, 0, 5, 6, 7, 4]
+    # world 6, size 4: tail group {4,5} still rings
+    assert backup_peer(4, 6, 4) == 5 and backup_peer(5, 6, 4) == 4
+    # odd world, pairs: rank 2 folds into the previous group {0,1,2}
+    peers = [backup_peer(r, 3, 2) for r in range(3)]
+    assert peers[2] != 2 and all(0 <= p < 3 for p in peers)
