@@ -120,6 +120,8 @@ class JobManager:
             self._transition(node, NodeStatus.SUCCEEDED)
             self._maybe_finish_job()
         elif event.event_type == NodeEventType.FAILED_EXITED:
+            if event.reason == NodeExitReason.OOM:
+                node.exit_reason = NodeExitReason.OOM
             node.exit_reason = node.exit_reason or NodeExitReason.UNKNOWN_ERROR
             self._transition(node, NodeStatus.FAILED)
             self._handle_node_failure(node, event.reason)
@@ -134,6 +136,8 @@ class JobManager:
             if meta.status and allowed_transition(node.status, meta.status):
                 node.update_status(meta.status)
                 if meta.status == NodeStatus.FAILED:
+                    if event.reason == NodeExitReason.OOM:
+                        node.exit_reason = NodeExitReason.OOM
                     self._handle_node_failure(node, event.reason)
 
     def _transition(self, node: Node, status: str):
@@ -339,8 +343,26 @@ class DistributedJobManager(JobManager):
             return
         self._relaunch_one(node, reason)
 
+    OOM_MEMORY_FACTOR = 2.0  # ref: PSTrainingAutoScaler OOM recovery cushion
+    OOM_MEMORY_CAP_MB = 1 << 20  # 1 TiB
+
     def _relaunch_one(self, node: Node, reason: str) -> Node:
         replacement = node.new_incarnation(self._next_node_id)
+        if node.exit_reason == NodeExitReason.OOM and (
+            replacement.config_resource.memory_mb > 0
+        ):
+            grown = int(
+                min(
+                    replacement.config_resource.memory_mb
+                    * self.OOM_MEMORY_FACTOR,
+                    self.OOM_MEMORY_CAP_MB,
+                )
+            )
+            logger.info(
+                "node %s OOMKilled: relaunching with memory %s -> %s MB",
+                node.id, replacement.config_resource.memory_mb, grown,
+            )
+            replacement.config_resource.memory_mb = grown
         self._next_node_id += 1
         self.ctx.update_node(replacement)
         logger.info("relaunching %s as %s (%s)", node, replacement, reason)

@@ -218,3 +218,40 @@ def test_group_relaunch_after_repeated_failures():
     # second failure within the window: the WHOLE group goes
     mgr._relaunch_node(nodes[1], "crash")
     assert len(sc.launched) >= 3, sc.launched  # node1 + surviving peers
+
+
+def test_oom_relaunch_grows_memory():
+    """OOMKilled nodes relaunch with a bigger memory request (ref
+    JobAutoScaler OOM recovery)."""
+    from dlrover_amd.common import comm
+    from dlrover_amd.common.constants import (
+        NodeEventType,
+        NodeExitReason,
+        NodeStatus,
+        NodeType,
+    )
+    from dlrover_amd.common.node import Node, NodeResource
+    from dlrover_amd.master.node.job_manager import DistributedJobManager
+
+    class FakeScaler:
+        launched = None
+
+        def launch_node(self, n):
+            FakeScaler.launched = n
+
+        def remove_node(self, n):
+            pass
+
+    ctx = _fresh_ctx()
+    mgr = DistributedJobManager(job_context=ctx, scaler=FakeScaler())
+    n = Node(NodeType.WORKER, 0, rank_index=0, max_relaunch_count=3,
+             config_resource=NodeResource(memory_mb=65536))
+    n.update_status(NodeStatus.RUNNING)
+    ctx.update_node(n)
+    mgr.on_node_event(comm.NodeEvent(
+        event_type=NodeEventType.FAILED_EXITED,
+        node=comm.NodeMeta(type=NodeType.WORKER, id=0, rank=0),
+        reason=NodeExitReason.OOM,
+    ))
+    assert FakeScaler.launched is not None
+    assert FakeScaler.launched.config_resource.memory_mb == 131072
