@@ -40,6 +40,8 @@ class _Handler(BaseHTTPRequestHandler):
                 self._send(200, json.dumps(self._metrics(master)).encode())
             elif self.path.startswith("/api/logs"):
                 self._send(200, json.dumps(self._logs()).encode())
+            elif self.path.startswith("/api/events"):
+                self._send(200, json.dumps(self._events()).encode())
             else:
                 self._send(404, b'{"error": "not found"}')
         except Exception as e:  # noqa: BLE001
@@ -105,6 +107,28 @@ class _Handler(BaseHTTPRequestHandler):
         return out
 
     @staticmethod
+    def _events(n: int = 200) -> list:
+        """Tail of the training-event JSONL spans (goodput/postmortem view;
+        ref: training_event exporters + dashboard)."""
+        import glob
+        import os
+
+        base = os.getenv("DLROVER_EVENT_DIR", "/tmp/dlrover_amd_events")
+        out = []
+        for path in sorted(glob.glob(os.path.join(base, "events_*.jsonl")))[-8:]:
+            try:
+                with open(path, errors="replace") as f:
+                    for line in f.readlines()[-n:]:
+                        try:
+                            out.append(json.loads(line))
+                        except ValueError:
+                            continue
+            except OSError:
+                continue
+        out.sort(key=lambda r: r.get("ts", 0))
+        return out[-n:]
+
+    @staticmethod
     def _metrics(master) -> dict:
         hangs = master.diagnosis_manager.data.latest_by_node("hang")
         return {
@@ -129,7 +153,8 @@ steps/s: {job['steps_per_sec']}</p>
 <tr><th>id</th><th>rank</th><th>status</th><th>relaunches</th><th>hb age s</th></tr>
 {rows}</table>
 <p>APIs: <a href=/api/job>/api/job</a> <a href=/api/nodes>/api/nodes</a>
-<a href=/api/metrics>/api/metrics</a></p></body></html>"""
+<a href=/api/metrics>/api/metrics</a> <a href=/api/events>/api/events</a>
+<a href=/api/logs>/api/logs</a></p></body></html>"""
 
 
 class Dashboard:

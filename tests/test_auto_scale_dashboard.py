@@ -143,3 +143,28 @@ def test_strategy_generator_versioned_growth():
     out2 = gen2.generate_parallel_config(0, current=cur)
     assert out2.dataloader.batch_size == 2
     assert out2.dataloader.version == 3
+
+
+def test_dashboard_events_tail(tmp_path, monkeypatch):
+    import json
+    import urllib.request
+
+    monkeypatch.setenv("DLROVER_EVENT_DIR", str(tmp_path))
+    (tmp_path / "events_agent_1.jsonl").write_text(
+        '{"ts": 1.0, "name": "rendezvous", "phase": "begin"}\n'
+        '{"ts": 2.0, "name": "rendezvous", "phase": "end"}\n'
+    )
+    from dlrover_amd.master.dashboard import Dashboard
+
+    class M:
+        pass
+
+    dash = Dashboard(M(), port=0, host="127.0.0.1").start()
+    try:
+        body = urllib.request.urlopen(
+            f"http://127.0.0.1:{dash.port}/api/events", timeout=10
+        ).read()
+        evs = json.loads(body)
+        assert len(evs) == 2 and evs[-1]["phase"] == "end"
+    finally:
+        dash.stop()
