@@ -297,3 +297,88 @@ class NetworkCheckRendezvousManager(RendezvousManager):
             self._straggler_nodes.clear()
             self._rdzv_nodes = {}
             self._waiting_nodes = {}
+
+
+class GroupNetworkCheckRendezvousManager(NetworkCheckRendezvousManager):
+    """Node-group (super-pod) aware probe pairing (ref:
+    GroupNodeNetworkCheckRendezvousManager, rdzv_manager.py:876-1070):
+    even rounds pair WITHIN each group (intra-group links: xGMI/ASW
+    domain), odd rounds pair ACROSS neighboring groups (inter-group
+    fabric). A node whose intra round passed but whose cross round failed
+    implicates the GROUP-PAIR link, not the node — exposed via
+    suspect_group_links()."""
+
+    def __init__(self):
+        super().__init__()
+        self._groups: Dict[int, int] = {}  # rank -> group id
+        self._suspect_links: set = set()   # {(group_a, group_b)}
+
+    def set_node_groups(self, mapping: Dict[int, int]):
+        with self._lock:
+            self._groups = dict(mapping)
+
+    def _build_groups_locked(self) -> List[List[int]]:
+        if not self._groups:
+            return super()._build_groups_locked()
+        ranks = sorted(self._rdzv_nodes)
+        by_group: Dict[int, List[int]] = {}
+        for r in ranks:
+            by_group.setdefault(self._groups.get(r, -1), []).append(r)
+        pairs: List[List[int]] = []
+        if self._check_round % 2 == 0:
+            # intra-group adjacent pairs; odd tails merge into the previous
+            # pair of the SAME group (never across groups)
+            for g in sorted(by_group):
+                members = by_group[g]
+                gp = [members[i : i + 2] for i in range(0, len(members), 2)]
+                if gp and len(gp[-1]) == 1 and len(gp) > 1:
+                    gp[-2].extend(gp.pop())
+                pairs.extend(gp)
+        else:
+            # cross-group: i-th member of group k probes i-th of group k+1
+            gids = sorted(by_group)
+            for gi in range(0, len(gids) - 1, 2):
+                a, b = by_group[gids[gi]], by_group[gids[gi + 1]]
+                for x, y in zip(a, b):
+                    pairs.append(sorted([x, y]))
+                tail = a[len(b):] + b[len(a):]
+                if len(tail) > 1:
+                    pairs.append(tail)
+                elif tail and pairs:
+                    pairs[-1] = pairs[-1] + tail
+            if len(gids) % 2 == 1:
+                members = by_group[gids[-1]]
+                gp = [members[i : i + 2] for i in range(0, len(members), 2)]
+                if gp and len(gp[-1]) == 1 and len(gp) > 1:
+                    gp[-2].extend(gp.pop())
+                pairs.extend(gp)
+        merged: List[List[int]] = []
+        for p in pairs:
+            if len(p) == 1 and merged:
+                merged[-1].extend(p)
+            else:
+                merged.append(p)
+        return merged
+
+    def _analyse_locked(self):
+        super()._analyse_locked()
+        if not self._groups or self._check_round % 2 == 0:
+            return
+        # cross round: failures implicate the inter-group path probed
+        for r, ok in self._node_status.items():
+            if ok:
+                continue
+            g = self._groups.get(r, -1)
+            for pair in self._build_groups_locked():
+                if r in pair:
+                    for peer in pair:
+                        pg = self._groups.get(peer, -1)
+                        if pg != g:
+                            self._suspect_links.add(tuple(sorted((g, pg))))
+        if self._suspect_links:
+            logger.warning("[network-check] suspect inter-group links: %s",
+                           self._suspect_links)
+
+    def suspect_group_links(self) -> List[tuple]:
+        with self._lock:
+            return sorted(self._suspect_links)

@@ -59,3 +59,36 @@ def test_scale_respects_node_unit(master):
     assert set(world) == {0, 1}
     for c in clients.values():
         c.close()
+
+
+def test_group_network_check_pairing_and_link_suspects():
+    """Node-group-aware check: intra-group pairs in even rounds, cross-group
+    pairs in odd rounds; a cross-round-only failure implicates the group
+    LINK (ref GroupNodeNetworkCheckRendezvousManager)."""
+    from dlrover_amd.master.elastic.rdzv_manager import (
+        GroupNetworkCheckRendezvousManager,
+    )
+
+    mgr = GroupNetworkCheckRendezvousManager()
+    # 2 groups x 4 nodes
+    mgr._rdzv_nodes = {r: 1 for r in range(8)}
+    mgr.set_node_groups({r: r // 4 for r in range(8)})
+
+    pairs0 = mgr._build_groups_locked()
+    for p in pairs0:
+        gids = {r // 4 for r in p}
+        assert len(gids) == 1, f"round-0 pair crosses groups: {p}"
+
+    # everyone passes round 0 quickly
+    for r in range(8):
+        mgr.report_network_check_result(r, True, 1.0)
+    assert mgr._check_round == 1
+
+    pairs1 = mgr._build_groups_locked()
+    crossing = [p for p in pairs1 if len({r // 4 for r in p}) == 2]
+    assert crossing, pairs1
+
+    # round 1: node 5 fails its cross-group probe
+    for r in range(8):
+        mgr.report_network_check_result(r, r != 5, 1.0)
+    assert (0, 1) in mgr.suspect_group_links()
