@@ -94,9 +94,23 @@ class JobMaster:
     def elastic_run_config(self) -> Dict[str, str]:
         return dict(self._elastic_run_configs)
 
+    _precheck_chain = None
+
     def pre_check_result(self, node_id: int) -> comm.PreCheckResponse:
-        # pre-check operator chain (ref: precheck_operator.py); default: pass
-        return comm.PreCheckResponse(status=comm.PreCheckResponse.PASS)
+        """Pluggable operator chain (ref: precheck_operator.py; operators
+        named in Context.pre_check_operators / DLROVER_PRE_CHECK_OPS)."""
+        import os
+
+        if self._precheck_chain is None:
+            from dlrover_amd.common.global_context import Context
+            from dlrover_amd.master.precheck import PreCheckChain
+
+            names = list(Context.singleton_instance().pre_check_operators)
+            env = os.getenv("DLROVER_PRE_CHECK_OPS", "")
+            names += [n.strip() for n in env.split(",") if n.strip()]
+            self._precheck_chain = PreCheckChain(names)
+        status, reason = self._precheck_chain.evaluate(self)
+        return comm.PreCheckResponse(status=status, reason=reason)
 
     def ckpt_sync(self, node_id: int, step: int) -> bool:
         """Consensus that all nodes persisted `step` (ref:

@@ -98,3 +98,60 @@ def test_hang_diagnostician_step_path():
     assert action is not None and "hang" in action.reason
     JobContext._reset_for_tests()
     Context._reset_for_tests()
+
+
+def test_precheck_operator_chain(monkeypatch):
+    """Pluggable pre-check chain gates worker start (ref
+    precheck_operator.py): min_nodes holds CHECKING, fails terminally on
+    timeout, and the master serves the aggregated status."""
+    from dlrover_amd.master.precheck import (
+        CHECKING,
+        FAIL,
+        PASS,
+        MinNodesPreCheckOperator,
+        PreCheckChain,
+        PreCheckOperator,
+        register_precheck_operator,
+    )
+
+    class Master:
+        rdzv_managers = {}
+
+    chain = PreCheckChain(["no_check", "bogus_name"])
+    assert chain.evaluate(Master()) == (PASS, "")
+
+    class Rdzv:
+        _alive_nodes = [0]
+        _waiting_nodes = {}
+        min_nodes = 2
+
+    from dlrover_amd.common.constants import RendezvousName
+
+    m = Master()
+    m.rdzv_managers = {RendezvousName.TRAINING: Rdzv()}
+    chain = PreCheckChain(["min_nodes"])
+    status, msg = chain.evaluate(m)
+    assert status == CHECKING and "1/2" in msg
+    # timeout -> terminal FAIL (sticky)
+    chain.ops[0].timeout_s = 0.0
+    status, _ = chain.evaluate(m)
+    assert status == FAIL
+    Rdzv._waiting_nodes = {1: 1}
+    status, _ = chain.evaluate(m)
+    assert status == FAIL  # sticky once failed
+
+    # satisfied chain passes
+    chain2 = PreCheckChain(["min_nodes"])
+    Rdzv._alive_nodes = [0, 1]
+    assert chain2.evaluate(m)[0] == PASS
+
+    # custom registered operator
+    @register_precheck_operator
+    class Always(PreCheckOperator):
+        name = "always_fail"
+
+        def check(self, master):
+            return FAIL, "nope"
+
+    chain3 = PreCheckChain(["always_fail"])
+    assert chain3.evaluate(m)[0] == FAIL
