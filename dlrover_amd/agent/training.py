@@ -476,6 +476,9 @@ class ElasticTrainingAgent(LocalElasticAgent):
                 "worker python stacks at hang:\n%s",
                 py_tracer.aggregate_stacks(stacks),
             )
+            kstacks = py_tracer.read_kernel_stacks(pids)
+            for rank, ks in sorted(kstacks.items()):
+                logger.info("worker rank %s kernel stacks:\n%s", rank, ks)
         except Exception as e:  # noqa: BLE001 — diagnosis must not block recovery
             logger.warning("python stack dump failed: %s", e)
 

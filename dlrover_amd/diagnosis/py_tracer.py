@@ -148,3 +148,33 @@ def _fmt_ranks(ranks: Iterable[int]) -> str:
         start = prev = r
     spans.append((start, prev))
     return ",".join(f"{a}-{b}" if a != b else f"{a}" for a, b in spans)
+
+
+def read_kernel_stacks(pids: Dict[int, int], max_threads: int = 16) -> Dict[int, str]:
+    """Kernel-side stacks of worker processes (/proc/<pid>/task/*/stack,
+    root-readable). The MI355X-relevant signal: a rank wedged inside an
+    amdgpu wait ioctl shows up here even when its Python stack looks idle —
+    the native complement to the SIGUSR2 Python dumps (ref: the reference's
+    daemon-side native unwind, stack_util.cc)."""
+    import glob
+    import os
+
+    out: Dict[int, str] = {}
+    for rank, pid in pids.items():
+        frames = []
+        tasks = sorted(glob.glob(f"/proc/{pid}/task/*/stack"))[:max_threads]
+        for t in tasks:
+            tid = t.split("/")[-2]
+            try:
+                with open(t) as f:
+                    body = f.read().strip()
+            except OSError:
+                continue
+            if body:
+                frames.append(f"  tid {tid}:\n" + "\n".join(
+                    "    " + ln for ln in body.splitlines()[:12]))
+        if frames:
+            out[rank] = "\n".join(frames)
+        elif os.path.exists(f"/proc/{pid}"):
+            out[rank] = "  (kernel stacks unreadable)"
+    return out

@@ -126,3 +126,13 @@ def test_py_metrics_served_by_prometheus_exporter(tmp_path):
         assert 'py_dataloader_batches{rank="0"}' in body
     finally:
         exp.stop()
+
+
+def test_read_kernel_stacks_self():
+    import os
+
+    from dlrover_amd.diagnosis.py_tracer import read_kernel_stacks
+
+    out = read_kernel_stacks({0: os.getpid()})
+    # root in this container: our own kernel stacks are readable
+    assert 0 in out and out[0]
