@@ -65,3 +65,64 @@ def test_failure_path_persist(agent_env):
     torch.testing.assert_close(data["model"]["w"], torch.ones(4))
     engine.close()
     engine.shm_handler.unlink()
+
+
+def test_reader_reattaches_after_segment_regrowth(tmp_path, monkeypatch):
+    """ADVICE r01: a writer that outgrows its segment unlinks+recreates it;
+    a cached reader mapping must re-attach instead of silently reading the
+    dead segment forever."""
+    import uuid
+
+    import torch
+
+    from dlrover_amd.trainer.flash_checkpoint.shm_handler import (
+        SharedMemoryHandler,
+        shm_segment_name,
+    )
+
+    name = shm_segment_name(f"regrow{uuid.uuid4().hex[:6]}", 0)
+    writer = SharedMemoryHandler(name, host_pin=False)
+    writer.save_state_dict(1, {"t": torch.ones(8)})
+
+    reader = SharedMemoryHandler(name, host_pin=False)
+    assert reader.committed_step() == 1
+
+    # writer outgrows the segment -> unlink + recreate bigger
+    writer.save_state_dict(2, {"t": torch.ones(1 << 22)})
+    assert reader.committed_step() == 2  # re-attached, not stale
+    meta = reader.read_meta()
+    assert meta is not None and meta.tensors[0].nbytes == (1 << 22) * 4
+    writer.unlink()
+
+
+def test_persist_mismatched_step_keeps_tracker_consistent(tmp_path):
+    """when shm holds a DIFFERENT step than the event, the shard
+    directory must be derived from the shm step so the tracker never points
+    at a directory that does not exist."""
+    import uuid
+
+    import torch
+
+    from dlrover_amd.agent.ckpt_saver import persist_shm_to_storage
+    from dlrover_amd.common.storage import PosixDiskStorage, read_tracker_step
+    from dlrover_amd.trainer.flash_checkpoint.engine import CheckpointEvent
+    from dlrover_amd.trainer.flash_checkpoint.shm_handler import (
+        SharedMemoryHandler,
+        shm_segment_name,
+    )
+
+    name = shm_segment_name(f"mm{uuid.uuid4().hex[:6]}", 0)
+    h = SharedMemoryHandler(name, host_pin=False)
+    h.save_state_dict(5, {"t": torch.ones(4)},
+                      extra={"global_rank": 0, "expected_shards": 1})
+    ckpt_dir = str(tmp_path / "ckpt")
+    event = CheckpointEvent(step=9, path=f"{ckpt_dir}/9", local_rank=0,
+                            global_rank=0)
+    ok = persist_shm_to_storage(h, event, PosixDiskStorage(), ckpt_dir, 1)
+    assert ok
+    step = read_tracker_step(ckpt_dir)
+    assert step == 5
+    import os
+
+    assert os.path.exists(os.path.join(ckpt_dir, "5", "rank_00000.pt"))
+    h.unlink()
