@@ -55,6 +55,11 @@ class JobMaster:
             job_context=self.ctx, rdzv_managers=self.rdzv_managers
         )
         self.job_manager.rdzv_managers = self.rdzv_managers
+        # dead worker -> its in-flight data shards re-queue (ref:
+        # event_callback.py TaskRescheduleCallback)
+        self.job_manager.add_node_failure_callback(
+            lambda node, reason: self.task_manager.recover_tasks(node.id)
+        )
         self._elastic_run_configs = elastic_run_configs or {}
         self._service_type = service_type or cfg.master_service_type
         # port=0 means "bind an ephemeral port" (standalone); None = default

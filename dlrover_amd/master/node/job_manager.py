@@ -47,6 +47,14 @@ class JobManager:
         self._stop = threading.Event()
         self._threads: List[threading.Thread] = []
         self._failures: List[comm.NodeFailure] = []
+        # pluggable node-event callbacks (ref: event_callback.py —
+        # TaskRescheduleCallback re-shards a dead worker's data,
+        # AllReduceNodeHandlingCallback drives rdzv):
+        # each is called as cb(node, reason) when a node FAILS
+        self.node_failure_callbacks: List = []
+
+    def add_node_failure_callback(self, cb):
+        self.node_failure_callbacks.append(cb)
 
     # -- lifecycle -----------------------------------------------------------------
 
@@ -160,6 +168,11 @@ class JobManager:
     def _handle_node_failure(self, node: Node, reason: str = ""):
         for mgr in self.rdzv_managers.values():
             mgr.remove_alive_node(node.id)
+        for cb in self.node_failure_callbacks:
+            try:
+                cb(node, reason)
+            except Exception:  # noqa: BLE001 — callbacks must not block recovery
+                logger.exception("node-failure callback %s failed", cb)
         if node.exit_reason == NodeExitReason.NO_HEARTBEAT:
             # the node's AGENT is gone (scale-down, preemption, host death):
             # nobody is listening for a restart action. Shrink the job — the

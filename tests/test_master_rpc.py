@@ -102,8 +102,16 @@ def test_shard_recovery_on_failure(master, client):
     )
     t1 = client.get_task("ds2")
     assert not t1.empty
-    # node dies: master recovers its in-flight shard
-    master.task_manager.recover_tasks(node_id=0)
+    # node dies: the failure callback re-queues its in-flight shard
+    from dlrover_amd.common import comm as _c
+    from dlrover_amd.common.constants import NodeEventType as _NE
+    from dlrover_amd.common.constants import NodeType as _NT
+
+    master.job_manager.on_node_event(_c.NodeEvent(
+        event_type=_NE.FAILED_EXITED,
+        node=_c.NodeMeta(type=_NT.WORKER, id=0, rank=0),
+        reason="crash",
+    ))
     t2 = client.get_task("ds2")
     assert (t2.start, t2.end) == (t1.start, t1.end)
 
