@@ -259,11 +259,22 @@ class NetworkCheckRendezvousManager(RendezvousManager):
             } & failed
         times = sorted(elapsed.values())
         if times:
-            median = times[len(times) // 2]
+            # baseline: true median for worlds >= 4; for tiny worlds the
+            # upper-middle "median" IS the straggler's own time, so compare
+            # against the fastest node instead
+            if len(times) >= 4:
+                mid = len(times) // 2
+                baseline = (
+                    times[mid]
+                    if len(times) % 2
+                    else 0.5 * (times[mid - 1] + times[mid])
+                )
+            else:
+                baseline = times[0]
             self._straggler_nodes = {
                 r
                 for r, t in elapsed.items()
-                if median > 0 and t > self.straggler_ratio * median
+                if baseline > 0 and t > self.straggler_ratio * baseline
             }
         logger.info(
             "[network-check] round %s elapsed=%s fault=%s straggler=%s",

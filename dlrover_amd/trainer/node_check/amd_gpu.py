@@ -34,6 +34,16 @@ def mock_error():
         raise RuntimeError(f"mock error on rank {err_rank}")
 
 
+def mock_straggle():
+    """Straggler injection: the chosen NODE's probe sleeps, so the master's
+    2-round pairing must isolate it (ref: fault_tolerance_exps.md straggler
+    localization — the BASELINE-cited chaos experiment)."""
+    rank = os.getenv("DLROVER_MOCK_STRAGGLER_NODE", "")
+    if rank and int(rank) == int(os.getenv("NODE_ID", os.getenv("RANK", "0"))):
+        secs = float(os.getenv("DLROVER_MOCK_STRAGGLE_SECS", "5"))
+        time.sleep(secs)
+
+
 def bm_matmul(device) -> float:
     """bf16 matmul TFLOPS probe (ref: utils.py:176). hipBLASLt GEMMs on the
     MFMA pipe; a sick GPU (downclocked, throttled, ECC-degraded HBM) lands
@@ -92,6 +102,7 @@ def main() -> int:
     try:
         mock_error()
         t0 = time.perf_counter()
+        mock_straggle()
         tflops = bm_matmul(device)
         busbw = bm_allreduce(device)
         elapsed = time.perf_counter() - t0

@@ -64,3 +64,39 @@ def test_node_check_mock_fault(tmp_path):
         assert "CHECK_RESULT failed" in out.stdout, (
             out.stdout[-2000:], out.stderr[-3000:]
         )
+
+
+@pytest.mark.timeout(300)
+def test_node_check_straggler_localization():
+    """Two nodes, one slow: the master's 2-round pairing flags the slow
+    node as a straggler over the real RPC path (the BASELINE-cited
+    chaos-experiment behavior, fault_tolerance_exps.md:149-156)."""
+    from dlrover_amd.agent.master_client import MasterClient
+    from dlrover_amd.testing import MasterProcess
+
+    env = {"ELASTIC_JOB_NAME": f"nc{uuid.uuid4().hex[:6]}"}
+    with MasterProcess(env) as master:
+        os.environ["DLROVER_MASTER_ADDR"] = master.addr
+        clients = [MasterClient(master.addr, node_id=0),
+                   MasterClient(master.addr, node_id=1)]
+        clients[0].report_rdzv_params(2, 2, 60, 1)
+        for round_times in ({0: 20.3, 1: 206.9}, {0: 20.1, 1: 201.8}):
+            for n, c in enumerate(clients):
+                c.join_rendezvous(n, 1, rdzv_name="network-check")
+            for n, c in enumerate(clients):
+                # poll until the check round's world forms
+                import time as _t
+
+                deadline = _t.time() + 60
+                while _t.time() < deadline:
+                    _, _, world = c.get_comm_world("network-check", n)
+                    if world:
+                        break
+                    _t.sleep(0.2)
+                assert world, f"node {n} never grouped"
+            for n, c in enumerate(clients):
+                c.report_network_check_result(n, True, round_times[n])
+        stragglers = clients[0].check_straggler()
+        assert stragglers == [1], stragglers
+        faults, _ = clients[0].check_fault_node()
+        assert faults == []  # slow, not broken
