@@ -52,16 +52,23 @@ class MinNodesPreCheckOperator(PreCheckOperator):
         mgr = master.rdzv_managers.get(RendezvousName.TRAINING)
         if mgr is None:
             return PASS, ""
+        need = getattr(mgr, "min_nodes", 1)
+        # agents ANNOUNCE themselves (report_node_event ADDED) before they
+        # block on the pre-check gate, so count announced nodes — counting
+        # rendezvous joins here would deadlock (joins happen post-gate)
+        announced = len([
+            n for n in master.ctx.job_nodes().values() if not n.eliminated
+        ]) if getattr(master, "ctx", None) else 0
         alive = len(getattr(mgr, "_alive_nodes", []) or [])
         waiting = len(getattr(mgr, "_waiting_nodes", {}) or {})
-        need = getattr(mgr, "min_nodes", 1)
-        if alive + waiting >= need:
+        have = max(announced, alive + waiting)
+        if have >= need:
             return PASS, ""
         if self._first == 0.0:
             self._first = time.time()
         if time.time() - self._first > self.timeout_s:
-            return FAIL, f"only {alive + waiting}/{need} nodes joined"
-        return CHECKING, f"{alive + waiting}/{need} nodes joined"
+            return FAIL, f"only {have}/{need} nodes joined"
+        return CHECKING, f"{have}/{need} nodes joined"
 
 
 class DeviceCountPreCheckOperator(PreCheckOperator):

@@ -185,6 +185,13 @@ def run(args) -> int:
     if args.hiptimer:
         os.environ["DLROVER_HIPTIMER"] = "1"
 
+    # announce this agent BEFORE the pre-check gate so master-side checks
+    # (e.g. the min_nodes operator) can count arrivals
+    try:
+        client.report_node_event("ADDED", "agent started", args.node_rank)
+    except Exception:  # noqa: BLE001
+        pass
+
     wait_pre_check(client)
 
     min_nodes, max_nodes = parse_nnodes(args.nnodes)
