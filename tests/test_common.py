@@ -104,3 +104,24 @@ def test_serialize_rejects_dangerous_builtins():
         loads(evil)
     assert loads(dumps({"a", 1})) == {"a", 1}
     assert loads(dumps(bytearray(b"x"))) == bytearray(b"x")
+
+
+def test_keep_step_interval_strategy(tmp_path):
+    """Checkpoints off the keep-interval grid are deleted once committed
+    (ref storage.py KeepStepIntervalStrategy)."""
+    import os
+
+    from dlrover_amd.common.storage import (
+        KeepStepIntervalStrategy,
+        PosixStorageWithDeletion,
+    )
+
+    st = PosixStorageWithDeletion(
+        str(tmp_path), KeepStepIntervalStrategy(keep_interval=10)
+    )
+    for step in (5, 10, 15, 20):
+        os.makedirs(tmp_path / str(step), exist_ok=True)
+        (tmp_path / str(step) / "x.pt").write_bytes(b"d")
+        st.commit(step, True)
+    remaining = {d.name for d in tmp_path.iterdir() if d.is_dir()}
+    assert remaining == {"10", "20"}, remaining

@@ -210,3 +210,40 @@ def test_derived_disabled_by_env(tmp_path, monkeypatch):
     sd = eng.gather_state_dict(model, opt)
     assert sd["_derived"] == {}
     assert all(n in sd["model"] for n, _ in model.named_parameters())
+
+
+def test_save_to_memory_skips_when_peer_not_ready(monkeypatch):
+    """A rank that cannot proceed must not strand peers writing
+    inconsistent steps (ref engine.py:60 readiness all-reduce)."""
+    import uuid
+
+    import torch
+
+    from dlrover_amd.trainer.flash_checkpoint.engine import (
+        ShardedCheckpointEngine,
+    )
+
+    eng = ShardedCheckpointEngine(f"/tmp/nr{uuid.uuid4().hex[:6]}")
+    # single-process: force the readiness check to report a bad peer
+    monkeypatch.setattr(eng, "_check_all_ranks_ready", lambda ready: False)
+    blocking = eng.save_to_memory(3, {"t": torch.ones(2)})
+    assert blocking == 0.0
+    assert eng.shm_handler.committed_step() == 0  # nothing written
+    eng.shm_handler.unlink()
+
+
+def test_save_step_mismatch_raises(monkeypatch):
+    import uuid
+
+    import pytest
+    import torch
+
+    from dlrover_amd.trainer.flash_checkpoint.engine import (
+        ShardedCheckpointEngine,
+    )
+
+    eng = ShardedCheckpointEngine(f"/tmp/sm{uuid.uuid4().hex[:6]}")
+    monkeypatch.setattr(eng, "_check_step_consistent", lambda step: False)
+    with pytest.raises(RuntimeError, match="differs across ranks"):
+        eng.save_to_memory(3, {"t": torch.ones(2)})
+    eng.shm_handler.unlink()
