@@ -179,3 +179,44 @@ def test_tp2_pp2_llama_and_megatron_ckpt(tmp_path):
             p.join(timeout=360)
         outcomes = dict(results)
     assert all(outcomes.get(r) == "ok" for r in range(WS)), outcomes
+
+
+@pytest.mark.timeout(420)
+def test_tp_pp_example_e2e(tmp_path):
+    """The user-facing TP x PP example trains, checkpoints and resumes on
+    4 gloo ranks (config #3 entry point)."""
+    import json
+    import subprocess
+    import sys
+    import uuid
+
+    progress = tmp_path / "prog.jsonl"
+    env = dict(os.environ)
+    env.update({
+        "ELASTIC_JOB_NAME": f"tpex{uuid.uuid4().hex[:6]}",
+        "DLROVER_IPC_SOCKET_DIR": str(tmp_path / "ipc"),
+        "MASTER_ADDR": "127.0.0.1",
+    })
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run", "--standalone",
+        "--master-addr", "127.0.0.1", "--nproc-per-node", "4",
+        os.path.join(root, "examples", "train_llama_tp_pp.py"),
+        "--model", "tiny", "--tp", "2", "--pp", "2", "--steps", "4",
+        "--seq", "16", "--ckpt-interval", "2",
+        "--ckpt-dir", str(tmp_path / "ckpt"),
+        "--progress-file", str(progress),
+    ]
+    out = subprocess.run(cmd, cwd=root, env=env, capture_output=True,
+                         text=True, timeout=360)
+    assert out.returncode == 0, out.stderr[-4000:]
+    rows = [json.loads(l) for l in progress.read_text().splitlines()]
+    assert rows[-1]["step"] == 4 and rows[-1]["loss"] is not None
+    # resume: a second run continues from the committed step
+    cmd[cmd.index("--steps") + 1] = "6"
+    out2 = subprocess.run(cmd, cwd=root, env=env, capture_output=True,
+                          text=True, timeout=360)
+    assert out2.returncode == 0, out2.stderr[-4000:]
+    rows = [json.loads(l) for l in progress.read_text().splitlines()]
+    steps = [r["step"] for r in rows]
+    assert steps[-1] == 6 and 5 in steps and steps.count(4) == 1, steps
