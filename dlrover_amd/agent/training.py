@@ -251,6 +251,8 @@ class ElasticTrainingAgent(LocalElasticAgent):
     # -- failure-path checkpoint persist (ref: training.py:1533) --------------------
 
     def _save_ckpt_to_storage(self):
+        if os.getenv("DLROVER_NO_BREAKPOINT_SAVE", "") == "1":
+            return  # --save-at-breakpoint escape hatch (default: always on)
         saver = AsyncCheckpointSaver.get_ckpt_saver()
         if saver is not None:
             try:

@@ -69,7 +69,12 @@ def parse_args(argv: Optional[List[str]] = None):
     p.add_argument("--auto-tunning", action="store_true")
     p.add_argument("--numa-affinity", action="store_true")
     p.add_argument("--accelerator", default="amd.com/gpu")
-    p.add_argument("--save-at-breakpoint", "--save_at_breakpoint", action="store_true")
+    p.add_argument("--save-at-breakpoint", "--save_at_breakpoint",
+                   action="store_true",
+                   help="accepted for reference-CLI compatibility: the agent "
+                        "ALWAYS persists the latest shm checkpoint to "
+                        "storage on worker failure/SIGTERM in this build "
+                        "(set DLROVER_NO_BREAKPOINT_SAVE=1 to disable)")
     p.add_argument("--hiptimer", action="store_true",
                    help="LD_PRELOAD the hiptimer profiler into workers "
                         "(kernel/GEMM/RCCL timing + hang detection)")
