@@ -16,12 +16,16 @@ class ElasticDataLoader(DataLoader):
 
     def update_batch_size(self, batch_size: Optional[int] = None):
         """Apply a master-pushed batch size (takes effect on next epoch's
-        iterator). When batch_size is None, ask the master."""
+        iterator). When batch_size is None, ask the master; VERSIONED
+        suggestions are applied only once per version (--auto-tunning)."""
         if batch_size is None:
             try:
                 from dlrover_amd.agent.master_client import MasterClient
 
                 cfg = MasterClient.singleton_instance().get_paral_config()
+                if cfg.dataloader.version <= self._config_version:
+                    return
+                self._config_version = cfg.dataloader.version
                 batch_size = cfg.dataloader.batch_size or None
             except Exception:  # noqa: BLE001
                 return
@@ -31,3 +35,12 @@ class ElasticDataLoader(DataLoader):
             if self.batch_sampler is not None:
                 self.batch_sampler.batch_size = batch_size
             logger.info("dataloader batch size -> %s", batch_size)
+
+
+    def maybe_autotune(self):
+        """Poll the master's versioned suggestion when --auto-tunning is on
+        (DLROVER_AUTO_TUNE); call at epoch boundaries."""
+        import os
+
+        if os.getenv("DLROVER_AUTO_TUNE", "") == "1":
+            self.update_batch_size()

@@ -65,8 +65,12 @@ def parse_args(argv: Optional[List[str]] = None):
     p.add_argument("--comm-perf-test", action="store_true")
     p.add_argument("--node-unit", "--node_unit", type=int, default=1,
                    help="world size must be a multiple of this")
-    p.add_argument("--auto-config", action="store_true")
-    p.add_argument("--auto-tunning", action="store_true")
+    p.add_argument("--auto-config", action="store_true",
+                   help="fill nproc-per-node from the visible GPU count "
+                        "(ref auto_configure_params)")
+    p.add_argument("--auto-tunning", action="store_true",
+                   help="let workers poll master-generated hyperparameter "
+                        "suggestions (versioned ParallelConfig)")
     p.add_argument("--numa-affinity", action="store_true")
     p.add_argument("--accelerator", default="amd.com/gpu")
     p.add_argument("--save-at-breakpoint", "--save_at_breakpoint",
@@ -200,6 +204,10 @@ def run(args) -> int:
     wait_pre_check(client)
 
     min_nodes, max_nodes = parse_nnodes(args.nnodes)
+    if args.auto_config and args.nproc_per_node <= 0:
+        args.nproc_per_node = 0  # auto_configure() fills from device count
+    if args.auto_tunning:
+        os.environ["DLROVER_AUTO_TUNE"] = "1"
     config = ElasticLaunchConfig(
         min_nodes=min_nodes,
         max_nodes=max_nodes,
@@ -220,6 +228,7 @@ def run(args) -> int:
         run_id=args.run_id or os.getenv("ELASTIC_JOB_NAME", "dlrover"),
     )
 
+    config.auto_configure()
     script = args.training_script
     script_args = list(args.training_script_args)
     if script.endswith(".py"):
