@@ -202,6 +202,7 @@ class ElasticTrainingAgent(LocalElasticAgent):
         self.config = config
         self.client = client or MasterClient.singleton_instance()
         self._restart_requested = threading.Event()
+        self._last_descendants: List[int] = []
         self._abort_requested = threading.Event()
         self._hb_stop = threading.Event()
         self._hb_thread: Optional[threading.Thread] = None
@@ -337,6 +338,13 @@ class ElasticTrainingAgent(LocalElasticAgent):
                         "failure_restart",
                         {"restarts_left": self._remaining_restarts},
                     ):
+                        # dead workers can leave descendants (dataloader
+                        # procs) holding HIP contexts. The parents are
+                        # already gone, so sweep the set recorded on the
+                        # last HEALTHY tick — and BEFORE the new workers
+                        # spawn (no chance of touching fresh pids)
+                        self._sweep_orphans(self._last_descendants)
+                        self._last_descendants = []
                         self._restart_workers(self._worker_group)
                     continue
                 self._stop_workers(self._worker_group)
@@ -345,6 +353,9 @@ class ElasticTrainingAgent(LocalElasticAgent):
                 return run_result
 
             if state == WorkerState.HEALTHY:
+                # descendants snapshot for the failure-path orphan sweep
+                # (a dead parent's tree cannot be walked post-mortem)
+                self._last_descendants = self._worker_descendants()
                 if self._restart_requested.is_set():
                     self._restart_requested.clear()
                     self._dump_worker_py_stacks()
