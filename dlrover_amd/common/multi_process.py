@@ -34,6 +34,10 @@ def ipc_socket_path(job_name: str = "") -> str:
     job = job_name or os.getenv("ELASTIC_JOB_NAME", "default")
     d = os.path.join(base, job)
     os.makedirs(d, exist_ok=True)
+    try:
+        os.chmod(d, 0o700)  # raw pickle flows over this socket: owner-only
+    except OSError:
+        pass
     return os.path.join(d, "ipc.sock")
 
 
