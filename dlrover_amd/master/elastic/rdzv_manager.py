@@ -134,7 +134,10 @@ class RendezvousManager:
             waiting >= p.min_nodes
             and self._lastcall_time > 0
             and time.time() - self._lastcall_time > p.waiting_timeout
-            and waiting % p.node_unit == 0
+            # ref :183 TRUNCATES to the node_unit multiple on timeout rather
+            # than requiring an exact multiple (5 waiting @ unit 2 -> 4);
+            # only refuse when truncation would fall below min_nodes
+            and (waiting // p.node_unit) * p.node_unit >= p.min_nodes
         ):
             completed = True
         if not completed:

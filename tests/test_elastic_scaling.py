@@ -92,3 +92,27 @@ def test_group_network_check_pairing_and_link_suspects():
     for r in range(8):
         mgr.report_network_check_result(r, r != 5, 1.0)
     assert (0, 1) in mgr.suspect_group_links()
+
+
+def test_lastcall_timeout_truncates_to_node_unit(monkeypatch):
+    """5 nodes waiting with node_unit=2 at last-call timeout -> a world of 4
+    (truncate, ref rdzv_manager.py:183), not a stall."""
+    import time as _t
+
+    from dlrover_amd.master.elastic.rdzv_manager import (
+        ElasticTrainingRendezvousManager,
+    )
+
+    mgr = ElasticTrainingRendezvousManager()
+    mgr.update_rdzv_params(2, 8, waiting_timeout=0.2, node_unit=2)
+    for r in range(8):
+        mgr.add_alive_node(r)  # 8 alive, only 5 will join this round
+    for r in range(5):
+        mgr.join_rendezvous(r, 1)
+    # 5 of 8 alive joined: must wait for the last call
+    _, _, world = mgr.get_comm_world(0)
+    assert world == {}
+    _t.sleep(0.3)
+    _, _, world = mgr.get_comm_world(0)
+    assert sorted(world) == [0, 1, 2, 3], world  # truncated, rank 4 dropped
+    assert mgr.num_nodes_waiting() == 1  # the dropped node waits for next round
