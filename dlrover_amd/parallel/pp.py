@@ -47,10 +47,16 @@ class PipelineRunner:
     """Drives one optimizer step of a staged model across the PP group."""
 
     def __init__(self, stage, groups: ParallelGroups, hidden_size: int,
-                 act_dtype: torch.dtype = torch.float32):
+                 act_dtype: Optional[torch.dtype] = None):
         self.stage = stage
         self.g = groups
         self.hidden = hidden_size
+        # activation dtype must match what neighboring stages SEND — infer
+        # from the stage parameters (a bf16 stage exchanging into fp32 recv
+        # buffers would corrupt the pipeline)
+        if act_dtype is None:
+            p = next(stage.parameters(), None)
+            act_dtype = p.dtype if p is not None else torch.float32
         self.act_dtype = act_dtype
 
     def _stage_device(self):
