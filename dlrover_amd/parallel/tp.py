@@ -24,7 +24,10 @@ class _CopyToTP(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad):
-        if ctx.group is not None or dist.get_world_size() > 1:
+        # group None == tp degenerate (pgroups convention): NO collective.
+        # Falling through to the default group would all-reduce across pp/dp
+        # ranks — wrong gradients and a deadlock across pipeline stages.
+        if ctx.group is not None:
             grad = grad.contiguous()
             dist.all_reduce(grad, group=ctx.group)
         return grad, None
@@ -35,6 +38,8 @@ class _ReduceFromTP(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, group):
+        if group is None:  # tp degenerate: nothing to reduce
+            return x
         x = x.contiguous()
         dist.all_reduce(x, group=group)
         return x

@@ -220,3 +220,30 @@ def test_tp_pp_example_e2e(tmp_path):
     rows = [json.loads(l) for l in progress.read_text().splitlines()]
     steps = [r["step"] for r in rows]
     assert steps[-1] == 6 and 5 in steps and steps.count(4) == 1, steps
+
+
+@pytest.mark.timeout(420)
+def test_tp1_pp2_example_does_not_world_allreduce(tmp_path):
+    """tp=1 with pp=2 must not fall through to whole-world collectives
+    (group None == degenerate): the example trains cleanly at tp1 pp2."""
+    import subprocess
+    import sys
+    import uuid
+
+    env = dict(os.environ)
+    env.update({
+        "ELASTIC_JOB_NAME": f"tp1{uuid.uuid4().hex[:6]}",
+        "DLROVER_IPC_SOCKET_DIR": str(tmp_path / "ipc"),
+        "MASTER_ADDR": "127.0.0.1",
+    })
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--standalone",
+         "--master-addr", "127.0.0.1", "--nproc-per-node", "2",
+         os.path.join(root, "examples", "train_llama_tp_pp.py"),
+         "--model", "tiny", "--tp", "1", "--pp", "2", "--steps", "3",
+         "--seq", "16", "--ckpt-interval", "2",
+         "--ckpt-dir", str(tmp_path / "ckpt")],
+        cwd=root, env=env, capture_output=True, text=True, timeout=360,
+    )
+    assert out.returncode == 0, out.stderr[-4000:]
