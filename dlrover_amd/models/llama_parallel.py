@@ -58,6 +58,17 @@ class ParallelAttention(nn.Module):
         q = rope_rotate(q.view(B, S, nh, hd), pos, cos, sin)
         k = rope_rotate(k.view(B, S, nkv, hd), pos, cos, sin)
         v = v.view(B, S, nkv, hd)
+        if x.is_cuda and hd == 128 and S % 64 == 0:
+            # hand-written MFMA flash path (same conditions as llama.py —
+            # the kernels are GQA/stride-aware, per-TP-shard head counts)
+            from dlrover_amd.ops import flash_attention
+
+            out = flash_attention(
+                q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
+                1.0 / math.sqrt(hd),
+            )
+            out = out.transpose(1, 2).reshape(B, S, nh * hd)
+            return self.o_proj(out)
         rep = nh // nkv
         q = q.view(B, S, nkv, rep, hd).permute(0, 2, 3, 1, 4).reshape(
             B, nkv, rep * S, hd
