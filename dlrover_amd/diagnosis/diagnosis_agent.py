@@ -41,8 +41,15 @@ _FATAL_USER_PATTERNS = [
     r"ModuleNotFoundError",
     r"ImportError",
     r"SyntaxError",
+]
+
+# OOM is RELAUNCH, not abort: the master grows the replacement pod's memory
+# request (job_manager OOM recovery); repeated OOMs exhaust the relaunch
+# budget and only then abort
+_OOM_PATTERNS = [
     r"torch\.OutOfMemoryError",
     r"(HIP|CUDA) out of memory",
+    r"hipErrorOutOfMemory",
 ]
 
 
@@ -77,6 +84,9 @@ def classify_error(error_text: str) -> str:
     """-> RESTART_WORKER | RELAUNCH_NODE | ABORT_JOB."""
     for pat in _HARDWARE_PATTERNS:
         if re.search(pat, error_text, re.IGNORECASE):
+            return RELAUNCH_NODE
+    for pat in _OOM_PATTERNS:
+        if re.search(pat, error_text):
             return RELAUNCH_NODE
     for pat in _FATAL_USER_PATTERNS:
         if re.search(pat, error_text):

@@ -34,7 +34,9 @@ def test_classify_hardware_error():
 
 def test_classify_fatal_user_error():
     assert classify_error("ModuleNotFoundError: No module named 'foo'") == ABORT_JOB
-    assert classify_error("torch.OutOfMemoryError: HIP out of memory") == ABORT_JOB
+    # OOM relaunches (the master grows the replacement pod's memory);
+    # repeated OOMs exhaust the budget and only then abort
+    assert classify_error("torch.OutOfMemoryError: HIP out of memory") == RELAUNCH_NODE
 
 
 def test_restart_budget_escalates():
