@@ -217,12 +217,17 @@ class PodScaler:
         """Scale the worker group to `count` (ref: ScalePlan execution)."""
         alive = [n for n in current_nodes if n.is_alive()]
         if len(alive) < count:
+            # pod ids only ever grow, but rank indices must re-fill the holes
+            # a previous scale-down left so the final rank space is a
+            # contiguous 0..count-1 (torchrun node ranks)
             base = max((n.id for n in current_nodes), default=-1) + 1
-            for i in range(count - len(alive)):
+            used_ranks = {n.rank_index for n in alive}
+            free_ranks = [r for r in range(count) if r not in used_ranks]
+            for i, rank in enumerate(free_ranks[: count - len(alive)]):
                 node = Node(
                     NodeType.WORKER,
                     base + i,
-                    rank_index=base + i,
+                    rank_index=rank,
                     config_resource=NodeResource(gpu_num=8),
                 )
                 self.launch_node(node)

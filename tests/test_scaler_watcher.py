@@ -71,6 +71,31 @@ def test_scaler_scale_to():
         scaler.stop()
 
 
+def test_scaler_scale_up_refills_rank_holes():
+    """After a scale-down removed ranks 2-3, scaling back up must re-fill
+    ranks 2-3 (contiguous 0..count-1), not extend to 4-5."""
+    api = FakeK8sApi()
+    scaler = PodScaler("jobx", api=api)
+    try:
+        nodes = [
+            Node(NodeType.WORKER, i, status=NodeStatus.RUNNING) for i in range(4)
+        ]
+        for dead in nodes[2:]:
+            dead.status = NodeStatus.FAILED
+        scaler.scale_to(4, nodes)
+        assert _wait(lambda: len(api.created) == 2)
+        ranks = sorted(
+            int(api.pods[name]["metadata"]["labels"][
+                "elasticjob.dlrover/rank-index"])
+            for name in api.created
+        )
+        assert ranks == [2, 3]
+        # ids still grow monotonically (no pod-name reuse)
+        assert set(api.created) == {"jobx-worker-4", "jobx-worker-5"}
+    finally:
+        scaler.stop()
+
+
 def _pod(idx, phase, reason=""):
     pod = {
         "metadata": {
