@@ -27,6 +27,9 @@ class BrainOptimizer:
 
     def __init__(self):
         self._metrics: Dict[str, List[dict]] = defaultdict(list)
+        # job -> (node_count, per_node_speed) at the last grow suggestion,
+        # used to verify the growth actually scaled before growing again
+        self._last_grow: Dict[str, tuple] = {}
         self._lock = threading.Lock()
 
     def report(self, job: str, metrics: dict):
@@ -57,7 +60,14 @@ class BrainOptimizer:
         per_node = avg / cur
         plan = {"node_count": cur, "comment": "brain:hold"}
         max_nodes = int(stats.get("max_nodes", cur))
-        if per_node > 0 and cur < max_nodes:
+        prev = self._last_grow.get(job)
+        # grow only while scaling stays near-linear: per-node speed after the
+        # previous doubling must be within 15% of what it was before it
+        scaled_ok = (
+            prev is None or cur <= prev[0] or per_node >= 0.85 * prev[1]
+        )
+        if per_node > 0 and cur < max_nodes and scaled_ok:
+            self._last_grow[job] = (cur, per_node)
             plan = {"node_count": min(cur * 2, max_nodes),
                     "comment": "brain:grow"}
         elif avg == 0 and cur > 1:

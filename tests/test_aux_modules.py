@@ -213,5 +213,12 @@ def test_brain_service_client_roundtrip():
             "j2", "running", {"current_nodes": 4, "max_nodes": 8}
         )
         assert plan["comment"] == "brain:shrink" and plan["node_count"] == 2
+        # sub-linear scaling gate: after growing 2->4, a collapsed per-node
+        # speed (same aggregate steps/s on 2x nodes = 50% per-node) holds
+        # instead of growing again
+        plan = c.get_optimization_plan(
+            "j1", "running", {"current_nodes": 4, "max_nodes": 8}
+        )
+        assert plan == {"node_count": 4, "comment": "brain:hold"}
     finally:
         svc.stop()
