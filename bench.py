@@ -309,13 +309,14 @@ def main():
     t_total = time.perf_counter() - t_begin
     hb(f"timed window done: {t_total:.2f} s")
 
-    # ---- aggregate across ranks: total = max, useful = min (conservative)
-    stats = torch.tensor([t_total, useful], dtype=torch.float64)
+    # ---- aggregate across ranks: total = max, useful = min (conservative).
+    # tensor must live on the backend's device: the N>1 GPU group is
+    # NCCL-only, which cannot reduce a CPU tensor
+    stats = torch.tensor([t_total, useful], dtype=torch.float64,
+                         device=device if on_gpu else "cpu")
     if world > 1:
-        tot = stats.clone()
-        dist.all_reduce(tot[:1], op=dist.ReduceOp.MAX)
-        dist.all_reduce(tot[1:], op=dist.ReduceOp.MIN)
-        stats = tot
+        dist.all_reduce(stats[:1], op=dist.ReduceOp.MAX)
+        dist.all_reduce(stats[1:], op=dist.ReduceOp.MIN)
     t_total, useful = stats[0].item(), stats[1].item()
 
     goodput = 100.0 * useful / t_total

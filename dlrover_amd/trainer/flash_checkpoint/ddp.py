@@ -52,8 +52,14 @@ class DdpCheckpointer(Checkpointer):
             import torch
 
             my_step = int(sd.get("step", -1)) if sd is not None else -1
-            t = torch.tensor([my_step], dtype=torch.long)
-            dist.all_reduce(t, op=dist.ReduceOp.MIN)
+            # route through the engine's gloo side group: the default group
+            # on a GPU job is NCCL-only and cannot reduce this CPU tensor
+            group = getattr(self.engine, "_sync_group", None)
+            dev = "cpu"
+            if group is None and "nccl" in str(dist.get_backend()).lower():
+                dev = f"cuda:{torch.cuda.current_device()}"
+            t = torch.tensor([my_step], dtype=torch.long, device=dev)
+            dist.all_reduce(t, op=dist.ReduceOp.MIN, group=group)
             agreed = int(t.item())
             if agreed < 0:
                 return None  # at least one rank has nothing: fresh start

@@ -55,6 +55,19 @@ class ReplicaManager:
                  group_size: int = 0):
         self.handler = shm_handler
         self.group = group  # gloo group (collectives carry host bytes)
+        # on a GPU job the default group is NCCL-only and cannot move the
+        # host byte buffers these collectives carry: make a gloo side group
+        # (every rank constructs the manager, so new_group is collective-safe)
+        if (
+            self.group is None
+            and dist.is_available()
+            and dist.is_initialized()
+            and "gloo" not in str(dist.get_backend()).lower()
+        ):
+            try:
+                self.group = dist.new_group(backend="gloo")
+            except (RuntimeError, ValueError):
+                pass
         self.group_size = group_size
         self._backup_shm = None
 

@@ -84,8 +84,12 @@ def gather_new_offsets(local_rows: int, group=None) -> List[int]:
     import torch.distributed as dist
 
     ws = dist.get_world_size(group)
-    t = torch.tensor([local_rows], dtype=torch.long)
-    out = [torch.zeros(1, dtype=torch.long) for _ in range(ws)]
+    # an NCCL(-only) group cannot gather CPU tensors: stage on the device
+    dev = "cpu"
+    if "nccl" in str(dist.get_backend(group)).lower():
+        dev = f"cuda:{torch.cuda.current_device()}"
+    t = torch.tensor([local_rows], dtype=torch.long, device=dev)
+    out = [torch.zeros(1, dtype=torch.long, device=dev) for _ in range(ws)]
     dist.all_gather(out, t, group=group)
     rows = [int(o.item()) for o in out]
     offsets, acc = [], 0
