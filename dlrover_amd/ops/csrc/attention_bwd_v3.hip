@@ -292,10 +292,10 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v3_kernel(
 
 // shared prologue macro for the two kv-major kernels
 #define FB_DKV_PROLOGUE()                                                    \
-  __shared__ char q_lds[2][FB_T * FB_D * 2];                                 \
-  __shared__ char do_lds[2][FB_T * FB_D * 2];                                \
-  __shared__ float l_lds[2][FB_T];                                           \
-  __shared__ float d_lds[2][FB_T];                                           \
+  __shared__ char q_lds[FB_T * FB_D * 2];                                    \
+  __shared__ char do_lds[FB_T * FB_D * 2];                                   \
+  __shared__ float l_lds[FB_T];                                              \
+  __shared__ float d_lds[FB_T];                                              \
   const int tid = threadIdx.x;                                               \
   const int wave = tid >> 6;                                                 \
   const int lane = tid & 63;                                                 \
@@ -318,38 +318,26 @@ __global__ __launch_bounds__(512) void fa_bwd_dq_v3_kernel(
   const int n_qt = S / FB_T - qt_min;                                        \
   const int n_work = rep * n_qt;                                             \
   const int wave_kv_min = kvb0 + wave * 32;                                  \
-  const int st_row0 = (tid * 16) / FB_D;                                     \
-  const int st_col0 = (tid * 16) % FB_D;                                     \
-  bf16x8 st_q[2], st_do[2];                                                  \
-  float st_l = 0.f, st_d = 0.f;                                              \
-  auto issue_qdo = [&](int h, int qt) {                                      \
+  auto stage_qdo = [&](int h, int qt) {                                      \
     const short* qsrc = q + (long long)b * qs_b + (long long)h * qs_h +      \
                         (long long)(qt * FB_T) * qs_s;                       \
     const short* dsrc = dout + (long long)b * ds_b + (long long)h * ds_h +   \
                         (long long)(qt * FB_T) * ds_s;                       \
     _Pragma("unroll") for (int c = 0; c < 2; ++c) {                          \
-      st_q[c] = *reinterpret_cast<const bf16x8*>(                            \
-          qsrc + (long long)st_row0 * qs_s + st_col0 + c * 8);               \
-      st_do[c] = *reinterpret_cast<const bf16x8*>(                           \
-          dsrc + (long long)st_row0 * ds_s + st_col0 + c * 8);               \
+      const int linear = (tid * 16) + c * 8;                                 \
+      const int row = linear / FB_D;                                         \
+      const int col = linear % FB_D;                                         \
+      *reinterpret_cast<bf16x8*>(q_lds + row * 256 + swzb(row, col * 2)) =   \
+          *reinterpret_cast<const bf16x8*>(qsrc + (long long)row * qs_s +    \
+                                           col);                             \
+      *reinterpret_cast<bf16x8*>(do_lds + row * 256 + swzb(row, col * 2)) =  \
+          *reinterpret_cast<const bf16x8*>(dsrc + (long long)row * ds_s +    \
+                                           col);                             \
     }                                                                        \
     if (tid < FB_T) {                                                        \
       const long long lrow = ((long long)b * H + h) * S + qt * FB_T + tid;   \
-      st_l = lse[lrow];                                                      \
-      st_d = dvec[lrow];                                                     \
-    }                                                                        \
-  };                                                                         \
-  auto write_qdo = [&](int buf) {                                            \
-    _Pragma("unroll") for (int c = 0; c < 2; ++c) {                          \
-      const int col = st_col0 + c * 8;                                       \
-      *reinterpret_cast<bf16x8*>(q_lds[buf] + st_row0 * 256 +                \
-                                 swzb(st_row0, col * 2)) = st_q[c];          \
-      *reinterpret_cast<bf16x8*>(do_lds[buf] + st_row0 * 256 +               \
-                                 swzb(st_row0, col * 2)) = st_do[c];         \
-    }                                                                        \
-    if (tid < FB_T) {                                                        \
-      l_lds[buf][tid] = st_l;                                                \
-      d_lds[buf][tid] = st_d;                                                \
+      l_lds[tid] = lse[lrow];                                                \
+      d_lds[tid] = dvec[lrow];                                               \
     }                                                                        \
   }
 
@@ -373,17 +361,13 @@ __global__ __launch_bounds__(512) void fa_bwd_dv_v3_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) acc_dv[n][r] = 0.f;
 
-    int cur = 0;
-    if (chunk < n_work) {
-      issue_qdo(g * rep + chunk / n_qt, qt_min + chunk % n_qt);
-      write_qdo(0);
-      __syncthreads();
-    }
     for (int w = chunk; w < n_work; w += CH) {
+      const int h = g * rep + w / n_qt;
       const int qt = qt_min + w % n_qt;
-      const int wn = w + CH;
-      if (wn < n_work) issue_qdo(g * rep + wn / n_qt, qt_min + wn % n_qt);
-      if (qt * FB_T + FB_T - 1 >= wave_kv_min) {
+      __syncthreads();
+      stage_qdo(h, qt);
+      __syncthreads();
+      if (qt * FB_T + FB_T - 1 < wave_kv_min) continue;  // fully masked
 #pragma unroll
       for (int st = 0; st < 2; ++st) {  // q subtiles of 32
         const int q0 = qt * FB_T + st * 32;
@@ -394,9 +378,11 @@ __global__ __launch_bounds__(512) void fa_bwd_dv_v3_kernel(
         for (int r = 0; r < 16; ++r) acc_s[r] = 0.f;
 #pragma unroll
         for (int ks = 0; ks < 8; ++ks) {
+          const int row = st * 32 + kcol;  // q row index == lane&31 pattern
           bf16x8 aq = *reinterpret_cast<const bf16x8*>(
-              q_lds[cur] + (st * 32 + kcol) * 256 +
+              q_lds + (st * 32 + kcol) * 256 +
               swzb(st * 32 + kcol, (ks * 16 + half * 8) * 2));
+          (void)row;
           acc_s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(aq, bk[ks], acc_s,
                                                           0, 0, 0);
         }
@@ -408,7 +394,7 @@ __global__ __launch_bounds__(512) void fa_bwd_dv_v3_kernel(
           const int q_glob = qt * FB_T + qrow;
           pv[r] = (kv_own > q_glob)
                       ? 0.f
-                      : __expf(acc_s[r] * scale - l_lds[cur][qrow]);
+                      : __expf(acc_s[r] * scale - l_lds[qrow]);
         }
         bf16x8 bp[2];
         pairswap_frags(pv, half, bp);
@@ -418,19 +404,12 @@ __global__ __launch_bounds__(512) void fa_bwd_dv_v3_kernel(
 #pragma unroll
           for (int ks2 = 0; ks2 < 2; ++ks2) {
             bf16x8 adoT =
-                trT_frag(do_lds[cur], st * 32 + ks2 * 16, n, lane, half);
+                trT_frag(do_lds, st * 32 + ks2 * 16, n, lane, half);
             acc_dv[n] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 adoT, bp[ks2], acc_dv[n], 0, 0, 0);
           }
         }
       }
-      }
-      __syncthreads();
-      if (wn < n_work) {
-        write_qdo(cur ^ 1);
-        __syncthreads();
-      }
-      cur ^= 1;
     }
     // flush dV (fp32 atomics: chunks and — at CH=1 — nothing else touches
     // these elements, but atomics keep the chunked path correct)
@@ -473,17 +452,13 @@ __global__ __launch_bounds__(512) void fa_bwd_dk_v3_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) acc_dk[n][r] = 0.f;
 
-    int cur = 0;
-    if (chunk < n_work) {
-      issue_qdo(g * rep + chunk / n_qt, qt_min + chunk % n_qt);
-      write_qdo(0);
-      __syncthreads();
-    }
     for (int w = chunk; w < n_work; w += CH) {
+      const int h = g * rep + w / n_qt;
       const int qt = qt_min + w % n_qt;
-      const int wn = w + CH;
-      if (wn < n_work) issue_qdo(g * rep + wn / n_qt, qt_min + wn % n_qt);
-      if (qt * FB_T + FB_T - 1 >= wave_kv_min) {
+      __syncthreads();
+      stage_qdo(h, qt);
+      __syncthreads();
+      if (qt * FB_T + FB_T - 1 < wave_kv_min) continue;
 #pragma unroll
       for (int st = 0; st < 2; ++st) {
         const int q0 = qt * FB_T + st * 32;
@@ -499,11 +474,11 @@ __global__ __launch_bounds__(512) void fa_bwd_dk_v3_kernel(
           const int row = st * 32 + kcol;
           const int bc = (ks * 16 + half * 8) * 2;
           bf16x8 aq = *reinterpret_cast<const bf16x8*>(
-              q_lds[cur] + row * 256 + swzb(row, bc));
+              q_lds + row * 256 + swzb(row, bc));
           acc_s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(aq, bk[ks], acc_s,
                                                           0, 0, 0);
           bf16x8 ado = *reinterpret_cast<const bf16x8*>(
-              do_lds[cur] + row * 256 + swzb(row, bc));
+              do_lds + row * 256 + swzb(row, bc));
           acc_dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ado, bv[ks],
                                                            acc_dp, 0, 0, 0);
         }
@@ -515,8 +490,8 @@ __global__ __launch_bounds__(512) void fa_bwd_dk_v3_kernel(
           const int q_glob = qt * FB_T + qrow;
           const float p = (kv_own > q_glob)
                               ? 0.f
-                              : __expf(acc_s[r] * scale - l_lds[cur][qrow]);
-          dsv[r] = p * (acc_dp[r] - d_lds[cur][qrow]) * scale;
+                              : __expf(acc_s[r] * scale - l_lds[qrow]);
+          dsv[r] = p * (acc_dp[r] - d_lds[qrow]) * scale;
         }
         bf16x8 bds[2];
         pairswap_frags(dsv, half, bds);
@@ -525,20 +500,12 @@ __global__ __launch_bounds__(512) void fa_bwd_dk_v3_kernel(
         for (int n = 0; n < 4; ++n) {
 #pragma unroll
           for (int ks2 = 0; ks2 < 2; ++ks2) {
-            bf16x8 aqT =
-                trT_frag(q_lds[cur], st * 32 + ks2 * 16, n, lane, half);
+            bf16x8 aqT = trT_frag(q_lds, st * 32 + ks2 * 16, n, lane, half);
             acc_dk[n] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 aqT, bds[ks2], acc_dk[n], 0, 0, 0);
           }
         }
       }
-      }
-      __syncthreads();
-      if (wn < n_work) {
-        write_qdo(cur ^ 1);
-        __syncthreads();
-      }
-      cur ^= 1;
     }
     float* dk_base = dk32 + ((long long)bg * S + kv_own) * FB_D;
 #pragma unroll
