@@ -262,6 +262,7 @@ class DistributedJobManager(JobManager):
         self._next_node_id = 1000  # relaunched nodes get fresh ids
         self._pending_since: dict = {}  # node id -> first PENDING sighting
         self._group_failures: dict = {}  # group id -> [fail timestamps]
+        self._auto_scaler = None
 
     def start(self):
         super().start()
@@ -276,6 +277,26 @@ class DistributedJobManager(JobManager):
         )
         t.start()
         self._threads.append(t)
+        # periodic optimizer->scaler loop (ref: dist_job_manager starting
+        # AllreduceTrainingAutoScaler), opt-in via Context.auto_worker_enabled
+        if self.scaler is not None and self._config.auto_worker_enabled:
+            from dlrover_amd.master.auto_scale import (
+                JobAutoScaler,
+                LocalResourceOptimizer,
+            )
+
+            perf = getattr(self, "perf_monitor", None)
+            opt = LocalResourceOptimizer(
+                perf, ctx=self.ctx, min_nodes=self.min_nodes,
+                max_nodes=max(self.min_nodes,
+                              getattr(self, "max_nodes", self.min_nodes)),
+            )
+            self._auto_scaler = JobAutoScaler(opt, self.scaler).start()
+
+    def stop(self):
+        if self._auto_scaler is not None:
+            self._auto_scaler.stop()
+        super().stop()
 
     # -- early stop on unschedulable nodes (ref: dist_job_manager.py:386
     # -- _early_stop_part_of_pending) -----------------------------------------

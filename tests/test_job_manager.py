@@ -255,3 +255,40 @@ def test_oom_relaunch_grows_memory():
     ))
     assert FakeScaler.launched is not None
     assert FakeScaler.launched.config_resource.memory_mb == 131072
+
+
+def test_auto_scaler_started_when_enabled():
+    """DistributedJobManager starts the optimizer->scaler loop when
+    Context.auto_worker_enabled (ref: dist_job_manager starting
+    AllreduceTrainingAutoScaler) and tears it down on stop."""
+    from dlrover_amd.common.global_context import Context
+    from dlrover_amd.master.node.job_manager import DistributedJobManager
+
+    class FakeScaler:
+        def launch_node(self, n):
+            pass
+
+        def remove_node(self, n):
+            pass
+
+        def scale_to(self, count, nodes):
+            pass
+
+    cfg = Context.singleton_instance()
+    old = cfg.auto_worker_enabled
+    try:
+        cfg.auto_worker_enabled = True
+        mgr = DistributedJobManager(job_context=_fresh_ctx(), scaler=FakeScaler())
+        mgr.start()
+        assert mgr._auto_scaler is not None
+        plan = mgr._auto_scaler.execute_once()  # empty ctx -> grow/hold/no-op
+        assert plan is None or plan.node_count >= 0
+        mgr.stop()
+
+        cfg.auto_worker_enabled = False
+        mgr2 = DistributedJobManager(job_context=_fresh_ctx(), scaler=FakeScaler())
+        mgr2.start()
+        assert mgr2._auto_scaler is None
+        mgr2.stop()
+    finally:
+        cfg.auto_worker_enabled = old
