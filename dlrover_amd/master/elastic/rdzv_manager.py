@@ -159,6 +159,15 @@ class RendezvousManager:
             self._rdzv_nodes,
             dropped,
         )
+        # structured lifecycle event (ref: DLRoverMasterEvent rendezvous
+        # spans) — feeds /api/events and offline goodput postmortems
+        from dlrover_amd.common.events import master_events
+
+        master_events().instant(
+            "rdzv_complete",
+            {"rdzv": self.name, "round": self._rdzv_round,
+             "world": sorted(self._rdzv_nodes), "dropped": dropped},
+        )
         return True
 
     def get_comm_world(self, node_rank: int) -> Tuple[int, int, Dict[int, int]]:

@@ -137,6 +137,12 @@ class JobMaster:
         self.job_manager.start()
         self.diagnosis_manager.start()
         self.ctx.job_stage = JobStage.RUNNING
+        from dlrover_amd.common.events import master_events
+
+        master_events().instant(
+            "master_start",
+            {"port": self._port, "service": self._service_type},
+        )
         logger.info(
             "job master ready on port %s (%s)", self._port, self._service_type
         )
@@ -163,6 +169,12 @@ class JobMaster:
 
     def stop(self):
         self.ctx.job_stage = JobStage.STOPPED
+        from dlrover_amd.common.events import master_events
+
+        master_events().instant(
+            "master_exit",
+            {"reason": self.ctx.exit_reason, "code": self.ctx.exit_code},
+        )
         self.diagnosis_manager.stop()
         self.job_manager.stop()
         if self._server is not None:

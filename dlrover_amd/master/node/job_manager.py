@@ -166,6 +166,14 @@ class JobManager:
     # -- failure handling / relaunch ladder (ref: _should_relaunch :1083) ---------------
 
     def _handle_node_failure(self, node: Node, reason: str = ""):
+        from dlrover_amd.common.events import master_events
+
+        master_events().instant(
+            "node_fail",
+            {"node": node.id, "rank": node.rank_index,
+             "exit_reason": node.exit_reason, "reason": reason,
+             "relaunches": node.relaunch_count},
+        )
         for mgr in self.rdzv_managers.values():
             mgr.remove_alive_node(node.id)
         for cb in self.node_failure_callbacks:
@@ -400,6 +408,14 @@ class DistributedJobManager(JobManager):
         self._next_node_id += 1
         self.ctx.update_node(replacement)
         logger.info("relaunching %s as %s (%s)", node, replacement, reason)
+        from dlrover_amd.common.events import master_events
+
+        master_events().instant(
+            "node_relaunch",
+            {"node": node.id, "replacement": replacement.id,
+             "rank": node.rank_index, "reason": reason,
+             "memory_mb": replacement.config_resource.memory_mb},
+        )
         self.scaler.launch_node(replacement)
         self.scaler.remove_node(node)
         return replacement

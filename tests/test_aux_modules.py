@@ -222,3 +222,43 @@ def test_brain_service_client_roundtrip():
         assert plan == {"node_count": 4, "comment": "brain:hold"}
     finally:
         svc.stop()
+
+
+def test_master_lifecycle_events_emitted(tmp_path, monkeypatch):
+    """Master-side structured events (rdzv_complete / node_fail /
+    node_relaunch / master_start) land in the events JSONL the dashboard
+    tails (ref: training_event predefined DLRoverMasterEvent)."""
+    import json
+    import time
+
+    from dlrover_amd.common import events as ev
+
+    monkeypatch.setenv("DLROVER_EVENT_DIR", str(tmp_path))
+    monkeypatch.setattr(ev.AsyncExporter, "_instance", None)
+
+    from dlrover_amd.common.constants import NodeStatus, NodeType
+    from dlrover_amd.common.node import Node
+    from dlrover_amd.master.node.job_context import JobContext
+    from dlrover_amd.master.node.job_manager import LocalJobManager
+
+    mgr = LocalJobManager(job_context=JobContext())
+    n = Node(NodeType.WORKER, 3, rank_index=3)
+    n.update_status(NodeStatus.RUNNING)
+    mgr.ctx.update_node(n)
+    mgr._handle_node_failure(n, "test failure")
+
+    ev.AsyncExporter.get().close()
+    monkeypatch.setattr(ev.AsyncExporter, "_instance", None)
+
+    recs = []
+    deadline = time.time() + 5
+    while time.time() < deadline and not recs:
+        for p in tmp_path.glob("events_*.jsonl"):
+            recs += [json.loads(line) for line in p.read_text().splitlines()]
+        if not recs:
+            time.sleep(0.1)
+    names = {r["name"] for r in recs}
+    assert "node_fail" in names, recs
+    fail = next(r for r in recs if r["name"] == "node_fail")
+    assert fail["content"]["node"] == 3
+    assert fail["target"] == "dlrover-master"
