@@ -262,3 +262,18 @@ def test_master_lifecycle_events_emitted(tmp_path, monkeypatch):
     fail = next(r for r in recs if r["name"] == "node_fail")
     assert fail["content"]["node"] == 3
     assert fail["target"] == "dlrover-master"
+
+
+def test_hipmem_register_degrades_without_gpu():
+    """hipHostRegister wrapper: bool result, no raise, safe unregister on a
+    box with the HIP runtime but no device (rc=100 hipErrorNoDevice here) —
+    the checkpoint drain must fall back to pageable copies, not crash."""
+    import numpy as np
+
+    from dlrover_amd.utils import hipmem
+
+    a = np.zeros(4096, dtype=np.uint8)
+    ok = hipmem.host_register(a.ctypes.data, a.nbytes)
+    assert isinstance(ok, bool)
+    hipmem.host_unregister(a.ctypes.data)  # must never raise
+    assert hipmem.host_register(a.ctypes.data, 0) is False  # empty region
