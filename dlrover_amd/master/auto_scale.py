@@ -38,13 +38,45 @@ class LocalResourceOptimizer:
     allreduce/worker case that matters for GPU training)."""
 
     def __init__(self, perf: PerfMonitor, ctx: Optional[JobContext] = None,
-                 min_nodes: int = 1, max_nodes: int = 1):
+                 min_nodes: int = 1, max_nodes: int = 1, job_name: str = ""):
+        import os
+
         self.perf = perf
         self.ctx = ctx or JobContext.singleton_instance()
         self.min_nodes = min_nodes
         self.max_nodes = max_nodes
+        self.job_name = job_name or os.getenv("ELASTIC_JOB_NAME", "job")
+        # optimizeMode=cluster: a Brain endpoint outranks local heuristics
+        # (ref: brain_optimizer.py wrapper; falls back when unreachable)
+        self._brain = None
+        if os.getenv("DLROVER_BRAIN_ADDR", ""):
+            from dlrover_amd.brain_client import BrainClient
+
+            client = BrainClient()
+            self._brain = client if client.available else None
+
+    def _brain_plan(self) -> Optional[ResourcePlan]:
+        if self._brain is None:
+            return None
+        alive = [n for n in self.ctx.job_nodes().values() if n.is_alive()]
+        speed = self.perf.running_speed() if self.perf is not None else 0.0
+        self._brain.report_metrics(self.job_name, {"steps_per_sec": speed})
+        plan = self._brain.get_optimization_plan(
+            self.job_name, "running",
+            {"current_nodes": len(alive) or 1, "max_nodes": self.max_nodes},
+        )
+        if not plan or not plan.get("node_count"):
+            return None
+        count = max(self.min_nodes, min(int(plan["node_count"]), self.max_nodes))
+        if count == len(alive):
+            return None  # hold
+        return ResourcePlan(node_count=count,
+                            comment=plan.get("comment", "brain"))
 
     def generate_plan(self) -> Optional[ResourcePlan]:
+        brain = self._brain_plan()
+        if brain is not None:
+            return brain
         nodes = self.ctx.job_nodes()
         alive = [n for n in nodes.values() if n.is_alive()]
         pending = [n for n in alive if n.status == NodeStatus.PENDING]

@@ -168,3 +168,33 @@ def test_dashboard_events_tail(tmp_path, monkeypatch):
         assert len(evs) == 2 and evs[-1]["phase"] == "end"
     finally:
         dash.stop()
+
+
+def test_optimizer_consults_brain_when_configured(ctx, monkeypatch):
+    """optimizeMode=cluster: with DLROVER_BRAIN_ADDR set, the optimizer asks
+    the LIVE Brain service for the plan (and reports throughput) before any
+    local heuristic; the plan is clamped to [min_nodes, max_nodes]."""
+    from dlrover_amd.common.constants import NodeStatus, NodeType
+    from dlrover_amd.common.node import Node
+    from dlrover_amd.master.auto_scale import LocalResourceOptimizer
+    from dlrover_amd.master.brain_service import BrainService
+    from dlrover_amd.master.perf_monitor import PerfMonitor
+    import time
+
+    svc = BrainService(port=0, host="127.0.0.1").start()
+    try:
+        monkeypatch.setenv("DLROVER_BRAIN_ADDR", f"127.0.0.1:{svc.port}")
+        for i in range(2):
+            n = Node(NodeType.WORKER, i)
+            n.update_status(NodeStatus.RUNNING)
+            ctx.update_node(n)
+        perf = PerfMonitor()
+        perf.report_global_step(10, time.time() - 1)
+        perf.report_global_step(20, time.time())
+        opt = LocalResourceOptimizer(perf, ctx=ctx, min_nodes=1, max_nodes=8,
+                                     job_name="jbrain")
+        plan = opt.generate_plan()
+        assert plan is not None and plan.comment == "brain:grow"
+        assert plan.node_count == 4  # 2 alive -> brain doubles
+    finally:
+        svc.stop()
