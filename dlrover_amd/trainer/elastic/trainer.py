@@ -110,6 +110,20 @@ class ElasticTrainer:
         gs = self.gradient_state
         return gs.micro_step % gs.accum_steps == 0
 
+    def prepare(self, optimizer, lr_scheduler=None):
+        """Reference-API convenience (ref: trainer.py:229 prepare): returns
+        wrappers whose ``step()`` fires only on accumulation boundaries, so
+        a loop written for the reference runs unchanged:
+
+            optimizer, scheduler = elastic_trainer.prepare(opt, sched)
+            with elastic_trainer.step():
+                loss.backward(); optimizer.step(); optimizer.zero_grad()
+        """
+        opt = _BoundaryStepper(self, optimizer, zero_on_boundary_only=True)
+        if lr_scheduler is None:
+            return opt
+        return opt, _BoundaryStepper(self, lr_scheduler)
+
     def _report_step(self):
         try:
             rank = dist.get_rank() if dist.is_initialized() else 0
@@ -124,3 +138,27 @@ class _GradState:
     def __init__(self, accum_steps: int):
         self.accum_steps = accum_steps
         self.micro_step = 0
+
+
+class _BoundaryStepper:
+    """Proxy for an optimizer/scheduler whose step() only fires on the
+    trainer's accumulation boundary; everything else passes through."""
+
+    def __init__(self, trainer: ElasticTrainer, inner, zero_on_boundary_only=False):
+        self._trainer = trainer
+        self._inner = inner
+        self._zero_gate = zero_on_boundary_only
+
+    def step(self, *a, **kw):
+        if self._trainer.step_boundary:
+            return self._inner.step(*a, **kw)
+        return None
+
+    def zero_grad(self, *a, **kw):
+        # off-boundary zeroing would drop accumulated gradients
+        if not self._zero_gate or self._trainer.step_boundary:
+            return self._inner.zero_grad(*a, **kw)
+        return None
+
+    def __getattr__(self, name):
+        return getattr(self._inner, name)

@@ -71,3 +71,46 @@ def test_sampler_reshard_after_scale():
     # full coverage of the remaining samples; at most num_replicas-1 dupes
     assert set(out) == set(range(8, 24))
     assert len(out) - len(set(out)) < 3
+
+
+def test_prepare_gates_optimizer_on_accum_boundary(monkeypatch):
+    """Reference-API loop (prepare + step context): optimizer.step() and
+    zero_grad() fire only on accumulation boundaries; gradients accumulate
+    across the suppressed micro-steps."""
+    import torch
+
+    from dlrover_amd.trainer.elastic.trainer import ElasticTrainer
+
+    monkeypatch.setenv("DLROVER_MAX_WORKERS", "4")  # world 1 -> accum 4
+    model = torch.nn.Linear(4, 4)
+    trainer = ElasticTrainer(model)
+    assert trainer.gradient_state.accum_steps == 4
+
+    base_opt = torch.optim.SGD(model.parameters(), lr=0.1)
+    sched = torch.optim.lr_scheduler.StepLR(base_opt, step_size=1, gamma=0.5)
+    opt, sched = trainer.prepare(base_opt, sched)
+    assert opt.param_groups is base_opt.param_groups  # passthrough attrs
+
+    steps_fired = 0
+    for micro in range(8):
+        with trainer.step() as sync:
+            loss = model(torch.randn(2, 4)).sum()
+            loss.backward()
+            before = [p.clone() for p in model.parameters()]
+            opt.step()
+            opt.zero_grad()
+            changed = any(
+                not torch.equal(b, p)
+                for b, p in zip(before, model.parameters())
+            )
+            assert changed == sync
+            if sync:
+                steps_fired += 1
+                sched.step()
+            else:
+                # grads must survive the gated zero_grad
+                assert any(p.grad is not None and p.grad.abs().sum() > 0
+                           for p in model.parameters())
+    assert steps_fired == 2
+    assert trainer.global_step == 2
+    assert base_opt.param_groups[0]["lr"] == 0.1 * 0.5 ** 2
