@@ -92,9 +92,27 @@ def parse_args(argv: Optional[List[str]] = None):
     p.add_argument("--run-id", "--run_id", default="",
                    help="rendezvous run id (defaults to the job name)")
     p.add_argument("--service-type", default=CommServiceType.TCP)
+    # torchrun flags accepted for drop-in compatibility; rendezvous is the
+    # dlrover master here, so the c10d knobs are intentionally ignored
+    for kebab in ("rdzv-backend", "rdzv-id", "start-method", "local-addr",
+                  "master-addr", "master-port"):
+        p.add_argument(f"--{kebab}", f"--{kebab.replace('-', '_')}",
+                       default=None, help=argparse.SUPPRESS)
+    p.add_argument("--no-python", "--no_python", action="store_true",
+                   help="run the script directly (torchrun compat; implied "
+                        "for non-.py scripts)")
     p.add_argument("training_script", help="training program (.py or executable)")
     p.add_argument("training_script_args", nargs=argparse.REMAINDER)
-    return p.parse_args(argv)
+    args = p.parse_args(argv)
+    ignored = [f for f in ("rdzv_backend", "rdzv_id", "start_method",
+                           "local_addr", "master_addr", "master_port")
+               if getattr(args, f, None)]
+    if ignored:
+        logger.info(
+            "torchrun-compat flags ignored (the dlrover master handles "
+            "rendezvous): %s", ", ".join(ignored),
+        )
+    return args
 
 
 def _launch_local_master(service_type: str) -> Tuple[subprocess.Popen, str]:
@@ -231,7 +249,7 @@ def run(args) -> int:
     config.auto_configure()
     script = args.training_script
     script_args = list(args.training_script_args)
-    if script.endswith(".py"):
+    if script.endswith(".py") and not getattr(args, "no_python", False):
         entrypoint: object = sys.executable
         script_args = ["-u", script] + script_args
     else:

@@ -277,3 +277,21 @@ def test_hipmem_register_degrades_without_gpu():
     assert isinstance(ok, bool)
     hipmem.host_unregister(a.ctypes.data)  # must never raise
     assert hipmem.host_register(a.ctypes.data, 0) is False  # empty region
+
+
+def test_dlrover_run_accepts_torchrun_flags():
+    """A torchrun command line runs unchanged: c10d rendezvous knobs are
+    accepted (and ignored — the dlrover master owns rendezvous)."""
+    from dlrover_amd.trainer.elastic_run import parse_args
+
+    args = parse_args([
+        "--nnodes", "1", "--nproc_per_node", "2",
+        "--rdzv_backend", "c10d", "--rdzv_id", "job42",
+        "--master_addr", "10.0.0.1", "--master_port", "29500",
+        "--start_method", "spawn", "--no-python",
+        "train.sh", "--flag",
+    ])
+    assert args.rdzv_backend == "c10d" and args.master_addr == "10.0.0.1"
+    assert args.no_python
+    assert args.training_script == "train.sh"
+    assert args.training_script_args == ["--flag"]
