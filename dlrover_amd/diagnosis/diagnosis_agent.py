@@ -94,6 +94,27 @@ def classify_error(error_text: str) -> str:
     return RESTART_WORKER
 
 
+def _load_failover_extension():
+    """User-pluggable failover strategy (ref: torch/dynamic_failover.py
+    DynamicAgentFailoverExtension): DLROVER_FAILOVER_EXTENSION="pkg.mod:Cls"
+    names a class whose get_user_failover_strategy(failure_text,
+    restart_count) returns one of "restart"/"relaunch"/"abort" or None/"" to
+    defer to the built-in ladder."""
+    import importlib
+    import os
+
+    spec = os.getenv("DLROVER_FAILOVER_EXTENSION", "")
+    if not spec or ":" not in spec:
+        return None
+    mod_name, _, cls_name = spec.partition(":")
+    try:
+        cls = getattr(importlib.import_module(mod_name), cls_name)
+        return cls()
+    except Exception:  # noqa: BLE001 — a broken extension must not break failover
+        logger.exception("failover extension %s failed to load", spec)
+        return None
+
+
 class WorkerDiagnosisAgent:
     """Used by the elastic agent on worker failure (ref: diagnosis_agent
     .diagnose_training_failure :153)."""
@@ -101,12 +122,23 @@ class WorkerDiagnosisAgent:
     def __init__(self, client=None, log_file: str = ""):
         self.client = client
         self.collector = TrainingLogCollector(log_file)
+        self.extension = _load_failover_extension()
 
     def diagnose_training_failure(
         self, failure_text: str, restart_count: int, max_restarts: int
     ) -> str:
         log = self.collector.collect()
         combined = f"{failure_text}\n{log.text()}"
+        if self.extension is not None:
+            try:
+                user = self.extension.get_user_failover_strategy(
+                    combined, restart_count
+                )
+                if user in (RESTART_WORKER, RELAUNCH_NODE, ABORT_JOB):
+                    logger.info("failover extension verdict: %s", user)
+                    return user
+            except Exception:  # noqa: BLE001
+                logger.exception("failover extension errored — using ladder")
         verdict = classify_error(combined)
         if verdict == RESTART_WORKER and restart_count >= max_restarts:
             verdict = RELAUNCH_NODE

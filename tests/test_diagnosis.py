@@ -157,3 +157,39 @@ def test_precheck_operator_chain(monkeypatch):
 
     chain3 = PreCheckChain(["always_fail"])
     assert chain3.evaluate(m)[0] == FAIL
+
+
+def test_failover_extension_overrides_ladder(tmp_path, monkeypatch):
+    """DLROVER_FAILOVER_EXTENSION names a user class whose strategy overrides
+    the built-in ladder (ref: torch/dynamic_failover.py); invalid verdicts
+    and load failures fall back to the ladder."""
+    import sys
+
+    ext = tmp_path / "my_failover.py"
+    ext.write_text(
+        "class AlwaysRelaunch:\n"
+        "    def get_user_failover_strategy(self, text, restarts):\n"
+        "        return 'relaunch' if 'special' in text else None\n"
+    )
+    monkeypatch.syspath_prepend(str(tmp_path))
+    monkeypatch.setenv("DLROVER_FAILOVER_EXTENSION", "my_failover:AlwaysRelaunch")
+    sys.modules.pop("my_failover", None)
+
+    from dlrover_amd.diagnosis.diagnosis_agent import (
+        RELAUNCH_NODE,
+        RESTART_WORKER,
+        WorkerDiagnosisAgent,
+    )
+
+    agent = WorkerDiagnosisAgent()
+    assert agent.extension is not None
+    # extension claims this one
+    assert agent.diagnose_training_failure("a special failure", 0, 3) == RELAUNCH_NODE
+    # extension defers (returns None) -> ladder says restart
+    assert agent.diagnose_training_failure("ordinary traceback", 0, 3) == RESTART_WORKER
+
+    # broken spec -> no extension, ladder still works
+    monkeypatch.setenv("DLROVER_FAILOVER_EXTENSION", "nope.not.there:X")
+    agent2 = WorkerDiagnosisAgent()
+    assert agent2.extension is None
+    assert agent2.diagnose_training_failure("ordinary", 0, 3) == RESTART_WORKER
