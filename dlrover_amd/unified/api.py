@@ -84,6 +84,11 @@ class _RoleBuilder:
     def role(self, name: str) -> "_RoleBuilder":
         return self._parent.role(name)
 
+    def __getattr__(self, name):
+        # any other chain method (train, actor, rollout, ...) belongs to the
+        # job builder: delegate so role chains compose naturally
+        return getattr(self._parent, name)
+
     def with_collocation(self, *names: str) -> "DLJobBuilder":
         return self._parent.with_collocation(*names)
 
@@ -124,4 +129,60 @@ class DLJobBuilder:
         for r in self._job.roles.values():
             if r.entry_func is None and not r.entry_module:
                 raise ValueError(f"role {r.name} has no entrypoint")
+        self._validate(self._job)
         return self._job
+
+    def _validate(self, job: DLJob) -> None:  # extension hook
+        pass
+
+
+class RLJobBuilder(DLJobBuilder):
+    """Reinforcement-learning job builder (ref: api/builder/rl.py:149):
+    the RL role vocabulary (trainer/actor/reference/reward/critic/rollout)
+    with per-role helpers and build-time validation — 'actor' is mandatory,
+    unknown roles are rejected.
+
+        job = (RLJobBuilder()
+               .trainer().run(train_fn)
+               .actor(8).run(actor_fn)
+               .rollout(4).run(rollout_fn)
+               .with_collocation("actor", "rollout")
+               .build())
+    """
+
+    TRAINER_ROLE = "trainer"
+    ACTOR_ROLE = "actor"
+    REF_ROLE = "reference"
+    REW_ROLE = "reward"
+    CRITIC_ROLE = "critic"
+    ROLLOUT_ROLE = "rollout"
+    ROLES = [TRAINER_ROLE, ACTOR_ROLE, REF_ROLE, REW_ROLE, CRITIC_ROLE,
+             ROLLOUT_ROLE]
+
+    def trainer(self, total: int = 1) -> _RoleBuilder:
+        return self.role(self.TRAINER_ROLE).total(total)
+
+    def actor(self, total: int = 1) -> _RoleBuilder:
+        return self.role(self.ACTOR_ROLE).total(total)
+
+    def reference(self, total: int = 1) -> _RoleBuilder:
+        return self.role(self.REF_ROLE).total(total)
+
+    def reward(self, total: int = 1) -> _RoleBuilder:
+        return self.role(self.REW_ROLE).total(total)
+
+    def critic(self, total: int = 1) -> _RoleBuilder:
+        return self.role(self.CRITIC_ROLE).total(total)
+
+    def rollout(self, total: int = 1) -> _RoleBuilder:
+        return self.role(self.ROLLOUT_ROLE).total(total)
+
+    def _validate(self, job: DLJob) -> None:
+        if self.ACTOR_ROLE not in job.roles:
+            raise ValueError("'actor' must be configured for an RL job")
+        for role in job.roles:
+            if role not in self.ROLES:
+                raise ValueError(
+                    f"invalid role {role!r} for an RL job; supported: "
+                    f"{self.ROLES}"
+                )

@@ -165,3 +165,38 @@ def test_prime_master_self_recovery(tmp_path):
     m = PrimeMaster(job, state_path=str(state))
     budgets = {v.name: v.restarts for v in m.graph.vertices}
     assert budgets == {"train-0": 2, "train-1": 1}
+
+
+def _noop():
+    pass
+
+
+def test_rl_builder_roles_and_validation(tmp_path):
+    """RLJobBuilder (ref: api/builder/rl.py): role vocabulary + actor
+    mandatory + unknown roles rejected; a valid RL graph runs end-to-end on
+    the local backend."""
+    import pytest
+
+    from dlrover_amd.unified.api import RLJobBuilder
+
+    with pytest.raises(ValueError, match="'actor' must be configured"):
+        RLJobBuilder().trainer().run(_noop).build()
+
+    with pytest.raises(ValueError, match="invalid role"):
+        b = RLJobBuilder()
+        b.actor().run(_noop)
+        b.role("weird").run(_noop)
+        b.build()
+
+    job = (
+        RLJobBuilder("rl-smoke")
+        .trainer().run(_noop)
+        .actor(2).run(_noop)
+        .rollout(1).run(_noop)
+        .with_collocation("actor", "rollout")
+        .build()
+    )
+    assert set(job.roles) == {"trainer", "actor", "rollout"}
+    assert job.roles["actor"].total == 2
+    master = job.submit(blocking=True)
+    assert master.status == "SUCCEEDED", master.status
