@@ -143,6 +143,57 @@ class ScalePlanWatcher:
             yield plan
 
 
+class ElasticJobWatcher:
+    """Watches THIS job's ElasticJob CR for spec changes
+    (ref: K8sElasticJobWatcher, k8s_watcher.py:450): a replica-count edit
+    becomes a ResourcePlan, ``spec.suspend: true`` yields a stop signal.
+    Yields ("scale", ResourcePlan) / ("suspend", None) / ("resume", None).
+    """
+
+    def __init__(self, job_name: str, namespace: str = "default", source=None):
+        self.job_name = job_name
+        self.namespace = namespace
+        self._source = source or FakeEventSource()
+        self._last_replicas: Optional[int] = None
+        self._suspended = False
+
+    def watch(self):
+        from dlrover_amd.master.auto_scale import ResourcePlan
+
+        for event_type, crd in self._source.stream():
+            if not crd or crd.get("kind") != "ElasticJob":
+                continue
+            if event_type not in ("ADDED", "MODIFIED"):
+                continue
+            meta = crd.get("metadata", {})
+            if meta.get("name") not in (None, self.job_name):
+                continue
+            spec = crd.get("spec", {})
+            suspended = bool(spec.get("suspend", False))
+            if suspended != self._suspended:
+                self._suspended = suspended
+                yield ("suspend" if suspended else "resume", None)
+            specs = spec.get("replicaSpecs", {})
+            worker = specs.get("worker", specs.get("Worker", {}))
+            replicas = worker.get("replicas")
+            if replicas is None:
+                continue
+            replicas = int(replicas)
+            if self._last_replicas is None:
+                self._last_replicas = replicas  # initial spec, not a change
+                continue
+            if replicas != self._last_replicas:
+                self._last_replicas = replicas
+                yield (
+                    "scale",
+                    ResourcePlan(
+                        node_count=replicas,
+                        comment=f"elasticjob/{meta.get('name', self.job_name)}"
+                                " replicas edit",
+                    ),
+                )
+
+
 class PodWatcher:
     """Yields NodeEvents to DistributedJobManager._watch_events."""
 

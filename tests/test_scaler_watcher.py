@@ -226,3 +226,37 @@ def test_service_per_pod():
         assert svc["spec"]["clusterIP"] == "None"
     finally:
         sc.stop()
+
+
+def test_elasticjob_watcher_replicas_and_suspend():
+    """ElasticJob CR edits drive the job (ref: K8sElasticJobWatcher):
+    replica change -> scale plan; suspend toggles -> suspend/resume
+    signals; the initial spec is not a change."""
+    from dlrover_amd.master.watcher.k8s_watcher import (
+        ElasticJobWatcher,
+        FakeEventSource,
+    )
+
+    def cr(replicas, suspend=False):
+        return {
+            "kind": "ElasticJob",
+            "metadata": {"name": "jobx"},
+            "spec": {
+                "suspend": suspend,
+                "replicaSpecs": {"worker": {"replicas": replicas}},
+            },
+        }
+
+    src = FakeEventSource()
+    w = ElasticJobWatcher("jobx", source=src)
+    src.push("ADDED", cr(4))            # initial spec: no event
+    src.push("MODIFIED", cr(8))         # scale up
+    src.push("MODIFIED", cr(8, True))   # suspend
+    src.push("MODIFIED", cr(2, False))  # resume + scale down
+    src.push("MODIFIED", {"kind": "ElasticJob",
+                          "metadata": {"name": "other"},
+                          "spec": {"replicaSpecs": {"worker": {"replicas": 1}}}})
+    events = list(w.watch())
+    kinds = [(k, p.node_count if p else None) for k, p in events]
+    assert kinds == [("scale", 8), ("suspend", None), ("resume", None),
+                     ("scale", 2)], kinds
