@@ -262,10 +262,12 @@ class DistributedJobManager(JobManager):
     """Cluster mode: relaunch via the platform scaler and watch platform
     events (ref: dist_job_manager.py)."""
 
-    def __init__(self, scaler=None, watcher=None, min_nodes: int = 1, **kw):
+    def __init__(self, scaler=None, watcher=None, min_nodes: int = 1,
+                 job_watcher=None, **kw):
         super().__init__(**kw)
         self.scaler = scaler
         self.watcher = watcher
+        self.job_watcher = job_watcher  # ElasticJob CR spec edits
         self.min_nodes = min_nodes
         self._next_node_id = 1000  # relaunched nodes get fresh ids
         self._pending_since: dict = {}  # node id -> first PENDING sighting
@@ -277,6 +279,12 @@ class DistributedJobManager(JobManager):
         if self.watcher is not None:
             t = threading.Thread(
                 target=self._watch_events, name="node-watcher", daemon=True
+            )
+            t.start()
+            self._threads.append(t)
+        if self.job_watcher is not None:
+            t = threading.Thread(
+                target=self._watch_job_spec, name="job-watcher", daemon=True
             )
             t.start()
             self._threads.append(t)
@@ -368,6 +376,41 @@ class DistributedJobManager(JobManager):
                     self.on_node_event(event)
             except Exception:  # noqa: BLE001 — watch streams break routinely
                 logger.exception("node watcher stream broke; re-watching")
+                time.sleep(3)
+
+    def _watch_job_spec(self):
+        """Consume ElasticJob CR edits (ref: K8sElasticJobWatcher): replica
+        edits scale the worker group; suspend scales to 0 and resume
+        restores the pre-suspend size."""
+        pre_suspend = 0
+        while not self._stop.is_set():
+            try:
+                for kind, plan in self.job_watcher.watch():
+                    if self._stop.is_set():
+                        return
+                    nodes = list(self.ctx.job_nodes().values())
+                    alive = [n for n in nodes if n.is_alive()]
+                    if kind == "scale" and plan is not None:
+                        logger.info("ElasticJob spec: scale to %s (%s)",
+                                    plan.node_count, plan.comment)
+                        if self.scaler is not None:
+                            self.scaler.scale_to(plan.node_count, nodes)
+                    elif kind == "suspend":
+                        pre_suspend = len(alive)
+                        logger.warning(
+                            "ElasticJob suspended: releasing %s workers",
+                            pre_suspend,
+                        )
+                        if self.scaler is not None:
+                            self.scaler.scale_to(0, nodes)
+                    elif kind == "resume" and pre_suspend > 0:
+                        logger.info("ElasticJob resumed: scaling back to %s",
+                                    pre_suspend)
+                        if self.scaler is not None:
+                            self.scaler.scale_to(pre_suspend, nodes)
+                        pre_suspend = 0
+            except Exception:  # noqa: BLE001 — watch streams break routinely
+                logger.exception("job watcher stream broke; re-watching")
                 time.sleep(3)
 
     def _relaunch_node(self, node: Node, reason: str):

@@ -292,3 +292,55 @@ def test_auto_scaler_started_when_enabled():
         mgr2.stop()
     finally:
         cfg.auto_worker_enabled = old
+
+
+def test_job_watcher_drives_scaling():
+    """ElasticJob CR edits flow through the manager to the scaler: replica
+    edit scales, suspend releases workers, resume restores the size."""
+    import time
+
+    from dlrover_amd.common.constants import NodeStatus, NodeType
+    from dlrover_amd.common.node import Node
+    from dlrover_amd.master.node.job_manager import DistributedJobManager
+    from dlrover_amd.master.watcher.k8s_watcher import (
+        ElasticJobWatcher,
+        FakeEventSource,
+    )
+
+    calls = []
+
+    class FakeScaler:
+        def launch_node(self, n):
+            pass
+
+        def remove_node(self, n):
+            pass
+
+        def scale_to(self, count, nodes):
+            calls.append(count)
+
+    def cr(replicas, suspend=False):
+        return {"kind": "ElasticJob", "metadata": {"name": "jobx"},
+                "spec": {"suspend": suspend,
+                         "replicaSpecs": {"worker": {"replicas": replicas}}}}
+
+    src = FakeEventSource()
+    ctx = _fresh_ctx()
+    for i in range(2):
+        n = Node(NodeType.WORKER, i)
+        n.update_status(NodeStatus.RUNNING)
+        ctx.update_node(n)
+    mgr = DistributedJobManager(
+        job_context=ctx, scaler=FakeScaler(),
+        job_watcher=ElasticJobWatcher("jobx", source=src),
+    )
+    src.push("ADDED", cr(2))           # initial — no scale
+    src.push("MODIFIED", cr(4))        # scale to 4
+    src.push("MODIFIED", cr(4, True))  # suspend -> scale to 0
+    src.push("MODIFIED", cr(4, False))  # resume -> back to pre-suspend (2)
+    mgr.start()
+    deadline = time.time() + 10
+    while time.time() < deadline and len(calls) < 3:
+        time.sleep(0.05)
+    mgr.stop()
+    assert calls[:3] == [4, 0, 2], calls
