@@ -30,13 +30,17 @@ def run(args) -> int:
     if args.platform == PlatformType.KUBERNETES:
         from dlrover_amd.master.job_master import DistributedJobMaster
         from dlrover_amd.master.scaler.pod_scaler import PodScaler
-        from dlrover_amd.master.watcher.k8s_watcher import PodWatcher
+        from dlrover_amd.master.watcher.k8s_watcher import (
+            ElasticJobWatcher,
+            PodWatcher,
+        )
 
         scaler = PodScaler(args.job_name, args.namespace)
         watcher = PodWatcher(args.job_name, args.namespace)
+        job_watcher = ElasticJobWatcher(args.job_name, args.namespace)
         master = DistributedJobMaster(
-            scaler=scaler, watcher=watcher, port=args.port,
-            service_type=args.service_type,
+            scaler=scaler, watcher=watcher, job_watcher=job_watcher,
+            port=args.port, service_type=args.service_type,
         )
     else:
         from dlrover_amd.master.job_master import LocalJobMaster

@@ -143,6 +143,28 @@ class ScalePlanWatcher:
             yield plan
 
 
+class K8sCustomObjectSource:  # pragma: no cover - needs a cluster
+    """Watch stream over a custom-resource plural (ElasticJob / ScalePlan)."""
+
+    def __init__(self, job_name: str, namespace: str, plural: str):
+        self.job_name = job_name
+        self.namespace = namespace
+        self.plural = plural
+
+    def stream(self) -> Iterator:
+        from kubernetes import client, config, watch
+
+        config.load_incluster_config()
+        api = client.CustomObjectsApi()
+        w = watch.Watch()
+        for event in w.stream(
+            api.list_namespaced_custom_object,
+            "elastic.iml.github.io", "v1alpha1", self.namespace, self.plural,
+            timeout_seconds=3600,
+        ):
+            yield event["type"], event["object"]
+
+
 class ElasticJobWatcher:
     """Watches THIS job's ElasticJob CR for spec changes
     (ref: K8sElasticJobWatcher, k8s_watcher.py:450): a replica-count edit
@@ -153,7 +175,9 @@ class ElasticJobWatcher:
     def __init__(self, job_name: str, namespace: str = "default", source=None):
         self.job_name = job_name
         self.namespace = namespace
-        self._source = source or FakeEventSource()
+        self._source = source or K8sCustomObjectSource(
+            job_name, namespace, "elasticjobs"
+        )
         self._last_replicas: Optional[int] = None
         self._suspended = False
 
