@@ -188,10 +188,11 @@ class LocalJobMaster(JobMaster):
 class DistributedJobMaster(JobMaster):
     """Cluster master with platform scaler/watcher (ref: dist_master.py)."""
 
-    def __init__(self, scaler=None, watcher=None, job_watcher=None, **kw):
+    def __init__(self, scaler=None, watcher=None, job_watcher=None,
+                 scaleplan_watcher=None, **kw):
         ctx = JobContext.singleton_instance()
         jm = DistributedJobManager(
             scaler=scaler, watcher=watcher, job_watcher=job_watcher,
-            job_context=ctx,
+            scaleplan_watcher=scaleplan_watcher, job_context=ctx,
         )
         super().__init__(job_manager=jm, **kw)

@@ -33,13 +33,16 @@ def run(args) -> int:
         from dlrover_amd.master.watcher.k8s_watcher import (
             ElasticJobWatcher,
             PodWatcher,
+            ScalePlanWatcher,
         )
 
         scaler = PodScaler(args.job_name, args.namespace)
         watcher = PodWatcher(args.job_name, args.namespace)
         job_watcher = ElasticJobWatcher(args.job_name, args.namespace)
+        scaleplan_watcher = ScalePlanWatcher(args.job_name, args.namespace)
         master = DistributedJobMaster(
             scaler=scaler, watcher=watcher, job_watcher=job_watcher,
+            scaleplan_watcher=scaleplan_watcher,
             port=args.port, service_type=args.service_type,
         )
     else:

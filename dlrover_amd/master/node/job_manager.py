@@ -263,11 +263,12 @@ class DistributedJobManager(JobManager):
     events (ref: dist_job_manager.py)."""
 
     def __init__(self, scaler=None, watcher=None, min_nodes: int = 1,
-                 job_watcher=None, **kw):
+                 job_watcher=None, scaleplan_watcher=None, **kw):
         super().__init__(**kw)
         self.scaler = scaler
         self.watcher = watcher
         self.job_watcher = job_watcher  # ElasticJob CR spec edits
+        self.scaleplan_watcher = scaleplan_watcher  # manual ScalePlan CRs
         self.min_nodes = min_nodes
         self._next_node_id = 1000  # relaunched nodes get fresh ids
         self._pending_since: dict = {}  # node id -> first PENDING sighting
@@ -285,6 +286,13 @@ class DistributedJobManager(JobManager):
         if self.job_watcher is not None:
             t = threading.Thread(
                 target=self._watch_job_spec, name="job-watcher", daemon=True
+            )
+            t.start()
+            self._threads.append(t)
+        if self.scaleplan_watcher is not None:
+            t = threading.Thread(
+                target=self._watch_scaleplans, name="scaleplan-watcher",
+                daemon=True,
             )
             t.start()
             self._threads.append(t)
@@ -376,6 +384,25 @@ class DistributedJobManager(JobManager):
                     self.on_node_event(event)
             except Exception:  # noqa: BLE001 — watch streams break routinely
                 logger.exception("node watcher stream broke; re-watching")
+                time.sleep(3)
+
+    def _watch_scaleplans(self):
+        """Apply user-submitted manual ScalePlan CRs (ref:
+        K8sScalePlanWatcher -> JobAutoScaler execution)."""
+        while not self._stop.is_set():
+            try:
+                for plan in self.scaleplan_watcher.watch():
+                    if self._stop.is_set():
+                        return
+                    logger.info("manual ScalePlan: scale to %s (%s)",
+                                plan.node_count, plan.comment)
+                    if self.scaler is not None and plan.node_count > 0:
+                        self.scaler.scale_to(
+                            plan.node_count,
+                            list(self.ctx.job_nodes().values()),
+                        )
+            except Exception:  # noqa: BLE001 — watch streams break routinely
+                logger.exception("scaleplan watcher stream broke; re-watching")
                 time.sleep(3)
 
     def _watch_job_spec(self):

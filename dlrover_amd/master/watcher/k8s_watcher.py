@@ -105,7 +105,9 @@ class ScalePlanWatcher:
     def __init__(self, job_name: str, namespace: str = "default", source=None):
         self.job_name = job_name
         self.namespace = namespace
-        self._source = source or FakeEventSource()
+        self._source = source or K8sCustomObjectSource(
+            job_name, namespace, "scaleplans"
+        )
         self._used_uids: set = set()
 
     def watch(self):
