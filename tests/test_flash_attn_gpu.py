@@ -240,6 +240,11 @@ def test_fsdp_dtensor_ckpt_roundtrip_gpu(tmp_path, monkeypatch):
             p.to_local().add_(1.0)
         out = cp.engine.restore_into(model, opt)
         assert out is not None and out.get("step") == 3
+        # the bf16 params were OMITTED from the snapshot (derived-ckpt,
+        # default on) — this equality only holds if the device rederive
+        # from the restored fp32 master actually ran
+        assert out.get("_derived"), "snapshot did not use derived-ckpt"
+        assert name in out["_derived"]
         torch.testing.assert_close(p.to_local(), want)
         # training continues after restore
         model(ids, ids.clone()).backward()
