@@ -129,6 +129,18 @@ class MasterClient:
         resp = self.get(comm.WaitingNodeNumRequest(rdzv_name=rdzv_name))
         return resp.waiting_num
 
+    def block_rendezvous(
+        self, node_rank: int, blocked: bool,
+        rdzv_name: str = RendezvousName.TRAINING,
+    ):
+        """Hold/release the pending round while this node persists UCP
+        shards (ref: UcpRdzvManager blockable rendezvous)."""
+        self.report(
+            comm.RdzvBlockRequest(
+                node_rank=node_rank, blocked=blocked, rdzv_name=rdzv_name
+            )
+        )
+
     def report_rdzv_params(
         self, min_nodes: int, max_nodes: int, waiting_timeout: float, node_unit: int
     ):

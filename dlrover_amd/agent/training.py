@@ -406,7 +406,20 @@ class ElasticTrainingAgent(LocalElasticAgent):
         4. re-rendezvous + start the new worker group.
         """
         with agent_events().duration("membership_restart", {}):
-            self._save_ckpt_to_storage()
+            # hold the pending round open while this node persists: a
+            # last-call timeout must not form the new world before our
+            # shards are on storage (ref: UcpRdzvManager blockable rdzv)
+            try:
+                self.client.block_rendezvous(self.config.node_rank, True)
+            except Exception:  # noqa: BLE001 — blocking is best-effort
+                logger.warning("rendezvous block failed", exc_info=True)
+            try:
+                self._save_ckpt_to_storage()
+            finally:
+                try:
+                    self.client.block_rendezvous(self.config.node_rank, False)
+                except Exception:  # noqa: BLE001
+                    logger.warning("rendezvous unblock failed", exc_info=True)
             descendants = self._worker_descendants()
             self._stop_workers(self._worker_group)
             self._sweep_orphans(descendants)

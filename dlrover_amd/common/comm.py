@@ -72,6 +72,16 @@ class CommWorldResponse(Message):
 
 
 @dataclass
+class RdzvBlockRequest(Message):
+    """Take/release a completion hold on the pending rendezvous round while
+    this node persists shards for a UCP reshard (ref: UcpRdzvManager)."""
+
+    node_rank: int = 0
+    blocked: bool = True
+    rdzv_name: str = ""
+
+
+@dataclass
 class WaitingNodeNumRequest(Message):
     node_id: int = 0
     local_world_size: int = 1

@@ -59,6 +59,27 @@ class RendezvousManager:
         self._start_round_time = 0.0
         self._node_unit = 1
         self._latest_join_time: Dict[int, float] = {}
+        # blockable rendezvous (ref: UcpRdzvManager rdzv_manager.py:583):
+        # while any node is persisting shards for a UCP reshard, the next
+        # round must NOT complete — a last-call timeout could otherwise
+        # form the new world without the slow-persisting node's shards on
+        # storage. Holders are node ranks; empty set = unblocked.
+        self._block_holders: set = set()
+
+    def block_rendezvous(self, node_rank: int, blocked: bool):
+        """A node takes/releases a completion hold on the pending round."""
+        with self._lock:
+            if blocked:
+                self._block_holders.add(node_rank)
+            else:
+                self._block_holders.discard(node_rank)
+            logger.info(
+                "[%s] rendezvous %s by node %s (%s holders)",
+                self.name,
+                "blocked" if blocked else "unblocked",
+                node_rank,
+                len(self._block_holders),
+            )
 
     # -- configuration ---------------------------------------------------------
 
@@ -97,6 +118,7 @@ class RendezvousManager:
             self._alive_nodes.discard(node_rank)
             # a dead node can no longer hold up or belong to a pending round
             self._waiting_nodes.pop(node_rank, None)
+            self._block_holders.discard(node_rank)
 
     # -- join / completion --------------------------------------------------------
 
@@ -123,6 +145,10 @@ class RendezvousManager:
         node_unit multiple)."""
         waiting = len(self._waiting_nodes)
         if waiting == 0:
+            return False
+        if self._block_holders:
+            # a UCP persist is in flight somewhere: hold the round open
+            # (dead holders are cleared by remove_alive_node)
             return False
         p = self._params
         alive = max(len(self._alive_nodes), 1)

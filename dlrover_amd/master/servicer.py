@@ -89,6 +89,10 @@ class MasterServicer:
             rnd = mgr.join_rendezvous(msg.node_rank, msg.local_world_size)
             self.master.job_manager.on_node_joined(msg.node_rank, msg.node_ip)
             return comm.JoinRendezvousResponse(round=rnd)
+        if isinstance(msg, comm.RdzvBlockRequest):
+            mgr = self.master.rdzv_managers[msg.rdzv_name or RendezvousName.TRAINING]
+            mgr.block_rendezvous(msg.node_rank, msg.blocked)
+            return None
         if isinstance(msg, comm.RendezvousParams):
             for mgr in self.master.rdzv_managers.values():
                 mgr.update_rdzv_params(

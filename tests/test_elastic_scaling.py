@@ -116,3 +116,49 @@ def test_lastcall_timeout_truncates_to_node_unit(monkeypatch):
     _, _, world = mgr.get_comm_world(0)
     assert sorted(world) == [0, 1, 2, 3], world  # truncated, rank 4 dropped
     assert mgr.num_nodes_waiting() == 1  # the dropped node waits for next round
+
+
+def test_blockable_rendezvous_holds_round_open():
+    """A UCP-persist hold keeps the pending round from completing — even
+    past the last-call timeout — until every holder releases (or dies);
+    ref UcpRdzvManager rdzv_manager.py:583."""
+    import time as _time
+
+    from dlrover_amd.master.elastic.rdzv_manager import (
+        ElasticTrainingRendezvousManager,
+    )
+
+    mgr = ElasticTrainingRendezvousManager()
+    mgr.update_rdzv_params(1, 4, waiting_timeout=0.01, node_unit=1)
+    mgr.block_rendezvous(0, True)
+    mgr.join_rendezvous(0, 8)
+    mgr.join_rendezvous(1, 8)
+    _time.sleep(0.05)  # well past last-call
+    rnd, _, world = mgr.get_comm_world(0)
+    assert world == {}, "blocked round must not complete"
+    # a second holder: releasing only one keeps the hold
+    mgr.block_rendezvous(1, True)
+    mgr.block_rendezvous(0, False)
+    _, _, world = mgr.get_comm_world(0)
+    assert world == {}
+    mgr.block_rendezvous(1, False)
+    _, _, world = mgr.get_comm_world(0)
+    assert set(world) == {0, 1}
+
+
+def test_blockable_rendezvous_dead_holder_released():
+    """A holder that dies must not wedge the round forever: the job
+    manager's remove_alive_node clears its hold."""
+    from dlrover_amd.master.elastic.rdzv_manager import (
+        ElasticTrainingRendezvousManager,
+    )
+
+    mgr = ElasticTrainingRendezvousManager()
+    mgr.update_rdzv_params(1, 4, waiting_timeout=600, node_unit=1)
+    mgr.block_rendezvous(3, True)
+    mgr.join_rendezvous(0, 8)
+    _, _, world = mgr.get_comm_world(0)
+    assert world == {}
+    mgr.remove_alive_node(3)  # holder died
+    _, _, world = mgr.get_comm_world(0)
+    assert set(world) == {0}
