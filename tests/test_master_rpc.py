@@ -159,3 +159,19 @@ def test_barrier_and_sync(client):
     assert client.is_sync_finished("warmup")
     assert not client.barrier("b1")
     assert client.barrier("b1", notify=True)
+
+
+def test_rdzv_block_over_rpc(master, client):
+    """block_rendezvous travels the wire: a held round does not complete,
+    release lets it form (ref UcpRdzvManager)."""
+    client.report_rdzv_params(1, 4, 0.01, 1)
+    client.block_rendezvous(0, True)
+    client.join_rendezvous(0, 8)
+    import time as _t
+
+    _t.sleep(0.05)
+    _, _, world = client.get_comm_world(RendezvousName.TRAINING, 0)
+    assert world == {}
+    client.block_rendezvous(0, False)
+    _, _, world = client.get_comm_world(RendezvousName.TRAINING, 0)
+    assert set(world) == {0}
