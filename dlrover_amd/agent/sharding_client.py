@@ -12,7 +12,6 @@ from typing import Optional
 
 from dlrover_amd.agent.master_client import MasterClient
 from dlrover_amd.common import comm
-from dlrover_amd.common.log import logger
 
 
 class ShardingClient:

@@ -17,7 +17,6 @@ import time
 import uuid
 from typing import Any, Dict, Optional
 
-from dlrover_amd.common.log import logger
 
 
 class AsyncExporter:

@@ -10,7 +10,6 @@ from typing import Iterator, Optional
 
 from dlrover_amd.common import comm
 from dlrover_amd.common.constants import NodeEventType, NodeStatus
-from dlrover_amd.common.log import logger
 
 _PHASE_TO_STATUS = {
     "Pending": NodeStatus.PENDING,

@@ -14,7 +14,7 @@ so a full (tp=1) checkpoint reshards by row slicing (megatron engine).
 """
 
 import math
-from typing import List, Optional
+from typing import Optional
 
 import torch
 import torch.nn as nn
@@ -23,11 +23,7 @@ from dlrover_amd.models.llama import LlamaConfig, RMSNorm
 from dlrover_amd.ops import causal_softmax, cross_entropy_loss, rope_rotate, swiglu
 from dlrover_amd.ops.api import build_rope_cache
 from dlrover_amd.parallel.pgroups import ParallelGroups
-from dlrover_amd.parallel.tp import (
-    ColumnParallelLinear,
-    RowParallelLinear,
-    copy_to_tp,
-)
+from dlrover_amd.parallel.tp import ColumnParallelLinear, RowParallelLinear
 
 
 class ParallelAttention(nn.Module):

@@ -7,7 +7,6 @@ block needs exactly two all-reduces per direction (after o_proj and after
 down_proj) — the xGMI-friendly minimum.
 """
 
-from typing import Optional
 
 import torch
 import torch.distributed as dist

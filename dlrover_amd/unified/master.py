@@ -15,11 +15,11 @@ import multiprocessing as mp
 import os
 import threading
 import time
-from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from dataclasses import dataclass
+from typing import List, Optional
 
 from dlrover_amd.common.log import logger
-from dlrover_amd.unified.api import DLJob, WorkloadDesc
+from dlrover_amd.unified.api import DLJob
 from dlrover_amd.unified.scheduler import NodeSpec, Placement, Scheduler
 
 

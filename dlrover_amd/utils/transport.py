@@ -22,7 +22,6 @@ import threading
 import time
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
 from typing import Callable, Optional
-from urllib import error as urlerror
 from urllib import request as urlrequest
 
 from dlrover_amd.common import serialize
