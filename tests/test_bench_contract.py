@@ -62,3 +62,18 @@ def test_bench_under_torchrun(tmp_path):
     r = _parse_bench(out.stdout)
     assert r["n_gpus"] == 2
     assert r["config"]["global_batch"] == 2 * 2  # ws=2 x default batch 2 (weak scaling)
+
+
+@pytest.mark.timeout(120)
+def test_bench_refuses_world_size_mismatch():
+    """--gpus N with a conflicting torchrun WORLD_SIZE must refuse rather
+    than silently measure a different world size."""
+    env = dict(os.environ, RANK="0", WORLD_SIZE="1", LOCAL_RANK="0",
+               MASTER_ADDR="127.0.0.1", MASTER_PORT="29742")
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--gpus", "2", "--model", "tiny",
+         "--steps", "2", "--warmup", "0", "--seq", "16"],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=90,
+    )
+    assert out.returncode != 0
+    assert "refusing" in out.stderr or "WORLD_SIZE" in out.stderr, out.stderr[-1500:]
