@@ -114,3 +114,46 @@ def test_prepare_gates_optimizer_on_accum_boundary(monkeypatch):
     assert steps_fired == 2
     assert trainer.global_step == 2
     assert base_opt.param_groups[0]["lr"] == 0.1 * 0.5 ** 2
+
+
+def test_elastic_dataloader_applies_versioned_batch_size(monkeypatch):
+    """ElasticDataLoader applies a master-pushed batch size once per
+    version (ref: dataloader.py + ParallelConfig versioning); an equal or
+    older version is a no-op."""
+    import torch
+    from torch.utils.data import TensorDataset
+
+    from dlrover_amd.common import comm
+    from dlrover_amd.trainer.elastic.dataloader import ElasticDataLoader
+
+    ds = TensorDataset(torch.arange(64).float())
+    dl = ElasticDataLoader(ds, batch_size=4)
+
+    class FakeClient:
+        calls = 0
+
+        def get_paral_config(self):
+            FakeClient.calls += 1
+            return comm.ParallelConfig(
+                dataloader=comm.DataLoaderConfig(batch_size=16, version=1)
+            )
+
+    from dlrover_amd.agent import master_client as mc
+
+    monkeypatch.setattr(
+        mc.MasterClient, "singleton_instance", classmethod(
+            lambda cls: FakeClient()
+        ),
+    )
+    dl.update_batch_size()  # pulls version 1 -> applies 16
+    assert dl.batch_size == 16
+    assert next(iter(dl))[0].shape[0] == 16
+    dl.update_batch_size()  # same version -> no re-apply
+    assert FakeClient.calls == 2 and dl.batch_size == 16
+    # explicit direct set still works
+    dl.update_batch_size(8)
+    assert next(iter(dl))[0].shape[0] == 8
+    # autotune honors the env gate
+    monkeypatch.delenv("DLROVER_AUTO_TUNE", raising=False)
+    dl.maybe_autotune()
+    assert FakeClient.calls == 2  # gate off: no poll
