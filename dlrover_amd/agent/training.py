@@ -626,7 +626,11 @@ def launch_agent(
             except OSError as e:
                 logger.warning("prometheus endpoint unavailable: %s", e)
 
-    if config.network_check:
+    if config.network_check or config.comm_perf_test:
+        # --comm-perf-test runs the same probe rounds (the probe reports
+        # matmul TFLOPS + allreduce busbw — ref comm_perf_check
+        # training.py:2337); --network-check additionally acts on
+        # fault/straggler verdicts inside run_network_check
         from dlrover_amd.agent.node_check_agent import run_network_check
 
         run_network_check(config, client)
