@@ -33,18 +33,36 @@ class ResourcePlan:
     comment: str = ""
 
 
+class QuotaChecker:
+    """Free-capacity oracle bounding scale-up (ref: master/cluster/quota.py).
+    The default assumes capacity is always available; platforms plug their
+    own."""
+
+    def get_free_node_num(self) -> int:
+        import sys
+
+        return sys.maxsize
+
+
+class NoFreeQuotaChecker(QuotaChecker):
+    def get_free_node_num(self) -> int:
+        return 0
+
+
 class LocalResourceOptimizer:
     """Single-job heuristics (ref: PSLocalOptimizer, reduced to the
     allreduce/worker case that matters for GPU training)."""
 
     def __init__(self, perf: PerfMonitor, ctx: Optional[JobContext] = None,
-                 min_nodes: int = 1, max_nodes: int = 1, job_name: str = ""):
+                 min_nodes: int = 1, max_nodes: int = 1, job_name: str = "",
+                 quota: Optional[QuotaChecker] = None):
         import os
 
         self.perf = perf
         self.ctx = ctx or JobContext.singleton_instance()
         self.min_nodes = min_nodes
         self.max_nodes = max_nodes
+        self.quota = quota or QuotaChecker()
         self.job_name = job_name or os.getenv("ELASTIC_JOB_NAME", "job")
         # optimizeMode=cluster: a Brain endpoint outranks local heuristics
         # (ref: brain_optimizer.py wrapper; falls back when unreachable)
@@ -89,9 +107,12 @@ class LocalResourceOptimizer:
                 node_count=len(running),
                 comment=f"{len(pending)} nodes pending: shrink to running set",
             )
-        if not pending and len(alive) < self.max_nodes:
+        # scale-up bounded by the platform quota (ref: cluster/quota.py)
+        free = self.quota.get_free_node_num()
+        target = min(self.max_nodes, len(alive) + max(0, free))
+        if not pending and len(alive) < target:
             return ResourcePlan(
-                node_count=self.max_nodes,
+                node_count=target,
                 comment="capacity available: grow to max_nodes",
             )
         return None

@@ -198,3 +198,32 @@ def test_optimizer_consults_brain_when_configured(ctx, monkeypatch):
         assert plan.node_count == 4  # 2 alive -> brain doubles
     finally:
         svc.stop()
+
+
+def test_quota_bounds_growth(ctx):
+    """Scale-up is clamped by the platform quota (ref: cluster/quota.py):
+    zero free capacity means no grow plan even below max_nodes."""
+    from dlrover_amd.common.constants import NodeStatus, NodeType
+    from dlrover_amd.common.node import Node
+    from dlrover_amd.master.auto_scale import (
+        LocalResourceOptimizer,
+        NoFreeQuotaChecker,
+    )
+    from dlrover_amd.master.perf_monitor import PerfMonitor
+
+    for i in range(2):
+        n = Node(NodeType.WORKER, i)
+        n.update_status(NodeStatus.RUNNING)
+        ctx.update_node(n)
+    opt = LocalResourceOptimizer(PerfMonitor(), ctx=ctx, min_nodes=1,
+                                 max_nodes=8, quota=NoFreeQuotaChecker())
+    assert opt.generate_plan() is None  # no free nodes -> hold
+
+    class TwoFree:
+        def get_free_node_num(self):
+            return 2
+
+    opt2 = LocalResourceOptimizer(PerfMonitor(), ctx=ctx, min_nodes=1,
+                                  max_nodes=8, quota=TwoFree())
+    plan = opt2.generate_plan()
+    assert plan is not None and plan.node_count == 4  # 2 alive + 2 free
