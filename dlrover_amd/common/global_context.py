@@ -41,7 +41,7 @@ class Context:
             os.getenv("DLROVER_HANG_DOWNTIME", DefaultValues.HANG_DOWNTIME_SECS)
         )
         self.seconds_interval_collect = DefaultValues.SECONDS_INTERVAL_COLLECT
-        self.auto_worker_enabled = False
+        self.auto_worker_enabled = os.getenv("DLROVER_AUTO_WORKER", "") == "1"
         self.auto_ps_enabled = False
         self.pre_check_operators = []
         self.is_tfv1_ps = False
