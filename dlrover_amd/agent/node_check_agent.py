@@ -93,6 +93,14 @@ def run_network_check(
             )
     stragglers = client.check_straggler()
     if config.node_rank in stragglers:
+        if getattr(config, "exclude_straggler", False):
+            # ref: --exclude-straggler (elastic_run.py:170) — the agent
+            # refuses to train on a straggler node so the platform
+            # reschedules it; default is observe-and-continue
+            raise NodeCheckFailedError(
+                f"node {config.node_rank} is a straggler and "
+                "--exclude-straggler is set"
+            )
         logger.warning("this node is a straggler (continuing; master decides)")
     fault_nodes, _ = client.check_fault_node()
     if config.node_rank in fault_nodes:

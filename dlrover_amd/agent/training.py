@@ -68,6 +68,7 @@ class ElasticLaunchConfig:
     node_unit: int = 1
     network_check: bool = False
     comm_perf_test: bool = False
+    exclude_straggler: bool = False
     run_id: str = "dlrover"
     checkpoint_dir: str = "/tmp/dlrover_amd_ckpt"
     log_dir: Optional[str] = None

@@ -62,7 +62,19 @@ def parse_args(argv: Optional[List[str]] = None):
     p.add_argument("--waiting-timeout", type=float, default=60.0)
     p.add_argument("--network-check", "--network_check", action="store_true",
                    help="run matmul+allreduce probes before training")
-    p.add_argument("--comm-perf-test", action="store_true")
+    p.add_argument("--comm-perf-test", "--comm_perf_test", action="store_true")
+    p.add_argument("--precheck", type=int, default=0, choices=[0, 1, 2],
+                   help="0=off, 1=comm probes, 2=probes + straggler "
+                        "exclusion (ref precheck levels)")
+    p.add_argument("--exclude-straggler", "--exclude_straggler",
+                   action="store_true",
+                   help="refuse to train on a straggler node (the probe "
+                        "raises instead of observing)")
+    p.add_argument("--training-port", "--training_port", type=int, default=0,
+                   help="accepted for reference-CLI compatibility (NPU port "
+                        "sync; not needed on MI355X)")
+    p.add_argument("--membind-policy", "--membind_policy", default="",
+                   help="NUMA memory policy hint for --numa-affinity")
     p.add_argument("--node-unit", "--node_unit", type=int, default=1,
                    help="world size must be a multiple of this")
     p.add_argument("--auto-config", action="store_true",
@@ -221,6 +233,14 @@ def run(args) -> int:
 
     wait_pre_check(client)
 
+    # --precheck levels map onto the probe flags (ref precheck semantics)
+    if args.precheck >= 1:
+        args.network_check = True
+    if args.precheck >= 2:
+        args.exclude_straggler = True
+    if args.membind_policy:
+        os.environ["DLROVER_MEMBIND_POLICY"] = args.membind_policy
+
     min_nodes, max_nodes = parse_nnodes(args.nnodes)
     if args.auto_config and args.nproc_per_node <= 0:
         args.nproc_per_node = 0  # auto_configure() fills from device count
@@ -238,6 +258,7 @@ def run(args) -> int:
         node_unit=args.node_unit,
         network_check=args.network_check,
         comm_perf_test=args.comm_perf_test,
+        exclude_straggler=args.exclude_straggler,
         checkpoint_dir=args.checkpoint_dir,
         log_dir=args.log_dir,
         numa_affinity=args.numa_affinity,

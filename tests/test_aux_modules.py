@@ -295,3 +295,15 @@ def test_dlrover_run_accepts_torchrun_flags():
     assert args.no_python
     assert args.training_script == "train.sh"
     assert args.training_script_args == ["--flag"]
+
+
+def test_precheck_level_flags():
+    """--precheck levels map onto the probe flags; --exclude-straggler and
+    reference-compat flags are accepted (ref elastic_run.py:137-215)."""
+    from dlrover_amd.trainer.elastic_run import parse_args
+
+    args = parse_args(["--precheck", "2", "--training-port", "60001",
+                       "--membind-policy", "bind", "train.py"])
+    assert args.precheck == 2 and args.training_port == 60001
+    args2 = parse_args(["--exclude-straggler", "train.py"])
+    assert args2.exclude_straggler
