@@ -67,7 +67,7 @@ def job_args_from_elasticjob_cr(cr: dict) -> JobArgs:
         args.node_groups[role] = NodeGroupResource(
             count=replicas,
             node_resource=NodeResource(
-                cpu=float(str(res.get("cpu", 8)).rstrip("m") or 8),
+                cpu=_parse_cpu(res.get("cpu", 8)),
                 memory_mb=_parse_mem(res.get("memory", "65536Mi")),
                 gpu_num=int(res.get("amd.com/gpu", res.get("nvidia.com/gpu", 0)) or 0),
                 gpu_type="amd.com/gpu" if "amd.com/gpu" in res else "",
@@ -88,6 +88,17 @@ def _parse_mem(v) -> int:
         return int(float(s) / (1 << 20))
     except ValueError:
         return 65536
+
+
+def _parse_cpu(v) -> float:
+    """k8s CPU quantity: '500m' = 0.5 cores, '4' = 4 cores."""
+    s = str(v)
+    try:
+        if s.endswith("m"):
+            return float(s[:-1]) / 1000.0
+        return float(s)
+    except ValueError:
+        return 8.0
 
 
 def new_job_args(platform: str, cr: Optional[dict] = None) -> JobArgs:

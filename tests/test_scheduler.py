@@ -35,3 +35,23 @@ def test_job_args_from_cr():
     assert g.node_resource.gpu_num == 8
     assert g.node_resource.memory_mb == 1024 * 1024
     assert g.node_resource.gpu_type == "amd.com/gpu"
+
+
+def test_cpu_quantity_parsing():
+    """k8s millicore CPU quantities: '500m' is half a core, not 500."""
+    from dlrover_amd.scheduler import job_args_from_elasticjob_cr
+
+    cr = {
+        "metadata": {"name": "j"},
+        "spec": {"replicaSpecs": {"worker": {
+            "replicas": 2,
+            "template": {"spec": {"containers": [{
+                "resources": {"limits": {"cpu": "500m", "memory": "2Gi",
+                                         "amd.com/gpu": 1}}
+            }]}},
+        }}},
+    }
+    g = job_args_from_elasticjob_cr(cr).node_groups["worker"]
+    assert g.node_resource.cpu == 0.5
+    assert g.node_resource.memory_mb == 2048
+    assert g.node_resource.gpu_num == 1
