@@ -20,13 +20,16 @@ WORKER = os.path.join(ROOT, "tests", "workers", "ws2_worker.py")
 def _visible_devices() -> int:
     """Device count as a FRESH process sees it (CPX partitioning changes it
     under a running session; this test spawns fresh workers anyway)."""
-    out = subprocess.run(
-        [sys.executable, "-c", "import torch; print(torch.cuda.device_count())"],
-        capture_output=True, text=True, timeout=120,
-    )
     try:
+        out = subprocess.run(
+            [sys.executable, "-c",
+             "import torch; print(torch.cuda.device_count())"],
+            # a COLD `import torch` on a fresh box can take ~2 min of image
+            # page-in; a timeout here must degrade to skip, never error
+            capture_output=True, text=True, timeout=300,
+        )
         return int(out.stdout.strip().splitlines()[-1])
-    except (ValueError, IndexError):
+    except (ValueError, IndexError, subprocess.TimeoutExpired):
         return 0
 
 
